@@ -1,0 +1,127 @@
+"""Simulation context: wires cell, G-vectors, FFT grids, device, parallel env.
+
+Reference behavior: src/context/simulation_context.hpp:154 —
+Simulation_context owns communicators, FFT grids, Gvec sets, unit cell,
+eigensolvers, memory pools; initialize() at simulation_context.cpp:154.
+
+MI355X design: one process per GPU; `torch.distributed` (RCCL over xGMI
+when backend "nccl", gloo on CPU) replaces the reference's MPI
+communicator splits (simulation_context.cpp:1301-1334). The k-point
+data-parallel group is the world; band parallelism inside a k group is a
+planned split (comm_band) over the same process group.
+"""
+
+from __future__ import annotations
+
+import math
+
+import numpy as np
+import torch
+
+from .cell import UnitCell
+from .config import Config
+from .core.fft import SphericalFFT
+from .core.gvec import Gvec, fft_grid_dims
+from . import xc as xc_mod
+
+
+class SimulationContext:
+    def __init__(self, cfg: Config, unit_cell: UnitCell | None = None,
+                 base_dir: str = ".", device: str | None = None):
+        self.cfg = cfg
+        self.unit_cell = unit_cell or UnitCell.from_config(cfg, base_dir)
+
+        pu = cfg.control.processing_unit
+        if device is None:
+            if pu == "cpu":
+                device = "cpu"
+            elif pu == "gpu":
+                device = "cuda"
+            else:
+                device = "cuda" if torch.cuda.is_available() else "cpu"
+        self.device = torch.device(device)
+
+        p = cfg.parameters
+        self.pw_cutoff = float(p.pw_cutoff)
+        self.gk_cutoff = float(p.gk_cutoff)
+        if self.pw_cutoff < 2 * self.gk_cutoff:
+            # reference insists fine grid covers products of wavefunctions
+            self.pw_cutoff = 2 * self.gk_cutoff
+        self.num_mag_dims = int(p.num_mag_dims)
+        self.num_spins = 2 if self.num_mag_dims > 0 else 1
+        self.num_spinors = 2 if self.num_mag_dims == 3 else 1
+        self.xc_names = list(p.xc_functionals)
+        self.is_gga = xc_mod.is_gga(self.xc_names)
+
+        uc = self.unit_cell
+        # fine G set (density/potential), on fine FFT grid
+        fine_dims = fft_grid_dims(uc.lattice, self.pw_cutoff)
+        self.gvec_fine = Gvec(uc.recip, self.pw_cutoff, dims=fine_dims, device=device)
+        # coarse grid for wavefunction FFTs: cutoff 2*gk (simulation_context.hpp:494)
+        coarse_dims = fft_grid_dims(uc.lattice, 2 * self.gk_cutoff)
+        self.coarse_dims = coarse_dims
+        self.gvec_coarse = Gvec(uc.recip, 2 * self.gk_cutoff, dims=coarse_dims, device=device)
+        self.fft_fine = SphericalFFT(self.gvec_fine)
+        self.fft_coarse = SphericalFFT(self.gvec_coarse)
+        # coarse -> fine index map (gvec_base_mapping analogue)
+        self.coarse_to_fine = self.gvec_coarse.gvec_map_to(self.gvec_fine)
+
+        # number of bands (simulation_context.cpp:333-352)
+        nel = uc.num_electrons
+        nbnd = int(nel / 2.0) + max(10, int(0.1 * nel))
+        if self.num_mag_dims == 3:
+            nbnd *= 2
+        self.num_bands = int(p.num_fv_states) if p.num_fv_states > 0 else (
+            int(p.num_bands) if p.num_bands > 0 else nbnd)
+        self.max_occupancy = 2.0 if self.num_mag_dims == 0 else 1.0
+
+        # structure phase factors on the fine sphere: e^{iG·τ_a} per atom
+        self._phase_fine = None
+
+        self.dtype = torch.complex128
+        self.rdtype = torch.float64
+
+    # -- structure factors -------------------------------------------------
+
+    def phase_factors_fine(self) -> torch.Tensor:
+        """e^{-iG·τ_a}, shape [natom, nG_fine] (on device).
+
+        Convention: f(G) = (1/Ω) Σ_a ff_a(|G|) e^{-iG·τ_a}
+        (make_periodic_function.hpp:35-44 uses conj(e^{+iG·τ})).
+        """
+        if self._phase_fine is None:
+            uc = self.unit_cell
+            tau = uc.atom_positions_frac()  # fractional
+            # G·r = 2π m·τ_frac
+            m = self.gvec_fine.miller.astype(np.float64)
+            ph = np.exp(-2j * math.pi * (tau @ m.T))
+            self._phase_fine = torch.from_numpy(ph).to(self.device)
+        return self._phase_fine
+
+    def make_periodic_function(self, form_factors: dict[str, np.ndarray]) -> torch.Tensor:
+        """Assemble f(G) = (4π/Ω) Σ_a ff_{type(a)}(|G|) e^{-iG·τ_a} on the fine sphere.
+
+        form_factors: per atom-type label, values on the G-shells
+        (self.gvec_fine.shell_len); reference make_periodic_function.hpp:56-82.
+        """
+        uc = self.unit_cell
+        gl = self.gvec_fine
+        out = np.zeros(gl.num_gvec, dtype=np.complex128)
+        tau = uc.atom_positions_frac()
+        m = gl.miller.astype(np.float64)
+        for lab in uc.type_labels:
+            ia = uc.atoms_of_type(lab)
+            if len(ia) == 0:
+                continue
+            ph = np.exp(-2j * math.pi * (tau[ia] @ m.T)).sum(axis=0)  # [nG]
+            ff_g = form_factors[lab][gl.shell_of_g]
+            out += ff_g * ph
+        out *= 4 * math.pi / uc.omega
+        return torch.from_numpy(out).to(self.device)
+
+    # -- integration -------------------------------------------------------
+
+    def integrate_rg_fine(self, f: torch.Tensor) -> float:
+        """∫ f dΩ over the cell from fine-grid real-space values."""
+        n = self.fft_fine.size
+        return float(f.sum().real) * self.unit_cell.omega / n

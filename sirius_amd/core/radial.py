@@ -1,0 +1,129 @@
+"""Radial grids, spline quadrature and radial integrals vs |G|.
+
+Reference behavior: src/radial/spline.hpp (cubic-spline quadrature of
+radial integrands) and src/radial/radial_integrals.{hpp,cpp} (tabulated
+<j_l|f> form factors as functions of q=|G| or |G+k|).
+
+We vectorize: one scipy CubicSpline solve over a [nr, nq] integrand batch
+replaces the reference's per-q spline loop; evaluation at arbitrary q then
+goes through a dense table + cubic interpolation, or direct evaluation on
+the exact set of |G|-shell lengths (no interpolation error at all — the
+shell count is small enough on one node that we simply evaluate exactly).
+"""
+
+from __future__ import annotations
+
+import numpy as np
+from scipy.interpolate import CubicSpline
+
+
+def spline_integrate(r: np.ndarray, f: np.ndarray, axis: int = 0) -> np.ndarray:
+    """∫ f dr over the full grid via cubic-spline quadrature.
+
+    f may be a batch: shape [nr, ...] with axis=0.
+    """
+    cs = CubicSpline(r, f, axis=axis)
+    return cs.integrate(r[0], r[-1])
+
+
+def sbessel(l: int, x: np.ndarray) -> np.ndarray:
+    """Spherical Bessel j_l(x), stable at x→0 (reference: src/core/sf/sbessel.cpp)."""
+    out = np.empty_like(x)
+    small = np.abs(x) < 1e-9
+    xs = np.where(small, 1.0, x)
+    if l == 0:
+        out = np.sin(xs) / xs
+        out[small] = 1.0
+    elif l == 1:
+        out = np.sin(xs) / xs**2 - np.cos(xs) / xs
+        out[small] = 0.0
+    elif l == 2:
+        out = (3.0 / xs**2 - 1.0) * np.sin(xs) / xs - 3.0 * np.cos(xs) / xs**2
+        out[small] = 0.0
+    elif l == 3:
+        out = (15.0 / xs**3 - 6.0 / xs) * np.sin(xs) / xs - (15.0 / xs**2 - 1.0) * np.cos(xs) / xs
+        out[small] = 0.0
+    else:
+        from scipy.special import spherical_jn
+
+        out = spherical_jn(l, xs)
+        out[small] = 0.0
+    return out
+
+
+class RadialIntegrals:
+    """Form factors f(q) = ∫ f(r) j_l(q r) r^p dr on a set of q points.
+
+    All integrals are evaluated exactly at the requested q values with
+    spline quadrature (batched), rather than through the reference's
+    uniform-q interpolation table. ~O(nr*nq) memory per call.
+    """
+
+    @staticmethod
+    def sbessel_transform(l: int, r: np.ndarray, fr: np.ndarray, q: np.ndarray,
+                          rpow: int = 2) -> np.ndarray:
+        """∫ fr(r) j_l(q r) r^rpow dr for each q. fr given on grid r.
+
+        Returns array [nq].
+        """
+        q = np.asarray(q, dtype=np.float64)
+        x = np.outer(r, q)  # [nr, nq]
+        jl = sbessel(l, x)
+        integrand = jl * (fr * r**rpow)[:, None]
+        return spline_integrate(r, integrand, axis=0)
+
+    @staticmethod
+    def vloc_q(r: np.ndarray, vloc_r: np.ndarray, zn: float, q: np.ndarray,
+               r_cut: float = 10.0) -> np.ndarray:
+        """Local-potential form factor (without 4π/Ω and without structure phase).
+
+        Matches the reference exactly (src/radial/radial_integrals.cpp:240-304,
+        value() src/radial/radial_integrals.hpp:386-411):
+          q=0:  ∫ (r·V(r) + Z)·r dr
+          q>0:  [∫ (r·V(r) + Z·erf(r))·sin(q r) dr] / q  −  Z·exp(−q²/4)/q²
+        The integration is truncated at ~10 a.u. (QE-inherited hack for
+        diverging tails; settings.pseudo_grid_cutoff).
+        """
+        from scipy.special import erf
+
+        n = int(np.searchsorted(r, r_cut, side="right"))
+        n = max(n, 2)
+        rr = r[:n]
+        vv = vloc_r[:n]
+        q = np.asarray(q, dtype=np.float64)
+        out = np.empty_like(q)
+        nz = q > 1e-12
+        qnz = q[nz]
+        sin_qr = np.sin(np.outer(rr, qnz))  # [nr, nqnz]
+        integrand = (rr * vv + zn * erf(rr))[:, None] * sin_qr
+        vals = spline_integrate(rr, integrand, axis=0)
+        out[nz] = vals / qnz - zn * np.exp(-qnz**2 / 4) / qnz**2
+        if (~nz).any():
+            out[~nz] = spline_integrate(rr, (rr * vv + zn) * rr)
+        return out
+
+    @staticmethod
+    def rho_q(r: np.ndarray, rho4pir2: np.ndarray, q: np.ndarray) -> np.ndarray:
+        """Charge-density form factor from the UPF convention 4π r² ρ(r).
+
+        f(q) = ∫ ρ(r) j0(qr) r² dr = (1/4π) ∫ [4π r² ρ] j0(qr) dr
+        (reference: Radial_integrals_rho_pseudo::generate,
+        src/radial/radial_integrals.cpp:133-158 — the UPF
+        total_charge_density array already contains 4π r² ρ, and the
+        stored value divides by 4π). The caller assembles
+        ρ(G) = (4π/Ω) Σ_a f_a(|G|) e^{-iG·τ_a}.
+        """
+        q = np.asarray(q, dtype=np.float64)
+        out = np.empty_like(q)
+        nz = q > 1e-12
+        qnz = q[nz]
+        # j0(qr) r^2 rho = (4pi r^2 rho) * j0 / 4pi -> use sin(qr)/(qr)
+        x = np.outer(r, qnz)
+        j0 = np.ones_like(x)
+        mask = x > 1e-12
+        j0[mask] = np.sin(x[mask]) / x[mask]
+        integrand = rho4pir2[:, None] * j0
+        out[nz] = spline_integrate(r, integrand, axis=0)
+        if (~nz).any():
+            out[~nz] = spline_integrate(r, rho4pir2)
+        return out / (4 * np.pi)

@@ -1,0 +1,242 @@
+"""SCF ground-state driver.
+
+Reference behavior: src/dft/dft_ground_state.cpp —
+initial_state (:23): atomic-density superposition, potential, LCAO
+subspace init; find (:179): SCF loop = build H0 → diagonalize (Davidson)
+→ Fermi occupancies → new density → mix → adaptive solver tolerance
+(:253-262) → new potential → scf-correction energy (:320-323) → total
+energy (energy.cpp:142-165) → convergence on |ΔE| and RMS (:349-356).
+"""
+
+from __future__ import annotations
+
+import json
+import math
+import time
+
+import numpy as np
+import torch
+
+from .core import ylm as ylm_mod
+from .core.radial import RadialIntegrals
+from .davidson import davidson
+from .density import Density
+from .hamiltonian import Hamiltonian0
+from .kpoint import KPointSet
+from .parallel import get_comm
+from .potential import Potential
+
+
+def atomic_orbitals(ctx, kp) -> torch.Tensor:
+    """LCAO trial orbitals χ_lm(G+k) for all atoms [n_ao, nGk]
+    (reference: initialize_subspace.hpp:27 via Radial_integrals_atomic_wf)."""
+    uc = ctx.unit_cell
+    g = kp.gkvec
+    glen = g.gk_len
+    _, theta, phi = ylm_mod.spherical_coords(g.gkvec_cart)
+    lmax = max((w.l for at in uc.atom_types.values() for w in at.atomic_wfs), default=0)
+    rl = ylm_mod.rlm(lmax, theta, phi)
+    cols_t = {}
+    for lab, at in uc.atom_types.items():
+        cols = []
+        for w in at.atomic_wfs:
+            f = RadialIntegrals.sbessel_transform(w.l, at.r, w.f_r, glen, rpow=1)
+            z = (-1j) ** w.l * (4 * math.pi / math.sqrt(uc.omega))
+            for m in range(-w.l, w.l + 1):
+                cols.append(z * rl[:, ylm_mod.lm_index(w.l, m)] * f)
+        cols_t[lab] = np.stack(cols, axis=1) if cols else np.zeros((len(glen), 0),
+                                                                   dtype=np.complex128)
+    blocks = []
+    mk = (g.miller + g.k_frac).astype(np.float64)
+    for lab, tau in uc.atoms:
+        if cols_t[lab].shape[1] == 0:
+            continue
+        phase = np.exp(-2j * math.pi * (mk @ tau))
+        blocks.append((cols_t[lab] * phase[:, None]).T)  # [nao_a, nGk]
+    if not blocks:
+        return torch.zeros(0, g.num_gvec, dtype=ctx.dtype, device=ctx.device)
+    return torch.from_numpy(np.concatenate(blocks, axis=0)).to(ctx.device)
+
+
+def initialize_subspace(ctx, kp, hk):
+    """LCAO + random trial basis, one Rayleigh-Ritz step
+    (initialize_subspace.hpp:27-150)."""
+    nb = ctx.num_bands
+    init = ctx.cfg.iterative_solver.init_subspace
+    phi = atomic_orbitals(ctx, kp) if init == "lcao" else \
+        torch.zeros(0, kp.num_gkvec, dtype=ctx.dtype, device=ctx.device)
+    n_ao = phi.shape[0]
+    if n_ao < nb:
+        gen = torch.Generator(device="cpu").manual_seed(12345 + kp.gkvec.num_gvec)
+        rnd = torch.randn(nb - n_ao, kp.num_gkvec, 2, generator=gen,
+                          dtype=torch.float64)
+        rnd = torch.view_as_complex(rnd.contiguous()).to(ctx.device)
+        # damp high-G components for smoother start
+        damp = 1.0 / (1.0 + kp.gkvec.gk2_t)
+        rnd = rnd * damp
+        phi = torch.cat([phi, rnd], dim=0)
+
+    from .davidson import _ortho_block
+    phi, _ = _ortho_block(phi, None, None, None)
+    hphi, sphi = hk.apply_h_s(phi)
+    if sphi is None:
+        H = phi.conj() @ hphi.T
+        H = 0.5 * (H + H.conj().T)
+        evals, Z = torch.linalg.eigh(H)
+    else:
+        H = phi.conj() @ hphi.T
+        S = phi.conj() @ sphi.T
+        H = 0.5 * (H + H.conj().T)
+        S = 0.5 * (S + S.conj().T)
+        L = torch.linalg.cholesky(S)
+        Li = torch.linalg.solve_triangular(
+            L, torch.eye(L.shape[0], dtype=L.dtype, device=L.device), upper=False)
+        Ht = Li @ H @ Li.conj().T
+        evals, Zt = torch.linalg.eigh(0.5 * (Ht + Ht.conj().T))
+        Z = Li.conj().T @ Zt
+    nsel = min(nb, phi.shape[0])
+    psi = Z[:, :nsel].T @ phi
+    if nsel < nb:
+        raise RuntimeError("not enough trial orbitals for requested bands")
+    for ispn in range(ctx.num_spins):
+        kp.psi[ispn] = psi
+        kp.eigvals[ispn] = evals[:nb].real.cpu().numpy()
+
+
+def diagonalize(ctx, h0: Hamiltonian0, kset: KPointSet, itsol_tol: float) -> bool:
+    """Davidson for all local k-points/spins (reference diagonalize.hpp)."""
+    itso = ctx.cfg.iterative_solver
+    empy_tol = max(itsol_tol * itso.tolerance_ratio, itso.empty_states_tolerance)
+    all_conv = True
+    for kp in kset:
+        hk = h0(kp)
+        h_diag = hk.h_diag()
+        o_diag = hk.o_diag()
+        for ispn in range(ctx.num_spins):
+            res = davidson(
+                lambda p: hk.apply_h_s(p, ispn),
+                kp.psi[ispn], h_diag, o_diag, occ=kp.occ[ispn],
+                tol_occ=itsol_tol, tol_empty=empy_tol,
+                num_steps=itso.num_steps, subspace_size=itso.subspace_size,
+                min_occupancy=itso.min_occupancy, extra_ortho=itso.extra_ortho)
+            kp.psi[ispn] = res.psi
+            kp.eigvals[ispn] = res.eval
+            all_conv = all_conv and res.converged
+    comm = get_comm()
+    if comm.active:
+        all_conv = bool(comm.allreduce_scalar(float(all_conv)) == comm.size)
+    return all_conv
+
+
+class DFTGroundState:
+    def __init__(self, kset: KPointSet):
+        self.kset = kset
+        self.ctx = kset.ctx
+        self.density = Density(self.ctx)
+        self.potential = Potential(self.ctx)
+        self.scf_correction_energy = 0.0
+
+    # -- energies (energy.cpp) --------------------------------------------
+
+    def energy_potential(self, rho_r: torch.Tensor) -> float:
+        """∫ρ V_eff (+ E_bxc + PAW/hubbard terms when present)."""
+        return self.ctx.integrate_rg_fine(rho_r * self.potential.veff_r)
+
+    def total_energy_components(self) -> dict:
+        d = {}
+        d["valence_eval_sum"] = self.kset.valence_eval_sum()
+        d["vxc"] = self.potential.energy_vxc(self.density)
+        d["bxc"] = 0.0
+        d["vha"] = self.potential.energy_vha
+        d["exc"] = self.potential.energy_exc(self.density)
+        d["vloc"] = self.potential.energy_vloc(self.density)
+        d["ewald"] = self.potential.ewald
+        d["entropy"] = self.kset.entropy_sum()
+        d["PAW_total_energy"] = 0.0
+        d["PAW_one_elec"] = 0.0
+        d["scf_correction"] = self.scf_correction_energy
+        d["fermi"] = self.kset.energy_fermi
+        return d
+
+    def total_energy(self) -> float:
+        """KS total energy, PP branch (energy.cpp:152-157) + scf correction."""
+        d = self.total_energy_components()
+        return (d["valence_eval_sum"] - d["vxc"] - d["bxc"] - d["PAW_one_elec"]
+                - 0.5 * d["vha"] + d["exc"] + d["PAW_total_energy"] + d["ewald"]
+                + d["scf_correction"])
+
+    # -- driver ------------------------------------------------------------
+
+    def initial_state(self):
+        self.density.initial_density()
+        self.potential.generate(self.density)
+        h0 = Hamiltonian0(self.ctx, self.potential)
+        for kp in self.kset:
+            initialize_subspace(self.ctx, kp, h0(kp))
+        self.kset.find_band_occupancies()
+        return self
+
+    def find(self, density_tol=None, energy_tol=None, itsol_tol=None,
+             num_dft_iter=None, callback=None) -> dict:
+        ctx = self.ctx
+        p = ctx.cfg.parameters
+        itso = ctx.cfg.iterative_solver
+        density_tol = density_tol if density_tol is not None else p.density_tol
+        energy_tol = energy_tol if energy_tol is not None else p.energy_tol
+        itsol_tol = itsol_tol if itsol_tol is not None else itso.energy_tolerance
+        num_dft_iter = num_dft_iter if num_dft_iter is not None else p.num_dft_iter
+
+        self.density.mixer_init(ctx.cfg.mixer)
+        eold = 0.0
+        etot_hist, rms_hist = [], []
+        num_iter = -1
+        t0 = time.time()
+
+        for it in range(num_dft_iter):
+            h0 = Hamiltonian0(ctx, self.potential, self.density)
+            bands_converged = diagonalize(ctx, h0, self.kset, itsol_tol)
+            self.kset.find_band_occupancies()
+            self.density.generate(self.kset, h0)
+
+            e1 = self.energy_potential(self.density.rho_r)
+            rho1_r = self.density.rho_r.clone()
+
+            rms = self.density.mix()
+
+            tol = rms
+            tol = min(itso.tolerance_scale[0] * tol,
+                      itso.tolerance_scale[1] * itsol_tol)
+            itsol_tol = max(itso.min_tolerance, tol)
+            itsol_converged = tol <= itso.min_tolerance
+
+            self.potential.generate(self.density)
+
+            e2 = self.ctx.integrate_rg_fine(rho1_r * self.potential.veff_r)
+            self.scf_correction_energy = e2 - e1
+
+            etot = self.total_energy()
+            etot_hist.append(etot)
+            rms_hist.append(rms)
+            if callback:
+                callback(it, etot, rms)
+
+            conv = (abs(eold - etot) < energy_tol) and bands_converged \
+                and itsol_converged and (rms < density_tol)
+            if conv:
+                num_iter = it
+                break
+            eold = etot
+
+        out = {
+            "converged": num_iter >= 0,
+            "num_scf_iterations": num_iter,
+            "energy": self.total_energy_components() | {
+                "total": self.total_energy(),
+                "free": self.total_energy() + self.kset.entropy_sum(),
+            },
+            "etot_history": etot_hist,
+            "rms_history": rms_hist,
+            "scf_time": time.time() - t0,
+            "efermi": self.kset.energy_fermi,
+        }
+        return out

@@ -1,0 +1,179 @@
+"""Density mixers.
+
+Reference behavior: src/mixer/mixer.hpp:275-420 (variadic Mixer base:
+set_input → residual = input − last_output → rms → mix_impl),
+anderson_mixer.hpp:54 (Anderson with history Gram least squares and beta
+rescaling), broyden2_mixer.hpp:88, linear.
+
+Here a mixed "function" is a dict of named torch tensors (complex or
+real); each name carries an inner-product callable and all components sum
+into the mixing Gram matrix exactly as the reference's multi-function
+mixer does. RMS = sqrt(Σ_components inner(res,res)/global_size), matching
+mixer.hpp:update_rms with normalize=true.
+"""
+
+from __future__ import annotations
+
+import torch
+
+
+def dot_re(x: torch.Tensor, y: torch.Tensor) -> float:
+    """Re <x|y> over all elements."""
+    if torch.is_complex(x):
+        return float(torch.vdot(x.reshape(-1), y.reshape(-1)).real)
+    return float(torch.dot(x.reshape(-1), y.reshape(-1)))
+
+
+class Component:
+    """One mixed quantity: inner product weight and normalization."""
+
+    def __init__(self, name: str, inner=None, global_size: float = 1.0):
+        self.name = name
+        self.inner = inner or dot_re
+        self.global_size = global_size
+
+
+class Mixer:
+    def __init__(self, components: list[Component], max_history: int = 8,
+                 beta: float = 0.7, beta0: float = 0.15,
+                 beta_scaling_factor: float = 1.0):
+        self.components = {c.name: c for c in components}
+        self.max_history = max_history
+        self.beta = beta
+        self.beta0 = beta0
+        self.beta_scaling_factor = beta_scaling_factor
+        self.step = 0
+        self.output_history: list[dict | None] = [None] * max_history
+        self.residual_history: list[dict | None] = [None] * max_history
+        self.rmse_history = [0.0] * max_history
+        self.input: dict | None = None
+
+    # -- helpers over dict-of-tensors ------------------------------------
+
+    def _copy(self, x: dict) -> dict:
+        return {k: v.clone() for k, v in x.items()}
+
+    def _axpy(self, a: float, x: dict, y: dict):
+        for k in y:
+            y[k] += a * x[k]
+
+    def _scale(self, a: float, x: dict):
+        for k in x:
+            x[k] *= a
+
+    def _inner(self, x: dict, y: dict, normalize: bool) -> float:
+        s = 0.0
+        for k in x:
+            c = self.components[k]
+            v = c.inner(x[k], y[k])
+            if normalize:
+                v /= c.global_size
+            s += v
+        return s
+
+    def _idx(self, step: int) -> int:
+        return step % self.max_history
+
+    # -- public ----------------------------------------------------------
+
+    def initialize(self, init_value: dict):
+        self.output_history[0] = self._copy(init_value)
+        self.input = self._copy(init_value)
+
+    def set_input(self, value: dict):
+        self.input = self._copy(value)
+
+    def get_output(self) -> dict:
+        return self._copy(self.output_history[self._idx(self.step)])
+
+    def mix(self, rms_min: float = 1e-16) -> float:
+        idx = self._idx(self.step)
+        res = self._copy(self.input)
+        self._axpy(-1.0, self.output_history[idx], res)
+        self.residual_history[idx] = res
+        rmse = self._inner(res, res, normalize=True) ** 0.5
+        self.rmse_history[idx] = rmse
+        if rmse < rms_min:
+            return rmse
+        self.mix_impl()
+        self.step += 1
+        return rmse
+
+
+class Linear(Mixer):
+    def mix_impl(self):
+        idx = self._idx(self.step)
+        nxt = self._copy(self.output_history[idx])
+        self._axpy(self.beta, self.residual_history[idx], nxt)
+        self.output_history[self._idx(self.step + 1)] = nxt
+
+
+class Anderson(Mixer):
+    """Anderson mixing (reference: anderson_mixer.hpp:77-168)."""
+
+    def __init__(self, *args, **kwargs):
+        super().__init__(*args, **kwargs)
+        self.history_size = 0
+        n = self.max_history - 1
+        self.S = torch.zeros(n, n, dtype=torch.float64)
+
+    def mix_impl(self):
+        idx = self._idx(self.step)
+        idx_prev = self._idx(self.step - 1)
+        hs = self.history_size
+
+        # adaptive beta rescaling
+        if self.step > self.max_history:
+            rmse_avg = sum(self.rmse_history) / len(self.rmse_history)
+            if self.rmse_history[idx] > rmse_avg:
+                self.beta = max(self.beta0, self.beta * self.beta_scaling_factor)
+
+        nxt = self._copy(self.output_history[idx])
+        self._axpy(self.beta, self.residual_history[idx], nxt)
+
+        if hs > 0:
+            # residual[prev] <- residual[step] - residual[prev]; same for output
+            self._scale(-1.0, self.residual_history[idx_prev])
+            self._axpy(1.0, self.residual_history[idx], self.residual_history[idx_prev])
+            self._scale(-1.0, self.output_history[idx_prev])
+            self._axpy(1.0, self.output_history[idx], self.output_history[idx_prev])
+
+            for i in range(hs):
+                j = self._idx(self.step - i - 1)
+                v = self._inner(self.residual_history[j], self.residual_history[idx_prev],
+                                normalize=False)
+                self.S[hs - 1, hs - i - 1] = v
+                self.S[hs - i - 1, hs - 1] = v
+
+            h = torch.zeros(hs, dtype=torch.float64)
+            for i in range(1, hs + 1):
+                j = self._idx(self.step - i)
+                h[hs - i] = self._inner(self.residual_history[j], self.residual_history[idx],
+                                        normalize=False)
+            try:
+                hsol = torch.linalg.solve(self.S[:hs, :hs], h)
+                for i in range(1, hs + 1):
+                    j = self._idx(self.step - i)
+                    self._axpy(-self.beta * float(hsol[hs - i]), self.residual_history[j], nxt)
+                    self._axpy(-float(hsol[hs - i]), self.output_history[j], nxt)
+            except Exception:
+                self.history_size = 0
+                hs = 0
+
+        if self.history_size == self.max_history - 1:
+            self.S[: hs - 1, : hs - 1] = self.S[1:hs, 1:hs].clone()
+
+        self.output_history[self._idx(self.step + 1)] = nxt
+        self.history_size = min(self.history_size + 1, self.max_history - 1)
+
+
+def make_mixer(cfg_mixer, components: list[Component]) -> Mixer:
+    kind = cfg_mixer.type
+    kw = dict(max_history=cfg_mixer.max_history, beta=cfg_mixer.beta,
+              beta0=cfg_mixer.beta0, beta_scaling_factor=cfg_mixer.beta_scaling_factor)
+    if kind == "linear":
+        return Linear(components, **kw)
+    if kind in ("anderson", "anderson_stable", "broyden2"):
+        # broyden2/anderson_stable fall back to anderson until implemented
+        return Anderson(components, **kw)
+    raise ValueError(f"unknown mixer type {kind}")
