@@ -5,10 +5,14 @@ fft.hpp:17) and Smooth_periodic_function::fft_transform
 (src/function3d/smooth_periodic_function.hpp:262).
 
 MI355X-native path: batched dense 3D FFTs through torch.fft (rocFFT) with
-sphere pack/unpack as indexed scatter/gather. The pack/unpack (and the
-fusions around them: kinetic-energy add, V_eff multiply) are the custom
-HIP kernel targets (sirius_amd/ops); this module provides the reference
-torch implementation used on CPU and as the numerics baseline.
+sphere pack/unpack as hand-written HIP kernels (sirius_amd/ops/src/
+scf_ops.hip) — including the fused local-operator pipeline
+pack → IFFT → ×V_eff(r) → FFT → unpack+kinetic, which is the Davidson
+hot loop (reference: Local_operator::apply_h, local_operator.cpp:275,
+GPU twins mul_by_veff_* / add_to_hphi_pw in local_operator.cu).
+
+On a GPU the HIP kernels are mandatory (loud failure if the extension is
+missing); the CPU path uses the equivalent torch indexing ops.
 
 Conventions: f(r) = Σ_G c_G e^{iG·r};  c_G = (1/N) Σ_r f(r) e^{-iG·r}.
 """
@@ -18,6 +22,15 @@ from __future__ import annotations
 import torch
 
 from .gvec import Gvec
+
+
+def _ext_for(t: torch.Tensor):
+    """Native extension when running on GPU (required there), else None."""
+    if t.is_cuda:
+        from .. import ops
+
+        return ops.get_ext(required=True)
+    return None
 
 
 class SphericalFFT:
@@ -30,15 +43,74 @@ class SphericalFFT:
         self.dims = gvec.dims
         self.size = gvec.dims[0] * gvec.dims[1] * gvec.dims[2]
 
+    # -- pack/unpack -------------------------------------------------------
+
+    def _pack(self, coeffs: torch.Tensor) -> torch.Tensor:
+        """[.., nG] -> dense flat [.., size] (zero-filled outside sphere)."""
+        batch = coeffs.shape[:-1]
+        grid = torch.zeros(*batch, self.size, dtype=coeffs.dtype,
+                           device=coeffs.device)
+        ext = _ext_for(coeffs)
+        if ext is not None:
+            ext.pack_sphere(coeffs.contiguous(), self.gvec.fft_index, grid)
+        else:
+            grid[..., self.gvec.fft_index] = coeffs
+        return grid
+
+    def _unpack(self, grid_flat: torch.Tensor) -> torch.Tensor:
+        ext = _ext_for(grid_flat)
+        if ext is not None:
+            out = torch.empty(*grid_flat.shape[:-1], self.gvec.num_gvec,
+                              dtype=grid_flat.dtype, device=grid_flat.device)
+            ext.unpack_sphere(grid_flat.contiguous(), self.gvec.fft_index, out)
+            return out
+        return grid_flat[..., self.gvec.fft_index]
+
+    # -- public ------------------------------------------------------------
+
     def to_real(self, coeffs: torch.Tensor) -> torch.Tensor:
         """[.., nG] complex -> [.., n1, n2, n3] complex."""
-        batch = coeffs.shape[:-1]
-        grid = torch.zeros(*batch, self.size, dtype=coeffs.dtype, device=coeffs.device)
-        grid[..., self.gvec.fft_index] = coeffs
-        grid = grid.reshape(*batch, *self.dims)
+        grid = self._pack(coeffs).reshape(*coeffs.shape[:-1], *self.dims)
         return torch.fft.ifftn(grid, dim=(-3, -2, -1), norm="forward")
 
     def to_pw(self, fr: torch.Tensor) -> torch.Tensor:
         """[.., n1, n2, n3] complex -> [.., nG] complex."""
         g = torch.fft.fftn(fr, dim=(-3, -2, -1), norm="forward")
-        return g.reshape(*fr.shape[:-3], self.size)[..., self.gvec.fft_index]
+        return self._unpack(g.reshape(*fr.shape[:-3], self.size))
+
+    def apply_veff_kinetic(self, psi: torch.Tensor, veff_r: torch.Tensor,
+                           gk2: torch.Tensor) -> torch.Tensor:
+        """Fused local-operator apply: FFT⁻¹ψ → ×V(r) → FFT → +½|G+k|²ψ.
+
+        psi [nb, nG], veff_r [n1,n2,n3] real, gk2 [nG] real.
+        Returns hpsi_local [nb, nG].
+        """
+        nb = psi.shape[0]
+        ext = _ext_for(psi)
+        grid = self._pack(psi).reshape(nb, *self.dims)
+        psi_r = torch.fft.ifftn(grid, dim=(-3, -2, -1), norm="forward")
+        if ext is not None:
+            ext.mul_veff(psi_r, veff_r.contiguous())
+            vpsi = psi_r
+        else:
+            vpsi = psi_r * veff_r
+        g = torch.fft.fftn(vpsi, dim=(-3, -2, -1), norm="forward")
+        g = g.reshape(nb, self.size)
+        if ext is not None:
+            out = torch.empty_like(psi)
+            ext.unpack_add_kinetic(g.contiguous(), self.gvec.fft_index,
+                                   gk2.contiguous(), psi.contiguous(), out)
+            return out
+        return g[..., self.gvec.fft_index] + gk2 * psi
+
+    def density_accumulate(self, psi: torch.Tensor, weights: torch.Tensor,
+                           rho_r: torch.Tensor):
+        """rho_r += Σ_b w_b |FFT⁻¹ψ_b(r)|² (in place; rho_r real [n1,n2,n3])."""
+        psi_r = self.to_real(psi)
+        ext = _ext_for(psi)
+        if ext is not None:
+            ext.density_acc(psi_r.contiguous(), weights.contiguous(),
+                            rho_r.reshape(-1))
+        else:
+            rho_r += torch.einsum("b,bxyz->xyz", weights,
+                                  psi_r.real**2 + psi_r.imag**2)

@@ -109,19 +109,29 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
         e = evals[idx].real
         hpsi = Zs.T @ hphi                                  # [n, nG]
         spsi = Zs.T @ (sphi if sphi is not None else phi)
-        res = hpsi - e[:, None].to(hpsi.dtype) * spsi
-        # drop residuals that are already tiny in norm
-        rn = torch.linalg.vector_norm(res, dim=1).real
+        if hpsi.is_cuda:
+            from . import ops
+
+            ext = ops.get_ext(required=True)
+            res = torch.empty_like(hpsi)
+            norms2 = ext.residual_precond(hpsi.contiguous(), spsi.contiguous(),
+                                          e.contiguous(), h_diag.contiguous(),
+                                          o_diag.contiguous(), res)
+            rn = torch.sqrt(norms2)
+        else:
+            res = hpsi - e[:, None].to(hpsi.dtype) * spsi
+            rn = torch.linalg.vector_norm(res, dim=1).real
+            # precondition (residuals_aux.cu apply_preconditioner)
+            t = h_diag[None, :] - e[:, None] * o_diag[None, :]
+            p = 0.5 * (1.0 + t + torch.sqrt(1.0 + (t - 1.0) ** 2))
+            res = res / p.to(res.dtype)
+        # drop residuals that are already tiny in (unpreconditioned) norm
         keep = rn > 1e-12
         if not bool(keep.any()):
             converged = True
             break
         res = res[keep]
         e = e[keep]
-        # precondition (residuals_aux.cu apply_preconditioner)
-        t = h_diag[None, :] - e[:, None] * o_diag[None, :]
-        p = 0.5 * (1.0 + t + torch.sqrt(1.0 + (t - 1.0) ** 2))
-        res = res / p.to(res.dtype)
         res = res / torch.linalg.vector_norm(res, dim=1, keepdim=True).to(res.dtype)
 
         n_new = res.shape[0]

@@ -94,10 +94,8 @@ class Density:
                 sel = torch.nonzero(occ.abs() > min_occ).reshape(-1)
                 if len(sel) == 0:
                     continue
-                psi_r = kp.fft.to_real(kp.psi[ispn][sel])  # [nocc, n1,n2,n3]
-                w = (kp.weight / ctx.unit_cell.omega) * occ[sel]
-                rho_c += torch.einsum("b,bxyz->xyz", w.to(ctx.rdtype),
-                                      psi_r.real**2 + psi_r.imag**2)
+                w = ((kp.weight / ctx.unit_cell.omega) * occ[sel]).to(ctx.rdtype)
+                kp.fft.density_accumulate(kp.psi[ispn][sel], w, rho_c)
 
         if comm.active:
             comm.allreduce_(rho_c)

@@ -168,11 +168,8 @@ class HamiltonianK:
     def apply_h_s(self, psi: torch.Tensor, ispn: int = 0):
         """psi [nb, nGk] -> (hpsi, spsi). spsi is None for NC (S = I)."""
         kp = self.kp
-        # local: FFT → multiply V_eff(r) → FFT back; fused kinetic add
-        psi_r = kp.fft.to_real(psi)
-        vpsi = psi_r * self.h0.veff_r_coarse  # broadcast over bands
-        hpsi = kp.fft.to_pw(vpsi)
-        hpsi += self.ekin * psi
+        # fused local operator: FFT⁻¹ → ×V_eff(r) → FFT → +½|G+k|²ψ
+        hpsi = kp.fft.apply_veff_kinetic(psi, self.h0.veff_r_coarse, self.ekin)
 
         spsi = None
         if self.bp.num_beta_total:

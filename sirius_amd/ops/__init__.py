@@ -1,0 +1,71 @@
+"""HIP/CDNA4 kernel extension loader.
+
+Builds in-tree (sirius_amd/ops/_build) so the compiled .so travels with the
+repo snapshot to GPU boxes. On a GPU machine the native kernels are
+mandatory: a failed build/import raises instead of silently falling back
+to eager torch (the CPU path keeps the torch reference implementations).
+"""
+
+from __future__ import annotations
+
+import os
+
+import torch
+
+_ext = None
+_tried = False
+
+_SRC_DIR = os.path.join(os.path.dirname(os.path.abspath(__file__)), "src")
+_BUILD_DIR = os.path.join(os.path.dirname(os.path.abspath(__file__)), "_build")
+
+
+def build_extensions(verbose: bool = False):
+    """Compile the HIP extension for gfx950 (cross-compiles fine on CPU-only
+    boxes; hipcc needs no GPU present)."""
+    global _ext, _tried
+    _tried = True
+    os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+    os.makedirs(_BUILD_DIR, exist_ok=True)
+    from torch.utils.cpp_extension import load
+
+    _ext = load(
+        name="sirius_amd_ops",
+        sources=[os.path.join(_SRC_DIR, "scf_ops.hip")],
+        extra_cflags=["-O3"],
+        extra_cuda_cflags=["-O3"],
+        build_directory=_BUILD_DIR,
+        verbose=verbose,
+    )
+    return _ext
+
+
+def get_ext(required: bool | None = None):
+    """Return the extension module; None when unavailable on CPU.
+
+    required defaults to torch.cuda.is_available(): on a GPU box the HIP
+    kernels must load — failure raises loudly.
+    """
+    global _ext, _tried
+    if _ext is not None:
+        return _ext
+    if required is None:
+        required = torch.cuda.is_available()
+    if not _tried:
+        try:
+            build_extensions()
+        except Exception as e:
+            _ext = None
+            if required:
+                raise RuntimeError(
+                    f"sirius_amd HIP extension failed to build/load on a GPU "
+                    f"machine — native kernels are mandatory there: {e}") from e
+    if _ext is None and required:
+        raise RuntimeError("sirius_amd HIP extension unavailable on GPU machine")
+    return _ext
+
+
+def available() -> bool:
+    try:
+        return get_ext(required=False) is not None
+    except Exception:
+        return False

@@ -1,0 +1,145 @@
+"""Unit tests for core machinery: G-vectors, FFT round trip, spherical
+harmonics, radial integrals, smearing, mixer.
+
+Mirrors the reference unit-test tier (apps/unit_tests: test_fft_correctness_*,
+test_gvec, test_ylm/test_rlm, test_spline_*).
+"""
+
+import math
+
+import numpy as np
+import pytest
+import torch
+
+from sirius_amd.core.gvec import Gvec, fft_grid_dims, next_fft_size
+from sirius_amd.core.fft import SphericalFFT
+from sirius_amd.core import ylm as ylm_mod
+from sirius_amd.core.radial import RadialIntegrals, sbessel, spline_integrate
+from sirius_amd import smearing as sm
+
+
+def test_next_fft_size():
+    assert next_fft_size(17) == 18
+    assert next_fft_size(16) == 16
+    assert next_fft_size(11) == 12
+
+
+def _cubic(a=10.0):
+    lat = np.eye(3) * a
+    recip = 2 * math.pi * np.linalg.inv(lat).T
+    return lat, recip
+
+
+def test_gvec_sphere_count():
+    lat, recip = _cubic(10.0)
+    cutoff = 5.0
+    g = Gvec(recip, cutoff, dims=fft_grid_dims(lat, cutoff))
+    # expected count ~ volume of sphere / volume per G point
+    vol_per_g = (2 * math.pi / 10.0) ** 3
+    expect = 4.0 / 3.0 * math.pi * cutoff**3 / vol_per_g
+    assert abs(g.num_gvec - expect) / expect < 0.05
+    assert g.index_of_zero() >= 0
+    # all |G| <= cutoff
+    assert g.gk_len.max() <= cutoff + 1e-12
+
+
+def test_fft_roundtrip():
+    """Sphere -> real grid -> sphere is identity (test_fft_correctness_1)."""
+    lat, recip = _cubic(8.0)
+    g = Gvec(recip, 6.0, dims=fft_grid_dims(lat, 6.0))
+    f = SphericalFFT(g)
+    torch.manual_seed(0)
+    c = torch.randn(4, g.num_gvec, dtype=torch.complex128)
+    fr = f.to_real(c)
+    c2 = f.to_pw(fr)
+    assert torch.allclose(c, c2, atol=1e-12)
+
+
+def test_fft_plane_wave_value():
+    """A single G coefficient produces e^{iG·r} on the grid."""
+    lat, recip = _cubic(7.0)
+    g = Gvec(recip, 4.0, dims=fft_grid_dims(lat, 4.0))
+    f = SphericalFFT(g)
+    # pick some G
+    ig = 5
+    c = torch.zeros(g.num_gvec, dtype=torch.complex128)
+    c[ig] = 1.0
+    fr = f.to_real(c)
+    # value at grid point (1,2,3)
+    n1, n2, n3 = g.dims
+    r = np.array([1 / n1, 2 / n2, 3 / n3]) @ lat
+    expect = np.exp(1j * np.dot(g.g_cart[ig], r))
+    got = complex(fr[1, 2, 3])
+    assert abs(got - expect) < 1e-12
+
+
+def test_rlm_orthonormal():
+    """Real spherical harmonics are orthonormal on the sphere (test_rlm)."""
+    rng = np.random.default_rng(1)
+    n = 200000
+    u = rng.normal(size=(n, 3))
+    _, theta, phi = ylm_mod.spherical_coords(u)
+    lmax = 3
+    R = ylm_mod.rlm(lmax, theta, phi)
+    gram = R.T @ R / n * 4 * math.pi
+    assert np.allclose(gram, np.eye(ylm_mod.lmmax(lmax)), atol=0.05)
+
+
+def test_ylm_vs_rlm_l1():
+    theta = np.array([0.3, 1.2])
+    phi = np.array([0.5, 2.0])
+    R = ylm_mod.rlm(1, theta, phi)
+    # R_10 = sqrt(3/4pi) cos(theta)
+    assert np.allclose(R[:, ylm_mod.lm_index(1, 0)],
+                       math.sqrt(3 / (4 * math.pi)) * np.cos(theta))
+
+
+def test_sbessel_transform_gaussian():
+    """FT of a Gaussian density is analytic: ∫ e^{-r²/2σ²} j0(qr) r² dr."""
+    sigma = 0.7
+    r = np.linspace(1e-6, 12.0, 3000)
+    fr = np.exp(-(r**2) / (2 * sigma**2))
+    q = np.array([0.0, 0.5, 1.5, 3.0])
+    got = RadialIntegrals.sbessel_transform(0, r, fr, q, rpow=2)
+    expect = math.sqrt(math.pi / 2) * sigma**3 * np.exp(-(q**2) * sigma**2 / 2)
+    assert np.allclose(got, expect, rtol=1e-6)
+
+
+def test_vloc_q_coulomb_tail():
+    """For pure -Z/r potential the form factor must equal -4πZ/q²·(Ω/4π)… i.e.
+    vloc_q returns -Z/q² exactly (analytic FT)."""
+    r = np.geomspace(1e-7, 40.0, 6000)
+    z = 4.0
+    v = -z / r
+    q = np.array([0.8, 2.0, 5.0])
+    got = RadialIntegrals.vloc_q(r, v, z, q, r_cut=100.0)
+    assert np.allclose(got, -z / q**2, rtol=1e-5)
+
+
+def test_fermi_search():
+    eig = np.array([[0.0, 0.1, 0.2, 1.0]])
+    w = np.array([1.0])
+    mu = sm.find_fermi(eig, w, 4.0, "gaussian", 0.01, 2.0)
+    assert 0.1 < mu < 0.2  # 4 electrons fill 2 of 4 doubly-occupied states
+    occ = sm.occupancy("gaussian", mu - eig, 0.01) * 2.0
+    assert abs(occ.sum() - 4.0) < 1e-10
+
+
+def test_mixer_linear_convergence():
+    """Linear mixing of a contraction map converges."""
+    from sirius_amd.mixer import Linear, Anderson, Component
+
+    target = torch.tensor([1.0, 2.0, 3.0], dtype=torch.float64)
+
+    def step(x):
+        return target + 0.5 * (x - target)  # fixed point = target
+
+    for cls in (Linear, Anderson):
+        mx = cls([Component("x")], max_history=4, beta=0.7)
+        x = torch.zeros(3, dtype=torch.float64)
+        mx.initialize({"x": x})
+        for _ in range(60):
+            out = mx.get_output()["x"]
+            mx.set_input({"x": step(out)})
+            rms = mx.mix()
+        assert torch.allclose(mx.get_output()["x"], target, atol=1e-6), cls

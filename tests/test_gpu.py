@@ -1,0 +1,113 @@
+"""GPU tests: HIP kernel numerics vs plain torch fp64 reference, and the
+end-to-end SCF on device.
+
+Every test requires an MI355X (marker `gpu`); numerics compare the native
+kernels against the same math done with torch indexing ops on the GPU
+(fp64 reference, per repo test policy).
+"""
+
+import math
+
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+requires_gpu = pytest.mark.skipif(not torch.cuda.is_available(),
+                                  reason="no GPU")
+
+
+@requires_gpu
+class TestKernels:
+    def setup_method(self, _):
+        from sirius_amd import ops
+
+        self.ext = ops.get_ext(required=True)
+        torch.manual_seed(0)
+        self.ng = 7919
+        self.nb = 13
+        self.gs = 24 * 24 * 24
+        dev = "cuda:0"
+        perm = torch.randperm(self.gs, device=dev)[: self.ng]
+        self.idx = perm.to(torch.long)
+        self.coeff = torch.randn(self.nb, self.ng, 2, dtype=torch.float64,
+                                 device=dev)
+        self.coeff = torch.view_as_complex(self.coeff.contiguous())
+
+    def test_pack_unpack(self):
+        grid = torch.zeros(self.nb, self.gs, dtype=torch.complex128,
+                           device="cuda:0")
+        self.ext.pack_sphere(self.coeff, self.idx, grid)
+        ref = torch.zeros_like(grid)
+        ref[:, self.idx] = self.coeff
+        assert torch.equal(grid, ref)
+        out = torch.empty_like(self.coeff)
+        self.ext.unpack_sphere(grid, self.idx, out)
+        assert torch.equal(out, self.coeff)
+
+    def test_unpack_add_kinetic(self):
+        grid = torch.randn(self.nb, self.gs, 2, dtype=torch.float64,
+                           device="cuda:0")
+        grid = torch.view_as_complex(grid.contiguous())
+        ekin = torch.rand(self.ng, dtype=torch.float64, device="cuda:0")
+        out = torch.empty_like(self.coeff)
+        self.ext.unpack_add_kinetic(grid, self.idx, ekin, self.coeff, out)
+        ref = grid[:, self.idx] + ekin * self.coeff
+        assert torch.allclose(out, ref, atol=1e-14)
+
+    def test_mul_veff(self):
+        g = self.coeff.clone().reshape(self.nb, self.ng)
+        v = torch.rand(self.ng, dtype=torch.float64, device="cuda:0")
+        ref = g * v
+        self.ext.mul_veff(g, v)
+        assert torch.allclose(g, ref, atol=1e-15)
+
+    def test_density_acc(self):
+        rho = torch.rand(self.ng, dtype=torch.float64, device="cuda:0")
+        rho0 = rho.clone()
+        w = torch.rand(self.nb, dtype=torch.float64, device="cuda:0")
+        self.ext.density_acc(self.coeff, w, rho)
+        ref = rho0 + torch.einsum("b,bg->g", w, self.coeff.real**2 + self.coeff.imag**2)
+        assert torch.allclose(rho, ref, atol=1e-12)
+
+    def test_residual_precond(self):
+        hpsi = self.coeff.clone()
+        spsi = torch.randn_like(hpsi.real).to(torch.complex128) + self.coeff
+        e = torch.rand(self.nb, dtype=torch.float64, device="cuda:0")
+        hd = torch.rand(self.ng, dtype=torch.float64, device="cuda:0") + 1.0
+        od = torch.ones(self.ng, dtype=torch.float64, device="cuda:0")
+        res = torch.empty_like(hpsi)
+        n2 = self.ext.residual_precond(hpsi, spsi, e, hd, od, res)
+        r_ref = hpsi - e[:, None] * spsi
+        n2_ref = (r_ref.abs() ** 2).sum(dim=1)
+        t = hd[None, :] - e[:, None] * od[None, :]
+        p = 0.5 * (1 + t + torch.sqrt(1 + (t - 1) ** 2))
+        assert torch.allclose(n2, n2_ref, rtol=1e-12)
+        assert torch.allclose(res, r_ref / p, atol=1e-13)
+
+
+@requires_gpu
+def test_scf_gpu_matches_cpu():
+    """Full SCF on device: total energy equals the CPU torch-reference run."""
+    from sirius_amd.models.synthetic import make_context
+    from sirius_amd.kpoint import KPointSet
+    from sirius_amd.dft import DFTGroundState
+
+    res = {}
+    for dev in ("cuda:0", "cpu"):
+        ctx = make_context(natoms=2, gk_cutoff=4.0, pw_cutoff=10.0, device=dev)
+        kset = KPointSet(ctx)
+        dft = DFTGroundState(kset).initial_state()
+        res[dev] = dft.find(num_dft_iter=12)["energy"]["total"]
+    assert math.isfinite(res["cuda:0"])
+    assert abs(res["cuda:0"] - res["cpu"]) < 1e-6
+
+
+@requires_gpu
+def test_native_ops_loaded():
+    """The in-tree HIP extension (not a fallback) is what runs on GPU."""
+    from sirius_amd import ops
+
+    ext = ops.get_ext(required=True)
+    assert "sirius_amd/ops/_build" in ext.__file__, ext.__file__
