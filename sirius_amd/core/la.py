@@ -1,0 +1,56 @@
+"""Dense linear algebra helpers.
+
+Reference behavior: src/core/la (Eigensolver hierarchy eigenproblem.hpp:39,
+chosen by control.std_evp_solver_name; wf::orthogonalize Cholesky path
+wave_functions.hpp:1781).
+
+MI355X policy (measured, profiles/r01_si64_1gpu_kernel_stats.csv):
+rocSOLVER zheevd on subspace-sized matrices (N≈300) is launch-bound —
+~500 kernel dispatches per solve (sytd2/hemv/her2/larfg storms) eating
+~25% of SCF GPU time. For N ≤ EIGH_GPU_MIN we therefore solve on the host
+(LAPACK via torch-CPU, ~12 ms at N=306) and keep the GPU queue free for
+the GEMM/FFT stream; large N stays on rocSOLVER.
+"""
+
+from __future__ import annotations
+
+import torch
+
+EIGH_GPU_MIN = 1024  # below this, host LAPACK beats rocSOLVER's launch storm
+
+
+def eigh(H: torch.Tensor):
+    """Hermitian eigensolve returning (evals, evecs) on H's device."""
+    n = H.shape[-1]
+    if H.is_cuda and n < EIGH_GPU_MIN:
+        w, v = torch.linalg.eigh(H.cpu())
+        return w.to(H.device), v.to(H.device)
+    return torch.linalg.eigh(H)
+
+
+def cholesky(S: torch.Tensor):
+    n = S.shape[-1]
+    if S.is_cuda and n < EIGH_GPU_MIN:
+        return torch.linalg.cholesky(S.cpu()).to(S.device)
+    return torch.linalg.cholesky(S)
+
+
+def inv_lower(L: torch.Tensor) -> torch.Tensor:
+    """L^{-1} for lower-triangular L (small; host when on GPU)."""
+    eye = torch.eye(L.shape[-1], dtype=L.dtype, device=L.device)
+    if L.is_cuda and L.shape[-1] < EIGH_GPU_MIN:
+        out = torch.linalg.solve_triangular(L.cpu(), eye.cpu(), upper=False)
+        return out.to(L.device)
+    return torch.linalg.solve_triangular(L, eye, upper=False)
+
+
+def inner(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    """Gram block ⟨a_i|b_j⟩ = Σ_g conj(a[i,g]) b[j,g] as ONE zgemm.
+
+    a @ b.conj().T maps to a single rocBLAS zgemm with op='C' (the
+    conj-transpose view is BLAS-native); the final conj touches only the
+    small [na, nb] result. The naive a.conj() @ b.T form materializes a
+    full conjugated copy of a (seen as copyBuffer+elementwise traffic in
+    profiles/r01).
+    """
+    return (a @ b.conj().T).conj()

@@ -18,6 +18,8 @@ from dataclasses import dataclass
 import numpy as np
 import torch
 
+from .core import la
+
 
 @dataclass
 class DavidsonResult:
@@ -28,8 +30,8 @@ class DavidsonResult:
 
 
 def _inner(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
-    """⟨a_i|b_j⟩ Gram block: [na, nb]."""
-    return a.conj() @ b.T
+    """⟨a_i|b_j⟩ Gram block: [na, nb] (single zgemm, see core.la.inner)."""
+    return la.inner(a, b)
 
 
 def _ortho_block(new: torch.Tensor, snew, phi, sphi):
@@ -50,14 +52,14 @@ def _ortho_block(new: torch.Tensor, snew, phi, sphi):
     n = gram.shape[0]
     eye = torch.eye(n, dtype=gram.dtype, device=gram.device)
     try:
-        L = torch.linalg.cholesky(gram)
-        linv = torch.linalg.solve_triangular(L, eye, upper=False)
+        L = la.cholesky(gram)
+        linv = la.inv_lower(L)
         new = linv.conj() @ new
         snew = linv.conj() @ snew if snew is not None else None
         return new, snew
     except Exception:
         # rank-deficient: keep the well-conditioned subspace via eigh filter
-        w, v = torch.linalg.eigh(gram)
+        w, v = la.eigh(gram)
         keep = w > 1e-10
         t = (v[:, keep] / torch.sqrt(w[keep])).conj().T     # [nkeep, n]
         new = t @ new
@@ -88,7 +90,7 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
     sphi = sphi_new  # None => S = I
     H = _inner(phi, hphi)
     H = 0.5 * (H + H.conj().T)
-    evals, Z = torch.linalg.eigh(H)
+    evals, Z = la.eigh(H)
     eval_old = np.full(nb, 1e10)
     niter = 0
     converged = False
@@ -172,7 +174,7 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
         Hn[:, N:] = blk
         Hn[N:, :N] = blk[:N].conj().T
         H = 0.5 * (Hn + Hn.conj().T)
-        evals, Z = torch.linalg.eigh(H)
+        evals, Z = la.eigh(H)
 
     psi = Z[:, :nb].T @ phi
     return DavidsonResult(eval=evals[:nb].real.cpu().numpy(), psi=psi,

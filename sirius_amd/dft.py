@@ -17,6 +17,7 @@ import time
 import numpy as np
 import torch
 
+from .core import la
 from .core import ylm as ylm_mod
 from .core.radial import RadialIntegrals
 from .davidson import davidson
@@ -80,19 +81,18 @@ def initialize_subspace(ctx, kp, hk):
     phi, _ = _ortho_block(phi, None, None, None)
     hphi, sphi = hk.apply_h_s(phi)
     if sphi is None:
-        H = phi.conj() @ hphi.T
+        H = la.inner(phi, hphi)
         H = 0.5 * (H + H.conj().T)
-        evals, Z = torch.linalg.eigh(H)
+        evals, Z = la.eigh(H)
     else:
-        H = phi.conj() @ hphi.T
-        S = phi.conj() @ sphi.T
+        H = la.inner(phi, hphi)
+        S = la.inner(phi, sphi)
         H = 0.5 * (H + H.conj().T)
         S = 0.5 * (S + S.conj().T)
-        L = torch.linalg.cholesky(S)
-        Li = torch.linalg.solve_triangular(
-            L, torch.eye(L.shape[0], dtype=L.dtype, device=L.device), upper=False)
+        L = la.cholesky(S)
+        Li = la.inv_lower(L)
         Ht = Li @ H @ Li.conj().T
-        evals, Zt = torch.linalg.eigh(0.5 * (Ht + Ht.conj().T))
+        evals, Zt = la.eigh(0.5 * (Ht + Ht.conj().T))
         Z = Li.conj().T @ Zt
     nsel = min(nb, phi.shape[0])
     psi = Z[:, :nsel].T @ phi
