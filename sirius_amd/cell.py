@@ -88,7 +88,7 @@ class AtomType:
         at.rho_total_4pir2 = np.asarray(pp["total_charge_density"], dtype=np.float64)
         at.spin_orbit = bool(h.get("spin_orbit", False))
         at.core_correction = bool(h.get("core_correction", False))
-        at.is_ultrasoft = bool(h.get("is_ultrasoft", False))
+        at.is_ultrasoft = bool(h.get("is_ultrasoft", False)) or h.get("pseudo_type") in ("US", "USPP")
         ptype = h.get("pseudo_type", "NC")
         at.is_paw = ptype in ("PAW",)
         at.is_norm_conserving = not (at.is_ultrasoft or at.is_paw)
@@ -184,6 +184,11 @@ class AtomType:
     @property
     def num_beta(self) -> int:
         return len(self.beta)
+
+    @property
+    def augment(self) -> bool:
+        """True when the type carries augmentation charges (USPP/PAW)."""
+        return len(self.q_radial) > 0
 
     @property
     def num_beta_lm(self) -> int:

@@ -25,6 +25,82 @@ from .core.gvec import Gvec, fft_grid_dims
 from . import xc as xc_mod
 
 
+class RadialIntegralsCache:
+    """Tabulated radial form factors per atom type (reference:
+    Simulation_context::ri(), radial integral tables built at
+    simulation_context.cpp:1050-1100 with settings.nprii_* resolutions;
+    evaluation is spline interpolation in q — see core.radial.RITable)."""
+
+    def __init__(self, ctx):
+        self.ctx = ctx
+        self._vloc = {}
+        self._beta = {}
+        self._aug = {}
+        self._core = {}
+
+    def vloc(self, lab):
+        from .core.radial import VlocTable
+
+        if lab not in self._vloc:
+            s = self.ctx.cfg.settings
+            self._vloc[lab] = VlocTable(
+                self.ctx.unit_cell.atom_types[lab], self.ctx.pw_cutoff,
+                s.nprii_vloc, r_cut=s.pseudo_grid_cutoff)
+        return self._vloc[lab]
+
+    def beta(self, lab):
+        """RITable over [n_beta_radial, nq], qmax = gk_cutoff."""
+        from .core.radial import RITable, make_q_grid, RadialIntegrals
+
+        if lab not in self._beta:
+            at = self.ctx.unit_cell.atom_types[lab]
+            q = make_q_grid(self.ctx.gk_cutoff, self.ctx.cfg.settings.nprii_beta)
+            vals = np.stack([
+                RadialIntegrals.sbessel_transform(b.l, at.r, b.f_r, q, rpow=1)
+                for b in at.beta]) if at.num_beta else np.zeros((0, len(q)))
+            self._beta[lab] = RITable(q, vals)
+        return self._beta[lab]
+
+    def aug(self, lab):
+        """RITable over [n_rf_pairs, 2*lmax_beta+1, nq], qmax = pw_cutoff."""
+        from .core.radial import RITable, make_q_grid, RadialIntegrals
+
+        if lab not in self._aug:
+            at = self.ctx.unit_cell.atom_types[lab]
+            q = make_q_grid(self.ctx.pw_cutoff, self.ctx.cfg.settings.nprii_aug)
+            qmap = {}
+            for qq in at.q_radial:
+                i, j = min(qq.i, qq.j), max(qq.i, qq.j)
+                qmap[(i, j, qq.l)] = qq.f_r
+            nbrf = at.num_beta
+            lmax3 = 2 * max((b.l for b in at.beta), default=0)
+            vals = np.zeros((nbrf * (nbrf + 1) // 2, lmax3 + 1, len(q)))
+            for j in range(nbrf):
+                lj = at.beta[j].l
+                for i in range(j + 1):
+                    li = at.beta[i].l
+                    pair = j * (j + 1) // 2 + i
+                    for l3 in range(abs(li - lj), min(li + lj, lmax3) + 1):
+                        if (li + lj + l3) % 2 != 0:
+                            continue
+                        f = qmap.get((i, j, l3))
+                        if f is not None:
+                            vals[pair, l3] = RadialIntegrals.sbessel_transform(
+                                l3, at.r, f, q, rpow=0)
+            self._aug[lab] = RITable(q, vals)
+        return self._aug[lab]
+
+    def rho_core(self, lab):
+        from .core.radial import RITable, make_q_grid, RadialIntegrals
+
+        if lab not in self._core:
+            at = self.ctx.unit_cell.atom_types[lab]
+            q = make_q_grid(self.ctx.pw_cutoff, self.ctx.cfg.settings.nprii_rho_core)
+            vals = RadialIntegrals.sbessel_transform(0, at.r, at.rho_core_r, q, rpow=2)
+            self._core[lab] = RITable(q, vals)
+        return self._core[lab]
+
+
 class SimulationContext:
     def __init__(self, cfg: Config, unit_cell: UnitCell | None = None,
                  base_dir: str = ".", device: str | None = None):
@@ -75,11 +151,49 @@ class SimulationContext:
             int(p.num_bands) if p.num_bands > 0 else nbnd)
         self.max_occupancy = 2.0 if self.num_mag_dims == 0 else 1.0
 
+        # crystal symmetry (space group + IBZ; reference: Crystal_symmetry)
+        self.symmetry = None
+        if p.use_symmetry:
+            from .symmetry import CrystalSymmetry
+
+            self.symmetry = CrystalSymmetry(self.unit_cell)
+
         # structure phase factors on the fine sphere: e^{iG·τ_a} per atom
         self._phase_fine = None
+        self._phase_pos = {}
+        self._aug_ops = {}
 
         self.dtype = torch.complex128
         self.rdtype = torch.float64
+
+        self.ri = RadialIntegralsCache(self)
+
+    # -- augmentation (USPP/PAW) ------------------------------------------
+
+    @property
+    def has_aug(self) -> bool:
+        return any(at.augment for at in self.unit_cell.atom_types.values())
+
+    def aug_op(self, label: str):
+        """Cached Augmentation_operator analogue per atom type
+        (reference: Simulation_context::augmentation_op)."""
+        if label not in self._aug_ops:
+            from .augmentation import AugmentationOperator
+
+            self._aug_ops[label] = AugmentationOperator(
+                self, self.unit_cell.atom_types[label])
+        return self._aug_ops[label]
+
+    def phase_pos(self, label: str) -> torch.Tensor:
+        """e^{+iG·τ_a} on the fine sphere for atoms of one type: [na, nG]."""
+        if label not in self._phase_pos:
+            uc = self.unit_cell
+            ia = uc.atoms_of_type(label)
+            tau = uc.atom_positions_frac()[ia]
+            m = self.gvec_fine.miller.astype(np.float64)
+            ph = np.exp(2j * math.pi * (tau @ m.T))
+            self._phase_pos[label] = torch.from_numpy(ph).to(self.device)
+        return self._phase_pos[label]
 
     # -- structure factors -------------------------------------------------
 

@@ -20,10 +20,16 @@ import torch
 
 
 def next_fft_size(n: int) -> int:
-    """Smallest 5-smooth (2,3,5,7-factorable) size >= n (rocFFT-friendly)."""
+    """Smallest {2,3,5}-smooth size >= n.
+
+    Matches the reference's fft::Grid::find_grid_size (fft3d_grid.hpp:34-48)
+    so that real-space integration grids — and therefore grid-quadrature
+    XC energies — agree with the reference to the last digit. (rocFFT also
+    handles 7-smooth sizes, but grid parity wins.)
+    """
     while True:
         m = n
-        for p in (2, 3, 5, 7):
+        for p in (2, 3, 4, 5):
             while m % p == 0:
                 m //= p
         if m == 1:
@@ -32,14 +38,16 @@ def next_fft_size(n: int) -> int:
 
 
 def fft_grid_dims(lattice: np.ndarray, cutoff: float) -> tuple[int, int, int]:
-    """FFT grid large enough to hold the |G| <= cutoff sphere.
+    """FFT grid that circumscribes the |G| <= cutoff sphere.
 
-    |m_i| = |G·a_i|/2π <= cutoff·|a_i|/2π.
+    Mirrors fft::get_min_grid (fft3d_grid.hpp:160) + r3::find_translations
+    (r3.hpp:588-603): N_i = int(2·cutoff·|a_i|/2π) + 1 + 2, rounded up to
+    a {2,3,5}-smooth size.
     """
     dims = []
     for i in range(3):
-        mmax = int(cutoff * np.linalg.norm(lattice[i]) / (2 * math.pi)) + 1
-        dims.append(next_fft_size(2 * mmax + 1))
+        n = int(2 * cutoff * np.linalg.norm(lattice[i]) / (2 * math.pi)) + 3
+        dims.append(next_fft_size(n))
     return tuple(dims)
 
 

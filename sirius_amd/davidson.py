@@ -2,13 +2,17 @@
 
 Reference behavior: src/hamiltonian/davidson.hpp:130-856 (generic block
 Davidson for H|ψ⟩ = ε S|ψ⟩ with locking, restart, adaptive per-band
-energy tolerance), residuals.hpp (r = Hψ − εSψ, Teter-style diagonal
-preconditioner p = ½(1 + t + sqrt(1 + (t−1)²)), t = h_diag − ε o_diag —
+energy tolerance; expansion block gets apply_h_s first, then {phi, hphi,
+sphi} are orthogonalized together — davidson.hpp:771-781), residuals.hpp
+(r = Hψ − εSψ, Teter-style diagonal preconditioner
+p = ½(1 + t + sqrt(1 + (t−1)²)), t = h_diag − ε o_diag —
 residuals_aux.cu:300-315), convergence test |Δε_j| ≤ tol(j)
 (davidson.hpp:333-335, diagonalize.hpp:48-52).
 
-All tensors live on the compute device; subspace eigensolve via
-torch.linalg.eigh (rocSOLVER on MI355X), Gram/transform GEMMs via zgemm.
+All tensors live on the compute device; subspace eigensolves go through
+core.la (host LAPACK below the rocSOLVER-viability size), Gram/transform
+GEMMs are single zgemms; the fused residual+preconditioner+norm HIP
+kernel runs on GPU.
 """
 
 from __future__ import annotations
@@ -34,37 +38,37 @@ def _inner(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
     return la.inner(a, b)
 
 
-def _ortho_block(new: torch.Tensor, snew, phi, sphi):
-    """Project existing subspace out of `new` and S-orthonormalize it.
+def _ortho_joint(new, hnew, snew, phi, hphi, sphi):
+    """Project the existing S-orthonormal subspace out of `new` and
+    S-orthonormalize it, applying identical transforms to hnew/snew
+    (mirrors wf::orthogonalize, wave_functions.hpp:1781-2051).
 
-    Mirrors wf::orthogonalize (wave_functions.hpp:1781-2051): project-out,
-    then Gram + Cholesky + triangular solve. Returns orthonormal block (rows
-    may be fewer than input when rank-deficient).
+    snew is None ⇔ S = I. Returns (new, hnew, snew) with possibly fewer
+    rows when rank-deficient.
     """
+    s_of_new = snew if snew is not None else new
     if phi is not None and phi.shape[0]:
         ov = _inner(sphi if sphi is not None else phi, new)   # [N, n]
         new = new - ov.T @ phi
+        if hnew is not None:
+            hnew = hnew - ov.T @ hphi
         if snew is not None:
             snew = snew - ov.T @ sphi
-    s = snew if snew is not None else new
-    gram = _inner(new, s)
+        s_of_new = snew if snew is not None else new
+    gram = _inner(new, s_of_new)
     gram = 0.5 * (gram + gram.conj().T)
     n = gram.shape[0]
-    eye = torch.eye(n, dtype=gram.dtype, device=gram.device)
     try:
         L = la.cholesky(gram)
-        linv = la.inv_lower(L)
-        new = linv.conj() @ new
-        snew = linv.conj() @ snew if snew is not None else None
-        return new, snew
+        t = la.inv_lower(L).conj()
     except Exception:
-        # rank-deficient: keep the well-conditioned subspace via eigh filter
         w, v = la.eigh(gram)
         keep = w > 1e-10
         t = (v[:, keep] / torch.sqrt(w[keep])).conj().T     # [nkeep, n]
-        new = t @ new
-        snew = t @ snew if snew is not None else None
-        return new, snew
+    new = t @ new
+    hnew = t @ hnew if hnew is not None else None
+    snew = t @ snew if snew is not None else None
+    return new, hnew, snew
 
 
 def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
@@ -73,21 +77,20 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
              num_steps: int = 20, subspace_size: int = 2,
              min_occupancy: float = 1e-14,
              extra_ortho: bool = False) -> DavidsonResult:
-    """Solve for the `nb` lowest eigenpairs of H (S=I or USPP S).
+    """Solve for the `nb` lowest eigenpairs of H ψ = ε S ψ (S=I or USPP S).
 
     apply_h_s(phi [n, nG]) -> (hphi, sphi|None).
     """
     nb, nG = psi0.shape
-    num_phi_max = min(subspace_size * nb, nG // 2) if nG // 2 > nb else nb
-    num_phi_max = max(num_phi_max, nb + 1) if nG > nb + 1 else nb
+    num_phi_max = min(max(subspace_size * nb, nb + 1), max(nG // 2, nb))
 
     tol = np.full(nb, tol_occ)
     if occ is not None:
         tol = np.where(occ > min_occupancy, tol_occ, tol_occ + tol_empty)
 
-    phi, sphi = _ortho_block(psi0.clone(), None, None, None)
-    hphi, sphi_new = apply_h_s(phi)
-    sphi = sphi_new  # None => S = I
+    phi = psi0.clone()
+    hphi, sphi = apply_h_s(phi)
+    phi, hphi, sphi = _ortho_joint(phi, hphi, sphi, None, None, None)
     H = _inner(phi, hphi)
     H = 0.5 * (H + H.conj().T)
     evals, Z = la.eigh(H)
@@ -127,13 +130,12 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
             t = h_diag[None, :] - e[:, None] * o_diag[None, :]
             p = 0.5 * (1.0 + t + torch.sqrt(1.0 + (t - 1.0) ** 2))
             res = res / p.to(res.dtype)
-        # drop residuals that are already tiny in (unpreconditioned) norm
+        # drop residuals already tiny in (unpreconditioned) norm
         keep = rn > 1e-12
         if not bool(keep.any()):
             converged = True
             break
         res = res[keep]
-        e = e[keep]
         res = res / torch.linalg.vector_norm(res, dim=1, keepdim=True).to(res.dtype)
 
         n_new = res.shape[0]
@@ -153,17 +155,15 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
                 if n_new <= 0:
                     break
 
-        new, snew0 = _ortho_block(res, None, phi, sphi)
+        hnew, snew = apply_h_s(res)
+        res, hnew, snew = _ortho_joint(res, hnew, snew, phi, hphi, sphi)
         if extra_ortho:
-            new, snew0 = _ortho_block(new, snew0, phi, sphi)
-        if new.shape[0] == 0:
+            res, hnew, snew = _ortho_joint(res, hnew, snew, phi, hphi, sphi)
+        if res.shape[0] == 0:
             converged = True
             break
-        hnew, snew = apply_h_s(new)
-        if sphi is not None and snew is None:
-            snew = new
         # grow subspace
-        phi = torch.cat([phi, new], dim=0)
+        phi = torch.cat([phi, res], dim=0)
         hphi = torch.cat([hphi, hnew], dim=0)
         if sphi is not None:
             sphi = torch.cat([sphi, snew], dim=0)

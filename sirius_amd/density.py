@@ -45,7 +45,7 @@ class Density:
         ff = {}
         for lab, at in uc.atom_types.items():
             if at.core_correction and at.rho_core_r.any():
-                ff[lab] = RadialIntegrals.sbessel_transform(0, at.r, at.rho_core_r, q, rpow=2)
+                ff[lab] = ctx.ri.rho_core(lab)(q)
             else:
                 ff[lab] = np.zeros_like(q)
         self.rho_core_g = ctx.make_periodic_function(ff)
@@ -105,9 +105,93 @@ class Density:
         self.rho_g = torch.zeros(ctx.gvec_fine.num_gvec, dtype=ctx.dtype,
                                  device=ctx.device)
         self.rho_g[ctx.coarse_to_fine] = rho_cg
-        # augmentation charge (USPP/PAW) added here when implemented
+        # augmentation charge (USPP/PAW)
+        if ctx.has_aug:
+            dm = self.generate_density_matrix(kset)
+            if ctx.symmetry is not None:
+                from .symmetry import symmetrize_density_matrix
+
+                dm = symmetrize_density_matrix(dm, ctx, ctx.symmetry.ops)
+            self.density_matrix = dm
+            self.rho_g = self.rho_g + self.generate_rho_aug(dm)
+        if ctx.symmetry is not None:
+            from .symmetry import symmetrize_rho_g
+
+            self.rho_g = symmetrize_rho_g(self.rho_g, ctx.gvec_fine,
+                                          ctx.symmetry.ops)
         self.rho_r = ctx.fft_fine.to_real(self.rho_g).real
         return self
+
+    def generate_density_matrix(self, kset):
+        """dm_a(ξ1,ξ2,σ) = Σ_{k,n} w_k f_nk ⟨β_ξ1|ψ⟩⟨ψ|β_ξ2⟩ per atom
+        (reference: add_k_point_contribution_dm_pwpp_collinear,
+        density.cpp:845-900)."""
+        ctx = self.ctx
+        uc = ctx.unit_cell
+        from .hamiltonian import BetaProjectors
+
+        dm = {}
+        for lab, at in uc.atom_types.items():
+            na = len(uc.atoms_of_type(lab))
+            nbf = at.num_beta_lm
+            dm[lab] = torch.zeros(na, nbf, nbf, ctx.num_spins,
+                                  dtype=ctx.dtype, device=ctx.device)
+        min_occ = ctx.cfg.iterative_solver.min_occupancy
+        for kp in kset:
+            if kp.beta is None:
+                kp.beta = BetaProjectors(ctx, kp)
+            bp = kp.beta
+            for ispn in range(ctx.num_spins):
+                occ = torch.from_numpy(kp.occ[ispn]).to(ctx.device)
+                sel = torch.nonzero(occ.abs() > min_occ).reshape(-1)
+                if len(sel) == 0:
+                    continue
+                bpsi = bp.inner(kp.psi[ispn][sel])     # [nbf_tot, nocc]
+                wf = (kp.weight * occ[sel]).to(ctx.dtype)
+                for lab, at in uc.atom_types.items():
+                    nbf = at.num_beta_lm
+                    if nbf == 0:
+                        continue
+                    ia_list = uc.atoms_of_type(lab)
+                    offs = [bp.atom_offsets[ia] for ia in ia_list]
+                    rows = torch.tensor(
+                        [o + x for o in offs for x in range(nbf)],
+                        device=ctx.device)
+                    x = bpsi[rows].reshape(len(ia_list), nbf, -1)  # [na,nbf,nocc]
+                    dm[lab][..., ispn] += torch.einsum(
+                        "aib,b,ajb->aij", x, wf, x.conj())
+        comm = get_comm()
+        if comm.active:
+            for lab in dm:
+                comm.allreduce_(dm[lab])
+        return dm
+
+    def generate_rho_aug(self, dm) -> torch.Tensor:
+        """ρ_aug(G) = Σ_a Σ_{ξ1≤ξ2} symw·Q_{ξ1ξ2}(G)·dm_aux(ξ2,ξ1,a)·e^{-iG·τ_a}
+        (reference: generate_rho_aug, density.cpp:1395-1520; GPU twins
+        generate_dm_pw / sum_q_pw_dm_pw kernels)."""
+        ctx = self.ctx
+        uc = ctx.unit_cell
+        out = torch.zeros(ctx.gvec_fine.num_gvec, dtype=ctx.dtype,
+                          device=ctx.device)
+        for lab, at in uc.atom_types.items():
+            if not (at.augment and at.num_beta):
+                continue
+            aug = ctx.aug_op(lab)
+            nbf = aug.nbf
+            xi1_idx, xi2_idx = [], []
+            for xi2 in range(nbf):
+                for xi1 in range(xi2 + 1):
+                    xi1_idx.append(xi1)
+                    xi2_idx.append(xi2)
+            d = dm[lab]                                    # [na, nbf, nbf, nspin]
+            # non-magnetic / collinear total: sum over spins, Re part
+            dm_aux = d[:, xi2_idx, xi1_idx, :].sum(-1).real.T   # [nqlm, na]
+            phases = ctx.phase_pos(lab).conj()             # e^{-iGτ} [na, nG]
+            dm_pw = dm_aux.to(ctx.dtype) @ phases          # [nqlm, nG]
+            out += torch.einsum("q,qg,qg->g", aug.sym_weight.to(ctx.dtype),
+                                aug.q_pw, dm_pw)
+        return out
 
     def check_num_electrons(self) -> float:
         n = self.ctx.integrate_rg_fine(self.rho_r)

@@ -77,23 +77,12 @@ def initialize_subspace(ctx, kp, hk):
         rnd = rnd * damp
         phi = torch.cat([phi, rnd], dim=0)
 
-    from .davidson import _ortho_block
-    phi, _ = _ortho_block(phi, None, None, None)
+    from .davidson import _ortho_joint
     hphi, sphi = hk.apply_h_s(phi)
-    if sphi is None:
-        H = la.inner(phi, hphi)
-        H = 0.5 * (H + H.conj().T)
-        evals, Z = la.eigh(H)
-    else:
-        H = la.inner(phi, hphi)
-        S = la.inner(phi, sphi)
-        H = 0.5 * (H + H.conj().T)
-        S = 0.5 * (S + S.conj().T)
-        L = la.cholesky(S)
-        Li = la.inv_lower(L)
-        Ht = Li @ H @ Li.conj().T
-        evals, Zt = la.eigh(0.5 * (Ht + Ht.conj().T))
-        Z = Li.conj().T @ Zt
+    phi, hphi, sphi = _ortho_joint(phi, hphi, sphi, None, None, None)
+    H = la.inner(phi, hphi)
+    H = 0.5 * (H + H.conj().T)
+    evals, Z = la.eigh(H)
     nsel = min(nb, phi.shape[0])
     psi = Z[:, :nsel].T @ phi
     if nsel < nb:

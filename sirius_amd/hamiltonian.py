@@ -52,8 +52,7 @@ class BetaProjectors:
             if at.num_beta == 0:
                 cols_t[lab] = np.zeros((len(glen), 0), dtype=np.complex128)
                 continue
-            fl = [RadialIntegrals.sbessel_transform(b.l, at.r, b.f_r, glen, rpow=1)
-                  for b in at.beta]  # each [nGk]
+            fl = ctx.ri.beta(lab)(glen)  # [n_beta_radial, nGk]
             cols = []
             for (irf, l, m) in at.beta_lm_index():
                 z = (-1j) ** l * (4 * math.pi / math.sqrt(uc.omega))
@@ -87,17 +86,19 @@ class BetaProjectors:
         return self.beta.conj().T @ psi.T  # psi [nb, nGk] -> [nbf, nb]
 
 
-def block_diag_matrix(ctx, bp: BetaProjectors, per_type_matrix: dict[str, np.ndarray]):
-    """Assemble block-diagonal [nbf_tot, nbf_tot] operator from per-type blocks."""
+def block_diag_matrix(ctx, bp: BetaProjectors, per_atom_matrix):
+    """Assemble block-diagonal [nbf_tot, nbf_tot] operator.
+
+    per_atom_matrix: callable ia -> [nbf, nbf] numpy (or None for zero)."""
     uc = ctx.unit_cell
     n = bp.num_beta_total
     out = torch.zeros(n, n, dtype=ctx.dtype, device=ctx.device)
-    for ia, (lab, _) in enumerate(uc.atoms):
+    for ia in range(uc.num_atoms):
         o = bp.atom_offsets[ia]
         nb = bp.atom_nbf[ia]
-        if nb:
-            out[o:o + nb, o:o + nb] = torch.from_numpy(
-                per_type_matrix[lab]).to(ctx.device)
+        m = per_atom_matrix(ia)
+        if nb and m is not None:
+            out[o:o + nb, o:o + nb] = torch.from_numpy(np.ascontiguousarray(m)).to(ctx.device)
     return out
 
 
@@ -116,7 +117,12 @@ def d_matrix_lm(at) -> np.ndarray:
 
 class Hamiltonian0:
     """Potential-dependent, k-independent parts (reference hamiltonian.hpp:62):
-    V_eff on the coarse real grid + D-operator matrices."""
+    V_eff on the coarse real grid + D-operator matrices.
+
+    D assembly for augmented (USPP) types follows
+    generate_d_operator_matrix.cpp: D_a = Ω·Σ_G conj(Q_ij(G))·V_eff(G)·e^{iG·τ_a}
+    + D^ion δ_{lm,lm'}; for NC types D = D^ion alone
+    (non_local_operator.cpp:208-253)."""
 
     def __init__(self, ctx, potential, density=None):
         self.ctx = ctx
@@ -124,9 +130,31 @@ class Hamiltonian0:
         # map veff from fine sphere to coarse sphere, then to coarse real grid
         veff_coarse_g = potential.veff_g[ctx.coarse_to_fine]
         self.veff_r_coarse = ctx.fft_coarse.to_real(veff_coarse_g).real
-        # D matrices: NC = d_ion; USPP adds ∫ V_eff(r) Q_ij(r - τ) dr
-        self.d_per_type = {lab: d_matrix_lm(at)
-                           for lab, at in ctx.unit_cell.atom_types.items()}
+        uc = ctx.unit_cell
+        # per-atom D matrices [nbf, nbf] (numpy, real)
+        self.d_atom = [None] * uc.num_atoms
+        for lab, at in uc.atom_types.items():
+            ions = d_matrix_lm(at)
+            ia_list = uc.atoms_of_type(lab)
+            if at.augment and at.num_beta:
+                aug = ctx.aug_op(lab)
+                ph = ctx.phase_pos(lab)                    # [na, nG]
+                va = potential.veff_g[None, :] * ph         # [na, nG]
+                dt = (aug.q_pw.conj() @ va.T).real * uc.omega  # [nqlm, na]
+                dt = dt.cpu().numpy()
+                nbf = aug.nbf
+                for i, ia in enumerate(ia_list):
+                    d = np.array(ions)
+                    for xi2 in range(nbf):
+                        for xi1 in range(xi2 + 1):
+                            idx12 = xi2 * (xi2 + 1) // 2 + xi1
+                            d[xi1, xi2] += dt[idx12, i]
+                            if xi1 != xi2:
+                                d[xi2, xi1] += dt[idx12, i]
+                    self.d_atom[ia] = d
+            else:
+                for ia in ia_list:
+                    self.d_atom[ia] = ions
 
     def __call__(self, kp) -> "HamiltonianK":
         return HamiltonianK(self, kp)
@@ -143,8 +171,18 @@ class HamiltonianK:
         if kp.beta is None:
             kp.beta = BetaProjectors(self.ctx, kp)
         self.bp = kp.beta
-        self.D = block_diag_matrix(self.ctx, self.bp, h0.d_per_type)
+        ctx = self.ctx
+        self.D = block_diag_matrix(ctx, self.bp, lambda ia: h0.d_atom[ia])
         self.Q = None  # USPP overlap augmentation (else S = I)
+        if ctx.has_aug:
+            uc = ctx.unit_cell
+
+            def qmat(ia):
+                lab = uc.atoms[ia][0]
+                at = uc.atom_types[lab]
+                return ctx.aug_op(lab).q_mtrx if at.augment else None
+
+            self.Q = block_diag_matrix(ctx, self.bp, qmat)
         self.ekin = 0.5 * kp.gkvec.gk2_t  # [nGk] float64
 
     def h_diag(self, ispn: int = 0) -> torch.Tensor:

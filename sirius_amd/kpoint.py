@@ -98,9 +98,11 @@ class KPointSet:
         self.comm = get_comm()
         p = ctx.cfg.parameters
         if vk is None:
-            if ctx.cfg.parameters.use_symmetry and getattr(ctx, "symmetry", None):
-                vk, weights = kmesh_ibz(ctx.unit_cell, p.ngridk, p.shiftk,
-                                        ctx.symmetry.rotations)
+            if p.use_symmetry and p.use_ibz and getattr(ctx, "symmetry", None):
+                from .symmetry import ibz_mesh
+
+                vk, weights = ibz_mesh(ctx.unit_cell, p.ngridk, p.shiftk,
+                                       ctx.symmetry.ops)
             else:
                 vk, weights = kmesh_full(p.ngridk, p.shiftk)
         self.vk = np.atleast_2d(vk)

@@ -47,9 +47,7 @@ class Potential:
         ff = {}
         q = ctx.gvec_fine.shell_len
         for lab, at in uc.atom_types.items():
-            ff[lab] = RadialIntegrals.vloc_q(
-                at.r, at.vloc_r, at.zn, q,
-                r_cut=ctx.cfg.settings.pseudo_grid_cutoff)
+            ff[lab] = ctx.ri.vloc(lab)(q)
         self.vloc_g = ctx.make_periodic_function(ff)
         self.vloc_r = ctx.fft_fine.to_real(self.vloc_g).real
 
