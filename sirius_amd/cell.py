@@ -219,6 +219,7 @@ class UnitCell:
         self.atom_types = atom_types
         self.type_labels = list(atom_types.keys())
         self.atoms = [(lab, np.asarray(pos, dtype=np.float64)) for lab, pos in positions]
+        self.vector_fields = np.zeros((len(self.atoms), 3))  # initial moments
 
     @classmethod
     def from_config(cls, cfg, base_dir: str = ".") -> "UnitCell":
@@ -231,11 +232,20 @@ class UnitCell:
             path = fname if os.path.isabs(fname) else os.path.join(base_dir, fname)
             types[lab] = AtomType.from_file(lab, path)
         positions = []
+        vfields = []
         for lab, plist in uc.atoms.items():
             for p in plist:
                 p = np.asarray(p, dtype=np.float64)
-                positions.append((lab, p[:3]))  # entries 4-6 (if any) are initial B field
-        return cls(lattice, types, positions)
+                positions.append((lab, p[:3]))
+                vf = np.zeros(3)
+                if len(p) >= 6:
+                    vf = p[3:6]
+                elif len(p) == 4:
+                    vf[2] = p[3]
+                vfields.append(vf)
+        cell = cls(lattice, types, positions)
+        cell.vector_fields = np.array(vfields)  # initial moments (atom entries 4-6)
+        return cell
 
     # -- queries -----------------------------------------------------------
 

@@ -27,9 +27,12 @@ class Density:
         dev = ctx.device
         self.rho_g = torch.zeros(g.num_gvec, dtype=ctx.dtype, device=dev)
         self.rho_r = torch.zeros(*ctx.fft_fine.dims, dtype=ctx.rdtype, device=dev)
-        self.mag_r = []  # magnetization (collinear: [mz(r)]), fine grid
+        # collinear magnetization m_z (num_mag_dims == 1)
+        self.mag_g = torch.zeros_like(self.rho_g) if ctx.num_mag_dims else None
+        self.mag_r = torch.zeros_like(self.rho_r) if ctx.num_mag_dims else None
         self.rho_core_r = torch.zeros_like(self.rho_r)
         self.rho_core_g = None
+        self.density_matrix = None
         self.mixer = None
         self._gen_core_density()
 
@@ -72,6 +75,25 @@ class Density:
         # clamp negative interstitial values like the reference init does
         self.rho_r = torch.clamp(self.rho_r, min=0.0)
         self.rho_g = ctx.fft_fine.to_pw(self.rho_r.to(ctx.dtype))
+        if ctx.num_mag_dims:
+            # smooth G-space initial magnetization: per-atom Gaussian blob
+            # carrying the starting moment (density.cpp:215-244,
+            # settings.smooth_initial_mag branch)
+            import math as _m
+
+            g = ctx.gvec_fine
+            gw = np.exp(-g.gk_len**2 / 16.0)  # alpha = 4
+            mz = np.zeros(g.num_gvec, dtype=np.complex128)
+            m = g.miller.astype(np.float64)
+            tau = uc.atom_positions_frac()
+            for ia in range(uc.num_atoms):
+                v = uc.vector_fields[ia]
+                if abs(v[2]) < 1e-12:
+                    continue
+                ph = np.exp(-2j * _m.pi * (m @ tau[ia]))
+                mz += v[2] * gw * ph / uc.omega
+            self.mag_g = torch.from_numpy(mz).to(ctx.device)
+            self.mag_r = ctx.fft_fine.to_real(self.mag_g).real
         return self
 
     # -- generation from KS states ----------------------------------------
@@ -85,26 +107,37 @@ class Density:
         ctx = self.ctx
         comm = get_comm()
         dims = ctx.coarse_dims
-        rho_c = torch.zeros(*dims, dtype=ctx.rdtype, device=ctx.device)
+        nsp = ctx.num_spins
+        rho_c = [torch.zeros(*dims, dtype=ctx.rdtype, device=ctx.device)
+                 for _ in range(nsp)]
         min_occ = ctx.cfg.iterative_solver.min_occupancy
 
         for kp in kset:
-            for ispn in range(ctx.num_spins):
+            for ispn in range(nsp):
                 occ = torch.from_numpy(kp.occ[ispn]).to(ctx.device)
                 sel = torch.nonzero(occ.abs() > min_occ).reshape(-1)
                 if len(sel) == 0:
                     continue
                 w = ((kp.weight / ctx.unit_cell.omega) * occ[sel]).to(ctx.rdtype)
-                kp.fft.density_accumulate(kp.psi[ispn][sel], w, rho_c)
+                kp.fft.density_accumulate(kp.psi[ispn][sel], w, rho_c[ispn])
 
-        if comm.active:
-            comm.allreduce_(rho_c)
+        for t in rho_c:
+            if comm.active:
+                comm.allreduce_(t)
 
         # coarse real -> coarse sphere -> fine sphere
-        rho_cg = ctx.fft_coarse.to_pw(rho_c.to(ctx.dtype))
-        self.rho_g = torch.zeros(ctx.gvec_fine.num_gvec, dtype=ctx.dtype,
-                                 device=ctx.device)
-        self.rho_g[ctx.coarse_to_fine] = rho_cg
+        nGf = ctx.gvec_fine.num_gvec
+        def to_fine(t):
+            cg = ctx.fft_coarse.to_pw(t.to(ctx.dtype))
+            out = torch.zeros(nGf, dtype=ctx.dtype, device=ctx.device)
+            out[ctx.coarse_to_fine] = cg
+            return out
+
+        if nsp == 1:
+            self.rho_g = to_fine(rho_c[0])
+        else:
+            self.rho_g = to_fine(rho_c[0] + rho_c[1])
+            self.mag_g = to_fine(rho_c[0] - rho_c[1])
         # augmentation charge (USPP/PAW)
         if ctx.has_aug:
             dm = self.generate_density_matrix(kset)
@@ -113,13 +146,21 @@ class Density:
 
                 dm = symmetrize_density_matrix(dm, ctx, ctx.symmetry.ops)
             self.density_matrix = dm
-            self.rho_g = self.rho_g + self.generate_rho_aug(dm)
+            aug = self.generate_rho_aug(dm)
+            self.rho_g = self.rho_g + aug[0]
+            if nsp == 2:
+                self.mag_g = self.mag_g + aug[1]
         if ctx.symmetry is not None:
             from .symmetry import symmetrize_rho_g
 
             self.rho_g = symmetrize_rho_g(self.rho_g, ctx.gvec_fine,
                                           ctx.symmetry.ops)
+            if nsp == 2:
+                self.mag_g = symmetrize_rho_g(self.mag_g, ctx.gvec_fine,
+                                              ctx.symmetry.ops)
         self.rho_r = ctx.fft_fine.to_real(self.rho_g).real
+        if nsp == 2:
+            self.mag_r = ctx.fft_fine.to_real(self.mag_g).real
         return self
 
     def generate_density_matrix(self, kset):
@@ -172,8 +213,9 @@ class Density:
         generate_dm_pw / sum_q_pw_dm_pw kernels)."""
         ctx = self.ctx
         uc = ctx.unit_cell
-        out = torch.zeros(ctx.gvec_fine.num_gvec, dtype=ctx.dtype,
-                          device=ctx.device)
+        nch = 2 if ctx.num_spins == 2 else 1
+        out = [torch.zeros(ctx.gvec_fine.num_gvec, dtype=ctx.dtype,
+                           device=ctx.device) for _ in range(nch)]
         for lab, at in uc.atom_types.items():
             if not (at.augment and at.num_beta):
                 continue
@@ -185,12 +227,18 @@ class Density:
                     xi1_idx.append(xi1)
                     xi2_idx.append(xi2)
             d = dm[lab]                                    # [na, nbf, nbf, nspin]
-            # non-magnetic / collinear total: sum over spins, Re part
-            dm_aux = d[:, xi2_idx, xi1_idx, :].sum(-1).real.T   # [nqlm, na]
+            dpk = d[:, xi2_idx, xi1_idx, :]                # [na, nqlm, nspin]
+            # channels (density_matrix_aux, density.cpp:1782-1810):
+            # ch0 = Re Σ_σ dm, ch1 = Re(dm↑ − dm↓)
+            chans = [dpk.sum(-1).real.T]
+            if nch == 2:
+                chans.append((dpk[..., 0] - dpk[..., 1]).real.T)
             phases = ctx.phase_pos(lab).conj()             # e^{-iGτ} [na, nG]
-            dm_pw = dm_aux.to(ctx.dtype) @ phases          # [nqlm, nG]
-            out += torch.einsum("q,qg,qg->g", aug.sym_weight.to(ctx.dtype),
-                                aug.q_pw, dm_pw)
+            for ic, dm_aux in enumerate(chans):
+                dm_pw = dm_aux.to(ctx.dtype) @ phases      # [nqlm, nG]
+                out[ic] += torch.einsum("q,qg,qg->g",
+                                        aug.sym_weight.to(ctx.dtype),
+                                        aug.q_pw, dm_pw)
         return out
 
     def check_num_electrons(self) -> float:
@@ -209,13 +257,28 @@ class Density:
             return omega * float(torch.vdot(x, y).real)
 
         comps = [Component("rho_g", inner=inner_pw, global_size=omega)]
+        init = {"rho_g": self.rho_g}
+        if self.ctx.num_spins == 2:
+            comps.append(Component("mag_g", inner=inner_pw, global_size=omega))
+            init["mag_g"] = self.mag_g
         self.mixer = make_mixer(cfg_mixer, comps)
-        self.mixer.initialize({"rho_g": self.rho_g})
+        self.mixer.initialize(init)
 
     def mix(self) -> float:
-        self.mixer.set_input({"rho_g": self.rho_g})
+        inp = {"rho_g": self.rho_g}
+        if self.ctx.num_spins == 2:
+            inp["mag_g"] = self.mag_g
+        self.mixer.set_input(inp)
         rms = self.mixer.mix(self.ctx.cfg.mixer.rms_min)
         out = self.mixer.get_output()
         self.rho_g = out["rho_g"]
         self.rho_r = self.ctx.fft_fine.to_real(self.rho_g).real
+        if self.ctx.num_spins == 2:
+            self.mag_g = out["mag_g"]
+            self.mag_r = self.ctx.fft_fine.to_real(self.mag_g).real
         return rms
+
+    def total_magnetization(self) -> float:
+        if self.mag_r is None:
+            return 0.0
+        return self.ctx.integrate_rg_fine(self.mag_r)

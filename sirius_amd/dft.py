@@ -99,11 +99,11 @@ def diagonalize(ctx, h0: Hamiltonian0, kset: KPointSet, itsol_tol: float) -> boo
     all_conv = True
     for kp in kset:
         hk = h0(kp)
-        h_diag = hk.h_diag()
         o_diag = hk.o_diag()
         for ispn in range(ctx.num_spins):
+            h_diag = hk.h_diag(ispn)
             res = davidson(
-                lambda p: hk.apply_h_s(p, ispn),
+                lambda p, s=ispn: hk.apply_h_s(p, s),
                 kp.psi[ispn], h_diag, o_diag, occ=kp.occ[ispn],
                 tol_occ=itsol_tol, tol_empty=empy_tol,
                 num_steps=itso.num_steps, subspace_size=itso.subspace_size,
@@ -127,15 +127,19 @@ class DFTGroundState:
 
     # -- energies (energy.cpp) --------------------------------------------
 
-    def energy_potential(self, rho_r: torch.Tensor) -> float:
-        """∫ρ V_eff (+ E_bxc + PAW/hubbard terms when present)."""
-        return self.ctx.integrate_rg_fine(rho_r * self.potential.veff_r)
+    def energy_potential(self, rho_r: torch.Tensor, mag_r=None) -> float:
+        """∫ρ V_eff + ∫m·B (+ PAW/hubbard terms when present;
+        reference energy.cpp:251-257)."""
+        e = self.ctx.integrate_rg_fine(rho_r * self.potential.veff_r)
+        if mag_r is not None and self.potential.bz_r is not None:
+            e += self.ctx.integrate_rg_fine(mag_r * self.potential.bz_r)
+        return e
 
     def total_energy_components(self) -> dict:
         d = {}
         d["valence_eval_sum"] = self.kset.valence_eval_sum()
         d["vxc"] = self.potential.energy_vxc(self.density)
-        d["bxc"] = 0.0
+        d["bxc"] = self.potential.energy_bxc(self.density)
         d["vha"] = self.potential.energy_vha
         d["exc"] = self.potential.energy_exc(self.density)
         d["vloc"] = self.potential.energy_vloc(self.density)
@@ -187,8 +191,9 @@ class DFTGroundState:
             self.kset.find_band_occupancies()
             self.density.generate(self.kset, h0)
 
-            e1 = self.energy_potential(self.density.rho_r)
+            e1 = self.energy_potential(self.density.rho_r, self.density.mag_r)
             rho1_r = self.density.rho_r.clone()
+            mag1_r = self.density.mag_r.clone() if self.density.mag_r is not None else None
 
             rms = self.density.mix()
 
@@ -201,6 +206,8 @@ class DFTGroundState:
             self.potential.generate(self.density)
 
             e2 = self.ctx.integrate_rg_fine(rho1_r * self.potential.veff_r)
+            if mag1_r is not None and self.potential.bz_r is not None:
+                e2 += self.ctx.integrate_rg_fine(mag1_r * self.potential.bz_r)
             self.scf_correction_energy = e2 - e1
 
             etot = self.total_energy()
@@ -227,5 +234,7 @@ class DFTGroundState:
             "rms_history": rms_hist,
             "scf_time": time.time() - t0,
             "efermi": self.kset.energy_fermi,
+            "magnetization": self.density.total_magnetization()
+            if ctx.num_mag_dims else 0.0,
         }
         return out

@@ -127,34 +127,53 @@ class Hamiltonian0:
     def __init__(self, ctx, potential, density=None):
         self.ctx = ctx
         self.potential = potential
-        # map veff from fine sphere to coarse sphere, then to coarse real grid
-        veff_coarse_g = potential.veff_g[ctx.coarse_to_fine]
-        self.veff_r_coarse = ctx.fft_coarse.to_real(veff_coarse_g).real
+        nsp = ctx.num_spins
+
+        def to_coarse(fg):
+            return ctx.fft_coarse.to_real(fg[ctx.coarse_to_fine]).real
+
+        veff_c = to_coarse(potential.veff_g)
+        if nsp == 1:
+            self.veff_r_coarse = [veff_c]
+        else:
+            bz_c = to_coarse(potential.bz_g)
+            # spin 0 = up: V + Bz; spin 1 = dn: V − Bz (non_local_operator.cpp:236-238)
+            self.veff_r_coarse = [veff_c + bz_c, veff_c - bz_c]
         uc = ctx.unit_cell
-        # per-atom D matrices [nbf, nbf] (numpy, real)
-        self.d_atom = [None] * uc.num_atoms
+        # per-atom, per-spin D matrices [nbf, nbf] (numpy, real)
+        self.d_atom = [[None] * uc.num_atoms for _ in range(nsp)]
         for lab, at in uc.atom_types.items():
             ions = d_matrix_lm(at)
             ia_list = uc.atoms_of_type(lab)
             if at.augment and at.num_beta:
                 aug = ctx.aug_op(lab)
                 ph = ctx.phase_pos(lab)                    # [na, nG]
-                va = potential.veff_g[None, :] * ph         # [na, nG]
-                dt = (aug.q_pw.conj() @ va.T).real * uc.omega  # [nqlm, na]
-                dt = dt.cpu().numpy()
+
+                def dints(fg):
+                    va = fg[None, :] * ph
+                    return ((aug.q_pw.conj() @ va.T).real * uc.omega).cpu().numpy()
+
+                dt0 = dints(potential.veff_g)               # [nqlm, na]
+                dt1 = dints(potential.bz_g) if nsp == 2 else None
                 nbf = aug.nbf
+                iu, il = np.triu_indices(nbf)   # xi1 <= xi2 pairs (row=xi1)
                 for i, ia in enumerate(ia_list):
-                    d = np.array(ions)
-                    for xi2 in range(nbf):
-                        for xi1 in range(xi2 + 1):
-                            idx12 = xi2 * (xi2 + 1) // 2 + xi1
-                            d[xi1, xi2] += dt[idx12, i]
-                            if xi1 != xi2:
-                                d[xi2, xi1] += dt[idx12, i]
-                    self.d_atom[ia] = d
+                    for ispn in range(nsp):
+                        dt = dt0[:, i] if nsp == 1 else (
+                            dt0[:, i] + dt1[:, i] if ispn == 0 else dt0[:, i] - dt1[:, i])
+                        d = np.array(ions)
+                        k = 0
+                        for xi2 in range(nbf):
+                            for xi1 in range(xi2 + 1):
+                                idx12 = xi2 * (xi2 + 1) // 2 + xi1
+                                d[xi1, xi2] += dt[idx12]
+                                if xi1 != xi2:
+                                    d[xi2, xi1] += dt[idx12]
+                        self.d_atom[ispn][ia] = d
             else:
                 for ia in ia_list:
-                    self.d_atom[ia] = ions
+                    for ispn in range(nsp):
+                        self.d_atom[ispn][ia] = ions
 
     def __call__(self, kp) -> "HamiltonianK":
         return HamiltonianK(self, kp)
@@ -172,7 +191,9 @@ class HamiltonianK:
             kp.beta = BetaProjectors(self.ctx, kp)
         self.bp = kp.beta
         ctx = self.ctx
-        self.D = block_diag_matrix(ctx, self.bp, lambda ia: h0.d_atom[ia])
+        self.D = [block_diag_matrix(ctx, self.bp,
+                                    lambda ia, s=s: h0.d_atom[s][ia])
+                  for s in range(ctx.num_spins)]
         self.Q = None  # USPP overlap augmentation (else S = I)
         if ctx.has_aug:
             uc = ctx.unit_cell
@@ -188,11 +209,11 @@ class HamiltonianK:
     def h_diag(self, ispn: int = 0) -> torch.Tensor:
         """Diagonal of H for the Davidson preconditioner
         (reference get_h_o_diag_pw, hamiltonian_k.cpp:67)."""
-        v0 = float(self.h0.veff_r_coarse.mean())
+        v0 = float(self.h0.veff_r_coarse[ispn].mean())
         d = self.ekin + v0
         if self.bp.num_beta_total:
             # sum_{ij} conj(β_i(G)) D_ij β_j(G) per G
-            bD = self.bp.beta @ self.D.T  # [nGk, nbf]
+            bD = self.bp.beta @ self.D[ispn].T  # [nGk, nbf]
             d = d + torch.einsum("gi,gi->g", bD, self.bp.beta.conj()).real
         return d
 
@@ -207,12 +228,13 @@ class HamiltonianK:
         """psi [nb, nGk] -> (hpsi, spsi). spsi is None for NC (S = I)."""
         kp = self.kp
         # fused local operator: FFT⁻¹ → ×V_eff(r) → FFT → +½|G+k|²ψ
-        hpsi = kp.fft.apply_veff_kinetic(psi, self.h0.veff_r_coarse, self.ekin)
+        hpsi = kp.fft.apply_veff_kinetic(psi, self.h0.veff_r_coarse[ispn],
+                                         self.ekin)
 
         spsi = None
         if self.bp.num_beta_total:
             bphi = self.bp.inner(psi)           # [nbf, nb]
-            hpsi += (self.bp.beta @ (self.D @ bphi)).T
+            hpsi += (self.bp.beta @ (self.D[ispn] @ bphi)).T
             if self.Q is not None:
                 spsi = psi + (self.bp.beta @ (self.Q @ bphi)).T
         return hpsi, spsi

@@ -32,7 +32,8 @@ class Potential:
         self.exc_r = None
         self.veff_r = None            # fine grid real values
         self.veff_g = None
-        self.bxc_r = []               # magnetic fields (collinear: [Bz])
+        self.bz_r = None              # collinear XC magnetic field B_z(r)
+        self.bz_g = None
         self.energy_vha = 0.0
         self.ewald = None
         self._ig0 = g.index_of_zero()
@@ -53,11 +54,30 @@ class Potential:
 
     # -- per-iteration generation ------------------------------------------
 
+    def _grad_r(self, f_r):
+        """∇f of a real fine-grid field via iG (3 FFTs)."""
+        ctx = self.ctx
+        fg = ctx.fft_fine.to_pw(f_r.to(ctx.dtype))
+        gvec = ctx.gvec_fine.gkvec_t
+        return [ctx.fft_fine.to_real(1j * gvec[:, d] * fg).real for d in range(3)]
+
+    def _div_r(self, F):
+        """∇·F from 3 real fields via iG."""
+        ctx = self.ctx
+        gvec = ctx.gvec_fine.gkvec_t
+        out = None
+        for d in range(3):
+            fg = ctx.fft_fine.to_pw(F[d].to(ctx.dtype))
+            t = ctx.fft_fine.to_real(1j * gvec[:, d] * fg).real
+            out = t if out is None else out + t
+        return out
+
     def generate(self, density):
-        """Build V_H, V_xc, V_eff from the current density.
+        """Build V_H, V_xc, V_eff (+B_z for collinear magnetism).
 
         Mirrors Potential::generate (potential.cpp:236): poisson on valence
-        ρ(G) (incl. augmentation), XC on ρ_valence+ρ_core real-space.
+        ρ(G) (incl. augmentation), XC on ρ_valence+ρ_core real-space
+        (xc_rg_nonmagnetic xc.cpp:26 / xc_rg_magnetic xc.cpp:197).
         """
         ctx = self.ctx
         g = ctx.gvec_fine
@@ -70,35 +90,51 @@ class Potential:
         vha_r = ctx.fft_fine.to_real(self.vha_g).real
         self.energy_vha = ctx.integrate_rg_fine(density.rho_r * vha_r)
 
-        # XC on rho_val + rho_core
         rho_xc = density.rho_r + density.rho_core_r
-        if ctx.is_gga:
-            # grad rho via iG on the fine sphere
-            rho_tot_g = ctx.fft_fine.to_pw(rho_xc.to(ctx.dtype))
-            gvec = g.gkvec_t  # [nG,3]
-            grads = []
-            for d in range(3):
-                gr = ctx.fft_fine.to_real(1j * gvec[:, d] * rho_tot_g).real
-                grads.append(gr)
-            sigma = grads[0] ** 2 + grads[1] ** 2 + grads[2] ** 2
-            eps, vrho, vsigma = xc_mod.evaluate(ctx.xc_names, rho_xc, sigma)
-            # v = vrho - 2 div(vsigma grad rho)
-            div = torch.zeros_like(vrho)
-            for d in range(3):
-                fg = ctx.fft_fine.to_pw((vsigma * grads[d]).to(ctx.dtype))
-                div = div + ctx.fft_fine.to_real(1j * gvec[:, d] * fg).real
-            self.vxc_r = vrho - 2.0 * div
-            self.exc_r = eps
+        if ctx.num_spins == 1:
+            if ctx.is_gga:
+                grads = self._grad_r(rho_xc)
+                sigma = grads[0] ** 2 + grads[1] ** 2 + grads[2] ** 2
+                eps, vrho, vsigma = xc_mod.evaluate(ctx.xc_names, rho_xc, sigma)
+                div = self._div_r([vsigma * gr for gr in grads])
+                self.vxc_r = vrho - 2.0 * div
+                self.exc_r = eps
+            else:
+                eps, vrho, _ = xc_mod.evaluate(ctx.xc_names, rho_xc)
+                self.vxc_r = vrho
+                self.exc_r = eps
+            self.bz_r = None
         else:
-            eps, vrho, _ = xc_mod.evaluate(ctx.xc_names, rho_xc)
-            self.vxc_r = vrho
+            # collinear: rho_up/dn = (rho + core ± m_z)/2 (get_rho_up_dn)
+            ru = 0.5 * (rho_xc + density.mag_r)
+            rd = 0.5 * (rho_xc - density.mag_r)
+            if ctx.is_gga:
+                gu = self._grad_r(ru)
+                gd = self._grad_r(rd)
+                s_uu = gu[0] ** 2 + gu[1] ** 2 + gu[2] ** 2
+                s_dd = gd[0] ** 2 + gd[1] ** 2 + gd[2] ** 2
+                gt = [gu[d] + gd[d] for d in range(3)]
+                s_tot = gt[0] ** 2 + gt[1] ** 2 + gt[2] ** 2
+                eps, vu, vd, vs_uu, vs_dd, vs_tot = xc_mod.evaluate_spin(
+                    ctx.xc_names, ru, rd, s_uu, s_dd, s_tot)
+                vu = vu - 2.0 * self._div_r([vs_uu * gu[d] for d in range(3)])                         - 2.0 * self._div_r([vs_tot * gt[d] for d in range(3)])
+                vd = vd - 2.0 * self._div_r([vs_dd * gd[d] for d in range(3)])                         - 2.0 * self._div_r([vs_tot * gt[d] for d in range(3)])
+            else:
+                eps, vu, vd, *_ = xc_mod.evaluate_spin(ctx.xc_names, ru, rd)
+            self.vxc_r = 0.5 * (vu + vd)
+            self.bz_r = 0.5 * (vu - vd)
             self.exc_r = eps
+            self.bz_g = ctx.fft_fine.to_pw(self.bz_r.to(ctx.dtype))
 
         self.veff_r = self.vloc_r + vha_r + self.vxc_r
         self.veff_g = ctx.fft_fine.to_pw(self.veff_r.to(ctx.dtype))
-
-        # collinear magnetism: Bxc = vxc_up - vxc_dn (added with spin support)
         return self
+
+    def energy_bxc(self, density) -> float:
+        """∫ m_z·B_z (reference energy.cpp:85-93)."""
+        if self.bz_r is None or density.mag_r is None:
+            return 0.0
+        return self.ctx.integrate_rg_fine(density.mag_r * self.bz_r)
 
     # -- energies ----------------------------------------------------------
 

@@ -159,28 +159,44 @@ def ibz_mesh(cell, ngridk, shiftk, ops: list[SymOp]):
     return np.array(irr), np.array(weights)
 
 
-def symmetrize_rho_g(rho_g, gvec, ops: list[SymOp], device=None):
-    """ρ_sym(m0) = (1/N) Σ_op e^{2πi m_op·t} ρ̂(m_op), m_op = m0·W^{-T}
-    (reference: symmetrize_pw_function.hpp via Gvec_shells)."""
-    import torch
+class RhoSymmetrizer:
+    """Precomputed gather indices + phases for PW symmetrization:
+    ρ_sym(m0) = (1/N) Σ_op e^{2πi m_op·t} ρ̂(m_op), m_op = m0·W^{-T}
+    (reference: symmetrize_pw_function.hpp via Gvec_shells). One batched
+    gather per call — GPU-friendly, no per-op host work."""
 
-    m = gvec.miller  # [nG, 3]
-    key = {tuple(mm): i for i, mm in enumerate(m)}
-    n = len(m)
-    acc = torch.zeros_like(rho_g)
-    cnt = 0
-    for op in ops:
-        Winvt = np.round(np.linalg.inv(op.W)).astype(np.int64).T
-        m_op = m @ Winvt
-        idx = np.array([key.get(tuple(mm), -1) for mm in m_op], dtype=np.int64)
-        ok = idx >= 0
-        phase = np.exp(2j * math.pi * (m_op @ op.t))
-        idx_t = torch.from_numpy(idx[ok]).to(rho_g.device)
-        ph_t = torch.from_numpy(phase[ok]).to(rho_g.device)
-        sel = torch.from_numpy(np.nonzero(ok)[0]).to(rho_g.device)
-        acc[sel] += ph_t * rho_g[idx_t]
-        cnt += 1
-    return acc / cnt
+    def __init__(self, gvec, ops: list[SymOp], device):
+        import torch
+
+        m = gvec.miller
+        n1, n2, n3 = gvec.dims
+        size = n1 * n2 * n3
+        inv = np.full(size, -1, dtype=np.int64)
+        lin0 = (np.mod(m[:, 0], n1) * n2 + np.mod(m[:, 1], n2)) * n3 + np.mod(m[:, 2], n3)
+        inv[lin0] = np.arange(len(m))
+        idx_all, ph_all = [], []
+        for op in ops:
+            Winvt = np.round(np.linalg.inv(op.W)).astype(np.int64).T
+            m_op = m @ Winvt
+            lin = (np.mod(m_op[:, 0], n1) * n2 + np.mod(m_op[:, 1], n2)) * n3 \
+                + np.mod(m_op[:, 2], n3)
+            idx = inv[lin]
+            assert (idx >= 0).all(), "rotated G left the sphere"
+            idx_all.append(idx)
+            ph_all.append(np.exp(2j * math.pi * (m_op @ op.t)))
+        self.idx = torch.from_numpy(np.stack(idx_all)).to(device)      # [nops, nG]
+        self.ph = torch.from_numpy(np.stack(ph_all)).to(device)        # [nops, nG]
+        self.nops = len(ops)
+
+    def __call__(self, rho_g):
+        return (self.ph * rho_g[self.idx]).sum(0) / self.nops
+
+
+def symmetrize_rho_g(rho_g, gvec, ops: list[SymOp], cache={}):
+    key = (id(gvec), id(ops), str(rho_g.device))
+    if key not in cache:
+        cache[key] = RhoSymmetrizer(gvec, ops, rho_g.device)
+    return cache[key](rho_g)
 
 
 def rlm_rotation_matrices(lmax: int, S: np.ndarray) -> list[np.ndarray]:
@@ -218,30 +234,43 @@ def symmetrize_density_matrix(dm: dict, ctx, ops: list[SymOp]):
     uc = ctx.unit_cell
     out = {lab: torch.zeros_like(d) for lab, d in dm.items()}
     nops = len(ops)
-    for op in ops:
+    key = ("dm_sym", id(ops))
+    cache = getattr(ctx, "_dm_sym_cache", None)
+    if cache is None or cache[0] != key:
+        # precompute per (op, label): beta rotation matrix T and atom index map
+        pre = {}
         for lab, d in dm.items():
             at = uc.atom_types[lab]
             if at.num_beta == 0:
                 continue
             lmax = max(b.l for b in at.beta)
-            Dl = rlm_rotation_matrices(lmax, op.S)
-            # block-diagonal transform over the lm-resolved beta index
             idxb = at.beta_lm_index()
             nbf = len(idxb)
-            T = np.zeros((nbf, nbf))
-            i = 0
-            while i < nbf:
-                irf, l, m = idxb[i]
-                T[i:i + 2 * l + 1, i:i + 2 * l + 1] = Dl[l]
-                i += 2 * l + 1
-            Tt = torch.from_numpy(T).to(d.device).to(d.dtype)
             ia_list = list(uc.atoms_of_type(lab))
             ia_pos = {ia: i for i, ia in enumerate(ia_list)}
-            for i, ia in enumerate(ia_list):
-                ja = int(op.perm[ia])
-                j = ia_pos[ja]
-                for ispn in range(d.shape[-1]):
-                    out[lab][i, :, :, ispn] += Tt.conj().T @ d[j, :, :, ispn] @ Tt
+            entries = []
+            for op in ops:
+                Dl = rlm_rotation_matrices(lmax, op.S)
+                T = np.zeros((nbf, nbf))
+                i = 0
+                while i < nbf:
+                    irf, l, m = idxb[i]
+                    T[i:i + 2 * l + 1, i:i + 2 * l + 1] = Dl[l]
+                    i += 2 * l + 1
+                Tt = torch.from_numpy(T).to(d.device).to(d.dtype)
+                src = torch.tensor([ia_pos[int(op.perm[ia])] for ia in ia_list],
+                                   device=d.device)
+                entries.append((Tt, src))
+            pre[lab] = entries
+        ctx._dm_sym_cache = (key, pre)
+        cache = ctx._dm_sym_cache
+    pre = cache[1]
+    for lab, d in dm.items():
+        if lab not in pre:
+            continue
+        for Tt, src in pre[lab]:
+            dp = d[src]                                   # [na, nbf, nbf, nspin]
+            out[lab] += torch.einsum("pq,aqrs,rt->apts", Tt.conj().T, dp, Tt)
     for lab in out:
         out[lab] /= nops
     return out

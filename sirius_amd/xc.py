@@ -166,8 +166,216 @@ def gga_c_pbe(rho: torch.Tensor, sigma: torch.Tensor):
     return eps, vrho, vsigma
 
 
+# -- spin-polarized variants (libxc-compatible conventions) -----------------
+# eps = energy density per particle of the total density; E = ∫ n·eps.
+
+_PZ_P = dict(A=0.01555, B=-0.0269, C=0.0007, D=-0.0048,
+             gamma=-0.0843, beta1=1.3981, beta2=0.2611)
+
+
+def _fzeta(z):
+    c = 1.0 / (2.0 ** (4.0 / 3.0) - 2.0)
+    return c * ((1 + z) ** (4.0 / 3.0) + (1 - z) ** (4.0 / 3.0) - 2.0)
+
+
+def _dfzeta(z):
+    c = 1.0 / (2.0 ** (4.0 / 3.0) - 2.0)
+    return c * (4.0 / 3.0) * ((1 + z) ** (1.0 / 3.0) - (1 - z) ** (1.0 / 3.0))
+
+
+def lda_x_spin(ru: torch.Tensor, rd: torch.Tensor):
+    """Exchange by exact spin scaling: E_x[n↑,n↓] = ½(E_x[2n↑]+E_x[2n↓])."""
+    ru = _safe_rho(ru)
+    rd = _safe_rho(rd)
+    n = ru + rd
+    cx = (3.0 / 4.0) * (3.0 / math.pi) ** _THIRD
+    eu = -cx * (2.0 * ru) ** _THIRD
+    ed = -cx * (2.0 * rd) ** _THIRD
+    eps = (ru * eu + rd * ed) / n
+    vu = -(4.0 / 3.0) * cx * (2.0 * ru) ** _THIRD
+    vd = -(4.0 / 3.0) * cx * (2.0 * rd) ** _THIRD
+    return eps, vu, vd
+
+
+def _pz_branch(rs, p):
+    lo = rs < 1.0
+    lnrs = torch.log(rs)
+    eps_lo = p["A"] * lnrs + p["B"] + p["C"] * rs * lnrs + p["D"] * rs
+    deps_lo = p["A"] / rs + p["C"] * (lnrs + 1.0) + p["D"]
+    sq = torch.sqrt(rs)
+    den = 1.0 + p["beta1"] * sq + p["beta2"] * rs
+    eps_hi = p["gamma"] / den
+    deps_hi = -p["gamma"] * (0.5 * p["beta1"] / sq + p["beta2"]) / den**2
+    return torch.where(lo, eps_lo, eps_hi), torch.where(lo, deps_lo, deps_hi)
+
+
+def lda_c_pz_spin(ru: torch.Tensor, rd: torch.Tensor):
+    ru = _safe_rho(ru)
+    rd = _safe_rho(rd)
+    n = ru + rd
+    z = torch.clamp((ru - rd) / n, -1.0 + 1e-15, 1.0 - 1e-15)
+    rs = (3.0 / (4.0 * math.pi * n)) ** _THIRD
+    eU, dU = _pz_branch(rs, _PZ_U)
+    eP, dP = _pz_branch(rs, _PZ_P)
+    f = _fzeta(z)
+    df = _dfzeta(z)
+    eps = eU + f * (eP - eU)
+    deps_drs = dU + f * (dP - dU)
+    deps_dz = df * (eP - eU)
+    common = eps - rs / 3.0 * deps_drs
+    vu = common + deps_dz * (1.0 - z)
+    vd = common - deps_dz * (1.0 + z)
+    return eps, vu, vd
+
+
+def lda_c_pw_spin(ru: torch.Tensor, rd: torch.Tensor):
+    """Perdew-Wang 92 with full ζ interpolation."""
+    ru = _safe_rho(ru)
+    rd = _safe_rho(rd)
+    n = ru + rd
+    z = torch.clamp((ru - rd) / n, -1.0 + 1e-15, 1.0 - 1e-15)
+    rs = (3.0 / (4.0 * math.pi * n)) ** _THIRD
+    ecU, dU = _pw_g(rs, 0.0310907, 0.21370, 7.5957, 3.5876, 1.6382, 0.49294)
+    ecP, dP = _pw_g(rs, 0.01554535, 0.20548, 14.1189, 6.1977, 3.3662, 0.62517)
+    mac, dmac = _pw_g(rs, 0.0168869, 0.11125, 10.357, 3.6231, 0.88026, 0.49671)
+    ac, dac = -mac, -dmac  # spin stiffness (PW fit returns -alpha_c)
+    f = _fzeta(z)
+    df = _dfzeta(z)
+    fdd0 = 4.0 / (9.0 * (2.0 ** (1.0 / 3.0) - 1.0))
+    z4 = z**4
+    eps = ecU + ac * f / fdd0 * (1.0 - z4) + (ecP - ecU) * f * z4
+    deps_drs = dU + dac * f / fdd0 * (1.0 - z4) + (dP - dU) * f * z4
+    deps_dz = ac / fdd0 * (df * (1.0 - z4) - 4.0 * z**3 * f) + \
+        (ecP - ecU) * (df * z4 + 4.0 * z**3 * f)
+    common = eps - rs / 3.0 * deps_drs
+    vu = common + deps_dz * (1.0 - z)
+    vd = common - deps_dz * (1.0 + z)
+    return eps, vu, vd
+
+
+def gga_x_pbe_spin(ru, rd, s_uu, s_dd):
+    """PBE exchange via exact spin scaling. Returns
+    (eps, vu, vd, vs_uu, vs_dd); vsigma_ud = 0 for exchange."""
+    ru = _safe_rho(ru)
+    rd = _safe_rho(rd)
+    n = ru + rd
+    e_u, v_u, vs_u = gga_x_pbe(2.0 * ru, 4.0 * s_uu)
+    e_d, v_d, vs_d = gga_x_pbe(2.0 * rd, 4.0 * s_dd)
+    eps = (ru * e_u + rd * e_d) / n
+    return eps, v_u, v_d, 2.0 * vs_u, 2.0 * vs_d
+
+
+def gga_c_pbe_spin(ru, rd, sigma):
+    """PBE correlation, spin-polarized; sigma = |∇(n↑+n↓)|².
+    Returns (eps, vu, vd, vsigma) with vsigma = ∂(n·eps)/∂σ (same for all
+    σ components: vsigma_uu = vsigma, vsigma_ud = 2·vsigma, vsigma_dd = vsigma
+    in libxc terms — handled by the caller using total-gradient form)."""
+    ru = _safe_rho(ru)
+    rd = _safe_rho(rd)
+    n = ru + rd
+    sigma = torch.clamp(sigma, min=1e-40)
+    z = torch.clamp((ru - rd) / n, -1.0 + 1e-12, 1.0 - 1e-12)
+    rs = (3.0 / (4.0 * math.pi * n)) ** _THIRD
+    phi = 0.5 * ((1 + z) ** (2.0 / 3.0) + (1 - z) ** (2.0 / 3.0))
+    beta, gamma = _PBE_BETA, _PBE_GAMMA
+
+    # need ec(rs, z) and its derivatives — reuse lda_c_pw_spin pieces
+    ecU, dU = _pw_g(rs, 0.0310907, 0.21370, 7.5957, 3.5876, 1.6382, 0.49294)
+    ecP, dP = _pw_g(rs, 0.01554535, 0.20548, 14.1189, 6.1977, 3.3662, 0.62517)
+    mac, dmac = _pw_g(rs, 0.0168869, 0.11125, 10.357, 3.6231, 0.88026, 0.49671)
+    ac, dac = -mac, -dmac
+    f = _fzeta(z)
+    df = _dfzeta(z)
+    fdd0 = 4.0 / (9.0 * (2.0 ** (1.0 / 3.0) - 1.0))
+    z4 = z**4
+    ec = ecU + ac * f / fdd0 * (1.0 - z4) + (ecP - ecU) * f * z4
+    dec_drs = dU + dac * f / fdd0 * (1.0 - z4) + (dP - dU) * f * z4
+    dec_dz = ac / fdd0 * (df * (1.0 - z4) - 4.0 * z**3 * f) + \
+        (ecP - ecU) * (df * z4 + 4.0 * z**3 * f)
+    dphi_dz = ((1 + z) ** (-1.0 / 3.0) - (1 - z) ** (-1.0 / 3.0)) / 3.0
+
+    kf = (3.0 * math.pi**2 * n) ** _THIRD
+    ks = torch.sqrt(4.0 * kf / math.pi)
+    t2 = sigma / (2.0 * phi * ks * n) ** 2
+    g3 = phi**3
+    expo = torch.exp(-ec / (gamma * g3))
+    A = beta / gamma / (expo - 1.0 + 1e-30)
+    At2 = A * t2
+    num = 1.0 + At2
+    den = 1.0 + At2 + At2 * At2
+    arg = 1.0 + beta / gamma * t2 * num / den
+    H = gamma * g3 * torch.log(arg)
+    eps = ec + H
+
+    # partials of H wrt t2, A, phi, ec
+    q = t2 * num / den
+    dq_dt2 = (num + t2 * A) / den - t2 * num * (A + 2.0 * A * At2) / den**2
+    dq_dA = t2 * (t2 * den - num * (t2 + 2.0 * t2 * At2)) / den**2
+    dH_darg = gamma * g3 / arg
+    dH_dt2 = dH_darg * beta / gamma * dq_dt2
+    dH_dA = dH_darg * beta / gamma * dq_dA
+    dA_dec = A * expo / (gamma * g3 * (expo - 1.0 + 1e-30))
+    # d expo/dphi = expo * 3 ec/(gamma g3 phi) => dA/dphi = -A/(expo-1)*dexpo_dphi
+    dexpo_dphi = expo * 3.0 * ec / (gamma * g3 * phi)
+    dA_dphi = -A * dexpo_dphi / (expo - 1.0 + 1e-30)
+    dH_dphi = 3.0 * H / phi + dH_dA * dA_dphi - dH_dt2 * 2.0 * t2 / phi
+    dH_dec = dH_dA * dA_dec
+
+    c_t2 = math.pi / 16.0 * (3.0 * math.pi**2) ** (-1.0 / 3.0)
+    # t2 = c_t2 * sigma * n^{-7/3} / phi^2
+    dt2_dn = -(7.0 / 3.0) * c_t2 * sigma * n ** (-10.0 / 3.0) / phi**2
+    dt2_dsigma = c_t2 * n ** (-7.0 / 3.0) / phi**2
+    drs_dn = -rs / (3.0 * n)
+
+    deps_dn_at_z = (dec_drs + dH_dec * dec_drs) * drs_dn + dH_dt2 * dt2_dn
+    deps_dz_tot = dec_dz + dH_dec * dec_dz + dH_dphi * dphi_dz
+    common = eps + n * deps_dn_at_z
+    vu = common + deps_dz_tot * (1.0 - z)
+    vd = common - deps_dz_tot * (1.0 + z)
+    vsigma = n * dH_dt2 * dt2_dsigma
+    return eps, vu, vd, vsigma
+
+
 _LDA = {"XC_LDA_X": lda_x, "XC_LDA_C_PZ": lda_c_pz, "XC_LDA_C_PW": lda_c_pw}
 _GGA = {"XC_GGA_X_PBE": gga_x_pbe, "XC_GGA_C_PBE": gga_c_pbe}
+_LDA_SPIN = {"XC_LDA_X": lda_x_spin, "XC_LDA_C_PZ": lda_c_pz_spin,
+             "XC_LDA_C_PW": lda_c_pw_spin}
+
+
+def evaluate_spin(names: list[str], ru: torch.Tensor, rd: torch.Tensor,
+                  s_uu=None, s_dd=None, s_tot=None):
+    """Collinear spin evaluation. Returns (eps, vu, vd, vs_uu, vs_dd, vs_tot)
+    where GGA exchange uses per-spin gradients (vs_uu/vs_dd) and GGA
+    correlation the total gradient (vs_tot)."""
+    eps = torch.zeros_like(ru)
+    vu = torch.zeros_like(ru)
+    vd = torch.zeros_like(ru)
+    vs_uu = torch.zeros_like(ru) if s_uu is not None else None
+    vs_dd = torch.zeros_like(ru) if s_uu is not None else None
+    vs_tot = torch.zeros_like(ru) if s_uu is not None else None
+    n = _safe_rho(ru + rd)
+    for name in names:
+        if name in _LDA_SPIN:
+            e, a, b = _LDA_SPIN[name](ru, rd)
+            eps = eps + e
+            vu = vu + a
+            vd = vd + b
+        elif name == "XC_GGA_X_PBE":
+            e, a, b, su, sd = gga_x_pbe_spin(ru, rd, s_uu, s_dd)
+            eps = eps + e
+            vu = vu + a
+            vd = vd + b
+            vs_uu = vs_uu + su
+            vs_dd = vs_dd + sd
+        elif name == "XC_GGA_C_PBE":
+            e, a, b, st = gga_c_pbe_spin(ru, rd, s_tot)
+            eps = eps + e
+            vu = vu + a
+            vd = vd + b
+            vs_tot = vs_tot + st
+        else:
+            raise ValueError(f"unsupported spin xc functional: {name}")
+    return eps, vu, vd, vs_uu, vs_dd, vs_tot
 
 
 def is_gga(names: list[str]) -> bool:

@@ -37,6 +37,47 @@ def test23_h_atom_nc_lda():
     assert abs(res["energy"]["total"] - eref) < 1e-5
 
 
+@requires_reference
+@pytest.mark.slow
+def test08_si_uspp_lda():
+    res, eref = run_case("test08")
+    assert res["converged"]
+    assert abs(res["energy"]["total"] - eref) < 1e-5
+
+
+@requires_reference
+@pytest.mark.slow
+def test06_fe_uspp_lda_fm():
+    res, eref = run_case("test06")
+    assert res["converged"]
+    assert abs(res["energy"]["total"] - eref) < 1e-5
+    assert abs(res["magnetization"] - 6.760705375907565) < 1e-3
+
+
+@requires_reference
+@pytest.mark.slow
+def test01_srvo3_uspp_lda():
+    res, eref = run_case("test01")
+    assert res["converged"]
+    assert abs(res["energy"]["total"] - eref) < 1e-5
+
+
+@requires_reference
+@pytest.mark.slow
+def test14_srvo3_uspp_pbe():
+    res, eref = run_case("test14")
+    assert res["converged"]
+    assert abs(res["energy"]["total"] - eref) < 1e-5
+
+
+@requires_reference
+@pytest.mark.slow
+def test07_ni_uspp_pbe_fm():
+    res, eref = run_case("test07")
+    assert res["converged"]
+    assert abs(res["energy"]["total"] - eref) < 1e-5
+
+
 def test_synthetic_si2_runs():
     """A tiny synthetic NC cell steps through the whole SCF machinery
     (no reference needed; exercises beta projectors + LCAO init)."""
