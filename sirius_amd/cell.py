@@ -231,12 +231,21 @@ class UnitCell:
             fname = uc.atom_files.get(lab, f"{lab}.json")
             path = fname if os.path.isabs(fname) else os.path.join(base_dir, fname)
             types[lab] = AtomType.from_file(lab, path)
+        units = uc.get("atom_coordinate_units", "lattice")
+        inv_lat = np.linalg.inv(lattice)
+        from .constants import bohr_to_ang
+
         positions = []
         vfields = []
         for lab, plist in uc.atoms.items():
             for p in plist:
                 p = np.asarray(p, dtype=np.float64)
-                positions.append((lab, p[:3]))
+                pos = p[:3]
+                if units in ("au", "a.u."):
+                    pos = pos @ inv_lat          # cartesian bohr -> fractional
+                elif units in ("A", "angstrom"):
+                    pos = (pos / bohr_to_ang) @ inv_lat
+                positions.append((lab, pos))
                 vf = np.zeros(3)
                 if len(p) >= 6:
                     vf = p[3:6]

@@ -76,24 +76,36 @@ class Density:
         self.rho_r = torch.clamp(self.rho_r, min=0.0)
         self.rho_g = ctx.fft_fine.to_pw(self.rho_r.to(ctx.dtype))
         if ctx.num_mag_dims:
-            # smooth G-space initial magnetization: per-atom Gaussian blob
-            # carrying the starting moment (density.cpp:215-244,
-            # settings.smooth_initial_mag branch)
-            import math as _m
-
-            g = ctx.gvec_fine
-            gw = np.exp(-g.gk_len**2 / 16.0)  # alpha = 4
-            mz = np.zeros(g.num_gvec, dtype=np.complex128)
-            m = g.miller.astype(np.float64)
+            # real-space initial magnetization: per-atom weight blob
+            # w(R,x) = (1−(x/R)²)e^{x/R} inside radius R carrying the atom's
+            # starting moment (density.cpp:246-272, the default
+            # !smooth_initial_mag branch; R from auto MT radii)
+            mz = np.zeros(ctx.fft_fine.dims)
+            dims = ctx.fft_fine.dims
+            frac_grid = np.stack(np.meshgrid(
+                np.arange(dims[0]) / dims[0], np.arange(dims[1]) / dims[1],
+                np.arange(dims[2]) / dims[2], indexing="ij"), axis=-1)
             tau = uc.atom_positions_frac()
+            # nearest-neighbour based MT radii (≈ find_mt_radii(1, true))
+            nn = uc.nearest_neighbours(8.0)
+            dmin = {}
+            for ia, ja, d in nn:
+                dmin[ia] = min(dmin.get(ia, 1e9), d)
             for ia in range(uc.num_atoms):
                 v = uc.vector_fields[ia]
-                if abs(v[2]) < 1e-12:
+                if np.abs(v).max() < 1e-12:
                     continue
-                ph = np.exp(-2j * _m.pi * (m @ tau[ia]))
-                mz += v[2] * gw * ph / uc.omega
-            self.mag_g = torch.from_numpy(mz).to(ctx.device)
-            self.mag_r = ctx.fft_fine.to_real(self.mag_g).real
+                R = min(ctx.cfg.control.rmt_max, 0.5 * dmin.get(ia, 4.0))
+                df = frac_grid - tau[ia]
+                df -= np.round(df)
+                r = np.linalg.norm(df @ uc.lattice, axis=-1)
+                w = np.where(r < R, (1 - (r / R) ** 2) * np.exp(r / R), 0.0)
+                s = w.sum() * uc.omega / np.prod(dims)
+                if s > 1e-12:
+                    mz += v[2] * w / s
+            self.mag_r = torch.from_numpy(mz).to(device=ctx.device,
+                                                 dtype=ctx.rdtype)
+            self.mag_g = ctx.fft_fine.to_pw(self.mag_r.to(ctx.dtype))
         return self
 
     # -- generation from KS states ----------------------------------------

@@ -173,7 +173,7 @@ def make_mixer(cfg_mixer, components: list[Component]) -> Mixer:
               beta0=cfg_mixer.beta0, beta_scaling_factor=cfg_mixer.beta_scaling_factor)
     if kind == "linear":
         return Linear(components, **kw)
-    if kind in ("anderson", "anderson_stable", "broyden2"):
-        # broyden2/anderson_stable fall back to anderson until implemented
+    if kind in ("anderson", "anderson_stable", "broyden1", "broyden2"):
+        # broyden1/broyden2/anderson_stable fall back to anderson until implemented
         return Anderson(components, **kw)
     raise ValueError(f"unknown mixer type {kind}")
