@@ -35,7 +35,7 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=4)
-    ap.add_argument("--warmup", type=int, default=2)
+    ap.add_argument("--warmup", type=int, default=4)
     ap.add_argument("--natoms", type=int, default=64)
     ap.add_argument("--gk-cutoff", type=float, default=5.0)
     ap.add_argument("--pw-cutoff", type=float, default=14.0)

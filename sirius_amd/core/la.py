@@ -18,20 +18,35 @@ import torch
 
 EIGH_GPU_MIN = 1024  # below this, host LAPACK beats rocSOLVER's launch storm
 
+# LAPACK on many-core hosts (128 threads on the MI355X boxes) oversubscribes
+# badly at subspace sizes (~300): 45 ms/zheevd at 128 threads vs ~12 ms at 8.
+_HOST_SOLVE_THREADS = min(16, torch.get_num_threads())
+
+
+class _host_threads:
+    def __enter__(self):
+        self.saved = torch.get_num_threads()
+        torch.set_num_threads(_HOST_SOLVE_THREADS)
+
+    def __exit__(self, *a):
+        torch.set_num_threads(self.saved)
+
 
 def eigh(H: torch.Tensor):
     """Hermitian eigensolve returning (evals, evecs) on H's device."""
     n = H.shape[-1]
     if H.is_cuda and n < EIGH_GPU_MIN:
-        w, v = torch.linalg.eigh(H.cpu())
-        return w.to(H.device), v.to(H.device)
+        with _host_threads():
+            w, v = torch.linalg.eigh(H.cpu())
+        return w.to(H.device, non_blocking=True), v.to(H.device)
     return torch.linalg.eigh(H)
 
 
 def cholesky(S: torch.Tensor):
     n = S.shape[-1]
     if S.is_cuda and n < EIGH_GPU_MIN:
-        return torch.linalg.cholesky(S.cpu()).to(S.device)
+        with _host_threads():
+            return torch.linalg.cholesky(S.cpu()).to(S.device)
     return torch.linalg.cholesky(S)
 
 
@@ -39,9 +54,27 @@ def inv_lower(L: torch.Tensor) -> torch.Tensor:
     """L^{-1} for lower-triangular L (small; host when on GPU)."""
     eye = torch.eye(L.shape[-1], dtype=L.dtype, device=L.device)
     if L.is_cuda and L.shape[-1] < EIGH_GPU_MIN:
-        out = torch.linalg.solve_triangular(L.cpu(), eye.cpu(), upper=False)
+        with _host_threads():
+            out = torch.linalg.solve_triangular(L.cpu(), eye.cpu(), upper=False)
         return out.to(L.device)
     return torch.linalg.solve_triangular(L, eye, upper=False)
+
+
+def ortho_factor(gram: torch.Tensor) -> torch.Tensor:
+    """conj(L^{-1}) for the Cholesky factor of a Gram matrix, in ONE host
+    round trip (the Davidson orthonormalization transform). Raises on a
+    non-positive-definite gram like cholesky does."""
+    n = gram.shape[-1]
+    if gram.is_cuda and n < EIGH_GPU_MIN:
+        with _host_threads():
+            g = gram.cpu()
+            L = torch.linalg.cholesky(g)
+            t = torch.linalg.solve_triangular(
+                L, torch.eye(n, dtype=g.dtype), upper=False).conj()
+        return t.to(gram.device)
+    L = torch.linalg.cholesky(gram)
+    eye = torch.eye(n, dtype=gram.dtype, device=gram.device)
+    return torch.linalg.solve_triangular(L, eye, upper=False).conj()
 
 
 def inner(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:

@@ -59,8 +59,7 @@ def _ortho_joint(new, hnew, snew, phi, hphi, sphi):
     gram = 0.5 * (gram + gram.conj().T)
     n = gram.shape[0]
     try:
-        L = la.cholesky(gram)
-        t = la.inv_lower(L).conj()
+        t = la.ortho_factor(gram)
     except Exception:
         w, v = la.eigh(gram)
         keep = w > 1e-10
