@@ -237,7 +237,11 @@ class UnitCell:
 
         positions = []
         vfields = []
-        for lab, plist in uc.atoms.items():
+        # global atom order follows the atom_types list (reference behavior:
+        # atoms are added per type in unit_cell initialization)
+        type_order = list(uc.atom_types) + [t for t in uc.atoms if t not in uc.atom_types]
+        for lab in type_order:
+            plist = uc.atoms.get(lab, [])
             for p in plist:
                 p = np.asarray(p, dtype=np.float64)
                 pos = p[:3]

@@ -168,6 +168,17 @@ class SimulationContext:
 
         self.ri = RadialIntegralsCache(self)
 
+        # DFT+U (reference: Hubbard class, src/hubbard/)
+        self.hubbard = None
+        if p.hubbard_correction:
+            from .hubbard import HubbardModule
+
+            self.hubbard = HubbardModule(self)
+            if self.hubbard.nonlocal_pairs and self.symmetry is not None:
+                # inter-site occupation symmetrization is not implemented;
+                # run the full k-mesh instead (physically equivalent)
+                self.symmetry = None
+
     # -- augmentation (USPP/PAW) ------------------------------------------
 
     @property

@@ -162,6 +162,8 @@ class Density:
             self.rho_g = self.rho_g + aug[0]
             if nsp == 2:
                 self.mag_g = self.mag_g + aug[1]
+        if ctx.hubbard is not None:
+            ctx.hubbard.generate_occupation_matrix(kset, hamiltonian0)
         if ctx.symmetry is not None:
             from .symmetry import symmetrize_rho_g
 
@@ -273,13 +275,50 @@ class Density:
         if self.ctx.num_spins == 2:
             comps.append(Component("mag_g", inner=inner_pw, global_size=omega))
             init["mag_g"] = self.mag_g
+        if self.ctx.hubbard is not None:
+            if self.ctx.hubbard.om_nl is None:
+                self.ctx.hubbard.om_nl = [
+                    __import__("torch").zeros(2 * p.il + 1, 2 * p.jl + 1,
+                                              self.ctx.num_spins,
+                                              dtype=self.ctx.dtype,
+                                              device=self.ctx.device)
+                    for p in self.ctx.hubbard.nonlocal_pairs]
+            comps.append(Component("hub_om"))
+            init["hub_om"] = self._pack_om()
         self.mixer = make_mixer(cfg_mixer, comps)
         self.mixer.initialize(init)
+
+    def _pack_om(self, om=None):
+        import torch as _t
+
+        hub = self.ctx.hubbard
+        parts = [t.reshape(-1) for t in hub.om]
+        if hub.om_nl:
+            parts += [t.reshape(-1) for t in hub.om_nl]
+        return _t.cat(parts)
+
+    def _unpack_om(self, flat):
+        hub = self.ctx.hubbard
+        off = 0
+        out = []
+        for t in hub.om:
+            n = t.numel()
+            out.append(flat[off:off + n].reshape(t.shape))
+            off += n
+        out_nl = []
+        if hub.om_nl:
+            for t in hub.om_nl:
+                n = t.numel()
+                out_nl.append(flat[off:off + n].reshape(t.shape))
+                off += n
+        return out, out_nl
 
     def mix(self) -> float:
         inp = {"rho_g": self.rho_g}
         if self.ctx.num_spins == 2:
             inp["mag_g"] = self.mag_g
+        if self.ctx.hubbard is not None:
+            inp["hub_om"] = self._pack_om()
         self.mixer.set_input(inp)
         rms = self.mixer.mix(self.ctx.cfg.mixer.rms_min)
         out = self.mixer.get_output()
@@ -288,6 +327,11 @@ class Density:
         if self.ctx.num_spins == 2:
             self.mag_g = out["mag_g"]
             self.mag_r = self.ctx.fft_fine.to_real(self.mag_g).real
+        if self.ctx.hubbard is not None:
+            om, om_nl = self._unpack_om(out["hub_om"])
+            self.ctx.hubbard.om = om
+            if om_nl:
+                self.ctx.hubbard.om_nl = om_nl
         return rms
 
     def total_magnetization(self) -> float:

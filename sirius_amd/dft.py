@@ -133,6 +133,10 @@ class DFTGroundState:
         e = self.ctx.integrate_rg_fine(rho_r * self.potential.veff_r)
         if mag_r is not None and self.potential.bz_r is not None:
             e += self.ctx.integrate_rg_fine(mag_r * self.potential.bz_r)
+        # NOTE: the reference's energy_potential also adds hubbard_energy
+        # (energy.cpp:251-257), but both SCF-correction evaluations use the
+        # SAME occupation matrix, so the Hubbard term cancels in e2-e1 and
+        # is deliberately left out here.
         return e
 
     def total_energy_components(self) -> dict:
@@ -149,6 +153,9 @@ class DFTGroundState:
         d["PAW_one_elec"] = 0.0
         d["scf_correction"] = self.scf_correction_energy
         d["fermi"] = self.kset.energy_fermi
+        hub = self.ctx.hubbard
+        d["hubbard_energy"] = hub.energy() if hub else 0.0
+        d["hubbard_one_el"] = hub.one_electron_energy() if hub else 0.0
         return d
 
     def total_energy(self) -> float:
@@ -156,12 +163,15 @@ class DFTGroundState:
         d = self.total_energy_components()
         return (d["valence_eval_sum"] - d["vxc"] - d["bxc"] - d["PAW_one_elec"]
                 - 0.5 * d["vha"] + d["exc"] + d["PAW_total_energy"] + d["ewald"]
+                + d["hubbard_energy"] - d["hubbard_one_el"]
                 + d["scf_correction"])
 
     # -- driver ------------------------------------------------------------
 
     def initial_state(self):
         self.density.initial_density()
+        if self.ctx.hubbard is not None:
+            self.ctx.hubbard.initial_occupation()
         self.potential.generate(self.density)
         h0 = Hamiltonian0(self.ctx, self.potential)
         for kp in self.kset:

@@ -237,4 +237,6 @@ class HamiltonianK:
             hpsi += (self.bp.beta @ (self.D[ispn] @ bphi)).T
             if self.Q is not None:
                 spsi = psi + (self.bp.beta @ (self.Q @ bphi)).T
+        if self.ctx.hubbard is not None:
+            self.ctx.hubbard.apply(kp, self, psi, hpsi, ispn)
         return hpsi, spsi

@@ -1,0 +1,467 @@
+"""DFT+U (Hubbard correction), collinear case.
+
+Reference behavior: src/hubbard/ —
+- hubbard orbitals: pseudo-atomic wavefunctions of the (n,l) channel,
+  optionally Löwdin-orthogonalized against the FULL atomic-wf set with the
+  S metric (hubbard_subspace_method "full_orthogonalization",
+  k_point.cpp:181-330); occupation uses S|φ⟩ (occupation_matrix.cpp:50-175):
+      n^σ_{m1m2}(a) = (w_k/max_occ) Σ_j ⟨Sφ_m1|ψ_j⟩ f_j ⟨ψ_j|Sφ_m2⟩
+- simplified (Dudarev) potential and energy
+  (hubbard_potential_energy.cpp:79-105, :251-273):
+      V^σ = (α + U_eff/2)·I − U_eff·n^σ,   U_eff = U − J0
+      E_U = Σ_σ [ (α + U_eff/2)·Tr n^σ − (U_eff/2)·Tr(n^σ n^σ) ]
+- apply (non_local_operator.cpp:450-519): hψ += Σ |Sφ_m1⟩ V_{m1m2} ⟨Sφ_m2|ψ⟩
+- one-electron double counting: Re Σ n·conj(V) (×2 for num_spins==1)
+  (hubbard_potential_energy.cpp:647-691).
+"""
+
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass
+
+import numpy as np
+import torch
+
+from .core import la
+
+
+def _rlm_ylm_rlm_table(l: int, k: int) -> np.ndarray:
+    """Re⟨R_{l m1}|Y_{k q}|R_{l m2}⟩ table [2l+1, 2k+1, 2l+1] by exact
+    quadrature (reference: SHT::gaunt_rlm_ylm_rlm, sht.hpp:449-466)."""
+    from .core import ylm as ylm_mod
+
+    ltot = 2 * l + k
+    nth = ltot // 2 + 2
+    nph = ltot + 2
+    x, wx = np.polynomial.legendre.leggauss(nth)
+    theta = np.arccos(x)
+    phi = np.arange(nph) * 2 * np.pi / nph
+    tt, pp = np.meshgrid(theta, phi, indexing="ij")
+    ww = np.broadcast_to(wx[:, None] * (2 * np.pi / nph), tt.shape).reshape(-1)
+    tt = tt.reshape(-1)
+    pp = pp.reshape(-1)
+    R = ylm_mod.rlm(l, tt, pp)[:, l * l:(l + 1) * (l + 1)]          # [np, 2l+1]
+    Y = ylm_mod.ylm(k, tt, pp)[:, k * k:(k + 1) * (k + 1)]          # [np, 2k+1]
+    out = np.einsum("p,pa,pq,pb->aqb", ww, R, Y, R).real
+    return out
+
+
+def hubbard_matrix_full(l: int, U: float, J: float, B: float = 0.0,
+                        E2: float = 0.0, E3: float = 0.0) -> np.ndarray:
+    """⟨m1 m2|V_ee|m3 m4⟩ stored as matrix(m1, m3, m2, m4), replicating the
+    reference verbatim (hubbard_orbitals_descriptor.hpp:66-169 — including
+    its k < 2l truncation of the Slater sum and F coefficients)."""
+    F = [U]
+    if l == 0:
+        F += [J]
+    elif l == 1:
+        F += [5.0 * J]
+    elif l == 2:
+        F += [5.0 * J + 31.5 * B, 9.0 * J - 31.5 * B]
+    elif l == 3:
+        F += [(225.0 / 54.0) * J + (32175.0 / 42.0) * E2 + (2475.0 / 42.0) * E3,
+              11.0 * J - (141570.0 / 77.0) * E2 + (4356.0 / 77.0) * E3,
+              (7361.640 / 594.0) * J + (36808.20 / 66.0) * E2 - 111.54 * E3]
+    mm = 2 * l + 1
+    hm = np.zeros((mm, mm, mm, mm))
+    for ki, k in enumerate(range(0, 2 * l, 2)):   # k = 0, 2, ..., 2l-2
+        g = _rlm_ylm_rlm_table(l, k)              # [mm, 2k+1, mm]
+        ak = 4.0 * math.pi / (2 * k + 1) * np.einsum("aqb,cqd->abcd", g, g)
+        # ak(m1, m2, m3, m4) with pairs (m1,m2) and (m3,m4);
+        # hubbard_matrix(m1, m2, m3, m4) += ak(m1, m3, m2, m4) F[k]
+        hm += np.transpose(ak, (0, 2, 1, 3)) * F[ki]
+    return hm
+
+
+@dataclass
+class NonlocalPair:
+    ia: int
+    ja: int
+    il: int
+    jl: int
+    n1: int
+    n2: int
+    V: float
+    T: tuple
+
+
+@dataclass
+class HubbardOrbital:
+    label: str          # atom type
+    n: int
+    l: int
+    U: float
+    J: float = 0.0
+    J0: float = 0.0
+    alpha: float = 0.0
+    beta: float = 0.0
+    initial_occupancy: float = 0.0
+
+
+class HubbardModule:
+    """Holds orbital descriptors, per-k Hubbard wavefunctions, the
+    occupation matrices and the U potential matrices."""
+
+    def __init__(self, ctx):
+        self.ctx = ctx
+        cfg = ctx.cfg.hubbard
+        self.simplified = bool(cfg.get("simplified", False))
+        self.subspace_method = cfg.get("hubbard_subspace_method", "none")
+        self.orbitals: list[HubbardOrbital] = []
+        for e in cfg.get("local", []):
+            self.orbitals.append(HubbardOrbital(
+                label=e["atom_type"], n=int(e.get("n", -1)), l=int(e["l"]),
+                U=float(e.get("U", 0.0)), J=float(e.get("J", 0.0)),
+                J0=float(e.get("J0", 0.0)),
+                alpha=float(e.get("alpha", 0.0)), beta=float(e.get("beta", 0.0)),
+                initial_occupancy=float(e.get("total_initial_occupancy", 0.0))))
+        self.nonlocal_pairs: list[NonlocalPair] = []
+        for e in cfg.get("nonlocal", []):
+            self.nonlocal_pairs.append(NonlocalPair(
+                ia=int(e["atom_pair"][0]), ja=int(e["atom_pair"][1]),
+                il=int(e["l"][0]), jl=int(e["l"][1]),
+                n1=int(e["n"][0]), n2=int(e["n"][1]),
+                V=float(e.get("V", 0.0)), T=tuple(int(x) for x in e["T"])))
+        self.desc_by_label = {o.label: o for o in self.orbitals}
+        # enumeration of (atom, orbital) levels: local-U orbitals plus
+        # V-only orbitals referenced by nonlocal pairs (reference:
+        # atom_type.cpp:1180-1208 adds them with use_for_calculation=false)
+        uc = ctx.unit_cell
+        self.levels = []            # (ia, HubbardOrbital)
+        self.level_use = []         # participates in the local U correction
+        for ia, (lab, _) in enumerate(uc.atoms):
+            if lab in self.desc_by_label:
+                self.levels.append((ia, self.desc_by_label[lab]))
+                self.level_use.append(True)
+        for p in self.nonlocal_pairs:
+            for (ia, n, l) in ((p.ia, p.n1, p.il), (p.ja, p.n2, p.jl)):
+                if not any(a == ia and o.n == n and o.l == l
+                           for a, o in self.levels):
+                    lab = uc.atoms[ia][0]
+                    self.levels.append((ia, HubbardOrbital(
+                        label=lab, n=n, l=l, U=0.0)))
+                    self.level_use.append(False)
+        self.offsets = []
+        off = 0
+        for ia, o in self.levels:
+            self.offsets.append(off)
+            off += 2 * o.l + 1
+        self.num_wf = off
+        self._lvl_key = {(ia, o.n, o.l): il for il, (ia, o) in enumerate(self.levels)}
+        # full-formula Coulomb matrices per orbital descriptor
+        self._vee = {}
+        if not self.simplified:
+            for o in self.orbitals:
+                self._vee[(o.label)] = torch.from_numpy(
+                    hubbard_matrix_full(o.l, o.U, o.J))
+        # unique translations for nonlocal (inter-site) occupation
+        self.T_list = sorted({p.T for p in self.nonlocal_pairs})
+        # occupation and potential matrices per level [mmax, mmax, nspin]
+        self.om = None
+        self.om_nl = None           # per nonlocal pair [2il+1, 2jl+1, nspin]
+        self.um = None
+        self.um_nl = None
+
+    def _find_level(self, ia, n, l):
+        il = self._lvl_key.get((ia, n, l))
+        if il is None:
+            raise RuntimeError(f"no hubbard level (atom {ia}, n={n}, l={l})")
+        return il
+
+    # -- per-k Hubbard wavefunctions --------------------------------------
+
+    def hubbard_wf_S(self, kp, hk) -> torch.Tensor:
+        """S|φ_hub⟩ rows [num_wf, nGk]; cached on the k-point."""
+        if getattr(kp, "_hub_wf_S", None) is not None:
+            return kp._hub_wf_S
+        ctx = self.ctx
+        from .dft import atomic_orbitals
+
+        phi_all = atomic_orbitals(ctx, kp)          # [n_ao_total, nGk]
+        sphi_all = self._apply_S(phi_all, hk)
+        if self.subspace_method == "full_orthogonalization":
+            ov = la.inner(phi_all, sphi_all)
+            ov = 0.5 * (ov + ov.conj().T)
+            w, v = la.eigh(ov)
+            B = (v / torch.sqrt(w.clamp(min=1e-12))) @ v.conj().T  # S^{-1/2}
+            phi_all = B.T @ phi_all
+            sphi_all = self._apply_S(phi_all, hk)
+        # extract the hubbard channels in level order
+        rows = []
+        uc = ctx.unit_cell
+        # atomic wf layout in atomic_orbitals: per atom, per wf, m-major
+        ao_offset = {}
+        off = 0
+        for ia, (lab, _) in enumerate(uc.atoms):
+            at = uc.atom_types[lab]
+            ao_offset[ia] = off
+            off += sum(2 * w.l + 1 for w in at.atomic_wfs)
+        for ia, o in self.levels:
+            lab = uc.atoms[ia][0]
+            at = uc.atom_types[lab]
+            woff = ao_offset[ia]
+            found = False
+            for w in at.atomic_wfs:
+                if w.l == o.l and (o.n < 0 or w.n < 0 or w.n == o.n):
+                    rows.extend(range(woff, woff + 2 * w.l + 1))
+                    found = True
+                    break
+                woff += 2 * w.l + 1
+            if not found:
+                raise RuntimeError(f"no atomic wf for hubbard orbital {o}")
+        idx = torch.tensor(rows, device=ctx.device)
+        kp._hub_wf_S = sphi_all[idx].contiguous()
+        return kp._hub_wf_S
+
+    def _apply_S(self, phi, hk):
+        if hk.Q is None or hk.bp.num_beta_total == 0:
+            return phi
+        bphi = hk.bp.inner(phi)
+        return phi + (hk.bp.beta @ (hk.Q @ bphi)).T
+
+    # -- occupation matrix -------------------------------------------------
+
+    def generate_occupation_matrix(self, kset, h0):
+        ctx = self.ctx
+        nsp = ctx.num_spins
+        om = [torch.zeros(2 * o.l + 1, 2 * o.l + 1, nsp, dtype=ctx.dtype,
+                          device=ctx.device) for _, o in self.levels]
+        occT = {T: torch.zeros(self.num_wf, self.num_wf, nsp, dtype=ctx.dtype,
+                               device=ctx.device) for T in self.T_list}
+        min_occ = ctx.cfg.iterative_solver.min_occupancy
+        for kp in kset:
+            hk = h0(kp)
+            swf = self.hubbard_wf_S(kp, hk)        # [nwf, nGk]
+            for ispn in range(nsp):
+                occ = torch.from_numpy(kp.occ[ispn]).to(ctx.device)
+                sel = torch.nonzero(occ.abs() > min_occ).reshape(-1)
+                if len(sel) == 0:
+                    continue
+                psi = kp.psi[ispn][sel]
+                f = (kp.weight / ctx.max_occupancy) * occ[sel]
+                proj = la.inner(swf, psi)           # ⟨Sφ_m|ψ_j⟩ [nwf, nocc]
+                full = torch.einsum("mj,j,nj->mn", proj, f.to(ctx.dtype),
+                                    proj.conj())
+                for il, (ia, o) in enumerate(self.levels):
+                    mm = 2 * o.l + 1
+                    off = self.offsets[il]
+                    om[il][..., ispn] += full[off:off + mm, off:off + mm]
+                for T in self.T_list:
+                    z = np.exp(-2j * math.pi * float(np.dot(kp.k_frac, T)))
+                    occT[T][..., ispn] += complex(z) * full
+        from .parallel import get_comm
+
+        comm = get_comm()
+        if comm.active:
+            for t in om:
+                comm.allreduce_(t)
+            for T in occT:
+                comm.allreduce_(occT[T])
+        if ctx.symmetry is not None:
+            om = self._symmetrize(om)
+        self.om = om
+        # inter-site blocks (update_nonlocal, occupation_matrix.cpp:421-453)
+        self.om_nl = []
+        for p in self.nonlocal_pairs:
+            a1 = self._find_level(p.ia, p.n1, p.il)
+            a2 = self._find_level(p.ja, p.n2, p.jl)
+            o1, o2 = self.offsets[a1], self.offsets[a2]
+            blk = occT[p.T][o1:o1 + 2 * p.il + 1, o2:o2 + 2 * p.jl + 1, :]
+            self.om_nl.append(blk.clone())
+        return om
+
+    def _symmetrize(self, om):
+        """Average over the space group (occupation_matrix symmetrization):
+        per-l real-SH rotation + atom permutation."""
+        from .symmetry import rlm_rotation_matrices
+
+        ctx = self.ctx
+        ops = ctx.symmetry.ops
+        lvl_of_atom = {ia: il for il, (ia, _) in enumerate(self.levels)}
+        out = [torch.zeros_like(t) for t in om]
+        for op in ops:
+            for il, (ia, o) in enumerate(self.levels):
+                ja = int(op.perm[ia])
+                jl = lvl_of_atom[ja]
+                D = rlm_rotation_matrices(o.l, op.S)[o.l]
+                T = torch.from_numpy(D).to(om[0].device).to(om[0].dtype)
+                for ispn in range(om[il].shape[-1]):
+                    out[il][..., ispn] += T.conj().T @ om[jl][..., ispn] @ T
+        for t in out:
+            t /= len(ops)
+        return out
+
+    def initial_occupation(self):
+        """Start from the configured total occupancy split over m and spins
+        (reference Occupation_matrix::init)."""
+        ctx = self.ctx
+        nsp = ctx.num_spins
+        om = []
+        for ia, o in self.levels:
+            mm = 2 * o.l + 1
+            t = torch.zeros(mm, mm, nsp, dtype=ctx.dtype, device=ctx.device)
+            occ = o.initial_occupancy
+            if nsp == 1:
+                for m in range(mm):
+                    t[m, m, 0] = min(1.0, occ / (2 * mm)) if occ else 0.0
+                t *= 0
+                if occ:
+                    t += torch.eye(mm, dtype=ctx.dtype,
+                                   device=ctx.device)[:, :, None] * (occ / 2.0 / mm)
+            else:
+                # polarized start: fill majority first (moment along z sign)
+                vz = ctx.unit_cell.vector_fields[ia][2]
+                up_first = vz >= 0
+                nup = min(mm, occ / 2 + abs(vz) / 2) if occ else 0
+                ndn = max(0.0, occ - nup) if occ else 0
+                for (s, nn) in ((0, nup), (1, ndn)) if up_first else ((1, nup), (0, ndn)):
+                    for m in range(mm):
+                        t[m, m, s] = min(1.0, max(0.0, nn / mm))
+            om.append(t)
+        self.om = om
+        return om
+
+    # -- potential ---------------------------------------------------------
+
+    def generate_potential(self, om=None):
+        """V from the occupation matrices: simplified (Dudarev) or full
+        (Liechtenstein; hubbard_potential_energy.cpp:126-167)."""
+        om = om if om is not None else self.om
+        ctx = self.ctx
+        nsp = ctx.num_spins
+        um = []
+        for il, (ia, o) in enumerate(self.levels):
+            mm = 2 * o.l + 1
+            t = torch.zeros_like(om[il])
+            if not self.level_use[il]:
+                um.append(t)
+                continue
+            if self.simplified:
+                u_eff = o.U - (o.J0 if abs(o.J0) > 1e-8 else 0.0)
+                for ispn in range(nsp):
+                    t[..., ispn] = -u_eff * om[il][..., ispn]
+                    t[..., ispn] += (o.alpha + 0.5 * u_eff) * torch.eye(
+                        mm, dtype=t.dtype, device=t.device)
+            else:
+                vee = self._vee[o.label].to(om[il].device)   # (m1,m3,m2,m4) layout
+                n_tot = sum(float(om[il][..., s].diagonal().real.sum())
+                            for s in range(nsp))
+                eye = torch.eye(mm, dtype=t.dtype, device=t.device)
+                n_sum = om[il].sum(-1) if nsp == 2 else 2.0 * om[il][..., 0]
+                if nsp == 1:
+                    n_tot *= 2.0
+                for ispn in range(nsp):
+                    n_s = float(om[il][..., ispn].diagonal().real.sum())
+                    t[..., ispn] += (o.J * n_s + 0.5 * (o.U - o.J)
+                                     - o.U * n_tot) * eye
+                    # Hartree: Σ V(m1,m3,m2,m4) n(m3,m4) over both spins
+                    t[..., ispn] += torch.einsum(
+                        "acbd,cd->ab", vee.to(t.dtype), n_sum)
+                    # exchange: − Σ V(m1,m3,m4,m2) n^σ(m3,m4)
+                    t[..., ispn] -= torch.einsum(
+                        "acdb,cd->ab", vee.to(t.dtype), om[il][..., ispn])
+            um.append(t)
+        self.um = um
+        # nonlocal: V_IJ = −V·n_IJ (generate_potential_collinear_nonlocal)
+        self.um_nl = [(-p.V) * self.om_nl[i] for i, p in enumerate(self.nonlocal_pairs)]             if self.om_nl is not None else None
+        return um
+
+    def u_matrix_full(self, ispn: int, k_frac=None) -> torch.Tensor:
+        """[num_wf, num_wf] U potential for one spin: block-diagonal local
+        parts + inter-site blocks × e^{+2πi k·T} (U_operator ctor,
+        non_local_operator.cpp:372-440)."""
+        ctx = self.ctx
+        out = torch.zeros(self.num_wf, self.num_wf, dtype=ctx.dtype,
+                          device=ctx.device)
+        for il, (ia, o) in enumerate(self.levels):
+            mm = 2 * o.l + 1
+            off = self.offsets[il]
+            out[off:off + mm, off:off + mm] = self.um[il][..., ispn]
+        if self.um_nl:
+            for i, p in enumerate(self.nonlocal_pairs):
+                a1 = self._find_level(p.ia, p.n1, p.il)
+                a2 = self._find_level(p.ja, p.n2, p.jl)
+                o1, o2 = self.offsets[a1], self.offsets[a2]
+                z = complex(np.exp(2j * math.pi * float(np.dot(k_frac, p.T))))                     if k_frac is not None else 1.0
+                out[o1:o1 + 2 * p.il + 1, o2:o2 + 2 * p.jl + 1] +=                     z * self.um_nl[i][..., ispn]
+        return out
+
+    def apply(self, kp, hk, psi, hpsi, ispn: int):
+        """hψ += Σ |Sφ⟩ V ⟨Sφ|ψ⟩ (apply_U_operator)."""
+        if self.um is None or self.num_wf == 0:
+            return
+        swf = self.hubbard_wf_S(kp, hk)
+        dm = la.inner(swf, psi)                  # [nwf, nb]
+        up = self.u_matrix_full(ispn, kp.k_frac) @ dm
+        hpsi += up.T @ swf
+
+    # -- energies ----------------------------------------------------------
+
+    def energy(self) -> float:
+        """E_U (calculate_energy_collinear_local + _nonlocal)."""
+        nsp = self.ctx.num_spins
+        e = 0.0
+        for il, (ia, o) in enumerate(self.levels):
+            if not self.level_use[il]:
+                continue
+            if self.simplified:
+                u_eff = o.U - (o.J0 if abs(o.J0) > 1e-8 else 0.0)
+                ea = 0.0
+                for ispn in range(nsp):
+                    n = self.om[il][..., ispn]
+                    ea += float(((o.alpha + 0.5 * u_eff) * torch.diagonal(n).sum()
+                                 - 0.5 * u_eff * (n @ n).diagonal().sum()).real)
+                if nsp == 1:
+                    ea *= 2.0
+                e += ea
+            else:
+                vee = self._vee[o.label].to(self.om[il].device)
+                n_tot = sum(float(self.om[il][..., s].diagonal().real.sum())
+                            for s in range(nsp))
+                n_ud = [float(self.om[il][..., s].diagonal().real.sum())
+                        for s in range(nsp)]
+                if nsp == 1:
+                    n_tot *= 2.0
+                    mag2 = 0.0
+                else:
+                    mag2 = (n_ud[0] - n_ud[1]) ** 2
+                e_dc = 0.5 * (o.U * n_tot * (n_tot - 1.0)
+                              - o.J * n_tot * (0.5 * n_tot - 1.0)
+                              - o.J * mag2 * 0.5)
+                e_u = 0.0
+                veet = vee.to(self.om[il].dtype)
+                for ispn in range(nsp):
+                    n_s = self.om[il][..., ispn]
+                    n_o = self.om[il][..., (ispn + 1) % 2] if nsp == 2                         else self.om[il][..., 0]
+                    # 0.5 Σ [(V(m1m2m3m4)−V(m1m2m4m3)) n^σ(m1,m3) n^σ(m2,m4)
+                    #        + V(m1m2m3m4) n^σ(m1,m3) n^σ'(m2,m4)]
+                    e_u += 0.5 * float((torch.einsum(
+                        "abcd,ac,bd->", veet - veet.permute(0, 1, 3, 2),
+                        n_s, n_s) + torch.einsum(
+                        "abcd,ac,bd->", veet, n_s, n_o)).real)
+                if nsp == 1:
+                    e_u *= 2.0
+                e += e_u - e_dc
+        for i, p in enumerate(self.nonlocal_pairs or []):
+            en = 0.0
+            for ispn in range(nsp):
+                n = self.om_nl[i][..., ispn]
+                en += float((n * n.conj()).sum().real) * p.V
+            if nsp == 1:
+                en *= 2.0
+            e += -0.5 * en
+        return e
+
+    def one_electron_energy(self) -> float:
+        """Re Σ n·conj(V) (×2 when num_spins == 1)."""
+        if self.om is None or self.um is None:
+            return 0.0
+        t = 0.0
+        for il in range(len(self.levels)):
+            t += float((self.om[il] * self.um[il].conj()).sum().real)
+        for i in range(len(self.nonlocal_pairs or [])):
+            t += float((self.om_nl[i] * self.um_nl[i].conj()).sum().real)
+        if self.ctx.num_spins == 1:
+            t *= 2.0
+        return t

@@ -128,6 +128,8 @@ class Potential:
 
         self.veff_r = self.vloc_r + vha_r + self.vxc_r
         self.veff_g = ctx.fft_fine.to_pw(self.veff_r.to(ctx.dtype))
+        if ctx.hubbard is not None:
+            ctx.hubbard.generate_potential()
         return self
 
     def energy_bxc(self, density) -> float:

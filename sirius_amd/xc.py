@@ -94,14 +94,15 @@ _PBE_BETA = 0.06672455060314922
 _PBE_GAMMA = (1.0 - math.log(2.0)) / math.pi**2
 
 
-def gga_x_pbe(rho: torch.Tensor, sigma: torch.Tensor):
-    """XC_GGA_X_PBE, unpolarized. Returns (eps, vrho, vsigma)."""
+def gga_x_pbe(rho: torch.Tensor, sigma: torch.Tensor, mu: float = _PBE_MU):
+    """XC_GGA_X_PBE (or PBEsol with mu=10/81), unpolarized.
+    Returns (eps, vrho, vsigma)."""
     rho = _safe_rho(rho)
     sigma = torch.clamp(sigma, min=1e-40)
     kf = (3.0 * math.pi**2 * rho) ** _THIRD
     # s^2 = sigma / (2 kf rho)^2
     s2 = sigma / (2.0 * kf * rho) ** 2
-    kappa, mu = _PBE_KAPPA, _PBE_MU
+    kappa = _PBE_KAPPA
     fdenom = 1.0 + mu * s2 / kappa
     fx = 1.0 + kappa - kappa / fdenom
     dfx_ds2 = mu / fdenom**2
@@ -120,13 +121,14 @@ def gga_x_pbe(rho: torch.Tensor, sigma: torch.Tensor):
     return eps, vrho, vsigma
 
 
-def gga_c_pbe(rho: torch.Tensor, sigma: torch.Tensor):
-    """XC_GGA_C_PBE, unpolarized. Returns (eps, vrho, vsigma)."""
+def gga_c_pbe(rho: torch.Tensor, sigma: torch.Tensor, beta: float = _PBE_BETA):
+    """XC_GGA_C_PBE (or PBEsol with beta=0.046), unpolarized.
+    Returns (eps, vrho, vsigma)."""
     rho = _safe_rho(rho)
     sigma = torch.clamp(sigma, min=1e-40)
     rs = (3.0 / (4.0 * math.pi * rho)) ** _THIRD
     ec, dec = _pw_g(rs, 0.0310907, 0.21370, 7.5957, 3.5876, 1.6382, 0.49294)
-    beta, gamma = _PBE_BETA, _PBE_GAMMA
+    gamma = _PBE_GAMMA
     kf = (3.0 * math.pi**2 * rho) ** _THIRD
     ks = torch.sqrt(4.0 * kf / math.pi)
     # t^2 = sigma / (2 ks rho)^2
@@ -253,19 +255,19 @@ def lda_c_pw_spin(ru: torch.Tensor, rd: torch.Tensor):
     return eps, vu, vd
 
 
-def gga_x_pbe_spin(ru, rd, s_uu, s_dd):
+def gga_x_pbe_spin(ru, rd, s_uu, s_dd, mu: float = _PBE_MU):
     """PBE exchange via exact spin scaling. Returns
     (eps, vu, vd, vs_uu, vs_dd); vsigma_ud = 0 for exchange."""
     ru = _safe_rho(ru)
     rd = _safe_rho(rd)
     n = ru + rd
-    e_u, v_u, vs_u = gga_x_pbe(2.0 * ru, 4.0 * s_uu)
-    e_d, v_d, vs_d = gga_x_pbe(2.0 * rd, 4.0 * s_dd)
+    e_u, v_u, vs_u = gga_x_pbe(2.0 * ru, 4.0 * s_uu, mu=mu)
+    e_d, v_d, vs_d = gga_x_pbe(2.0 * rd, 4.0 * s_dd, mu=mu)
     eps = (ru * e_u + rd * e_d) / n
     return eps, v_u, v_d, 2.0 * vs_u, 2.0 * vs_d
 
 
-def gga_c_pbe_spin(ru, rd, sigma):
+def gga_c_pbe_spin(ru, rd, sigma, beta_pbe: float = _PBE_BETA):
     """PBE correlation, spin-polarized; sigma = |∇(n↑+n↓)|².
     Returns (eps, vu, vd, vsigma) with vsigma = ∂(n·eps)/∂σ (same for all
     σ components: vsigma_uu = vsigma, vsigma_ud = 2·vsigma, vsigma_dd = vsigma
@@ -277,7 +279,7 @@ def gga_c_pbe_spin(ru, rd, sigma):
     z = torch.clamp((ru - rd) / n, -1.0 + 1e-12, 1.0 - 1e-12)
     rs = (3.0 / (4.0 * math.pi * n)) ** _THIRD
     phi = 0.5 * ((1 + z) ** (2.0 / 3.0) + (1 - z) ** (2.0 / 3.0))
-    beta, gamma = _PBE_BETA, _PBE_GAMMA
+    beta, gamma = beta_pbe, _PBE_GAMMA
 
     # need ec(rs, z) and its derivatives — reuse lda_c_pw_spin pieces
     ecU, dU = _pw_g(rs, 0.0310907, 0.21370, 7.5957, 3.5876, 1.6382, 0.49294)
@@ -336,8 +338,21 @@ def gga_c_pbe_spin(ru, rd, sigma):
     return eps, vu, vd, vsigma
 
 
+_PBESOL_MU = 10.0 / 81.0
+_PBESOL_BETA = 0.046
+
+
+def gga_x_pbesol(rho, sigma):
+    return gga_x_pbe(rho, sigma, mu=_PBESOL_MU)
+
+
+def gga_c_pbesol(rho, sigma):
+    return gga_c_pbe(rho, sigma, beta=_PBESOL_BETA)
+
+
 _LDA = {"XC_LDA_X": lda_x, "XC_LDA_C_PZ": lda_c_pz, "XC_LDA_C_PW": lda_c_pw}
-_GGA = {"XC_GGA_X_PBE": gga_x_pbe, "XC_GGA_C_PBE": gga_c_pbe}
+_GGA = {"XC_GGA_X_PBE": gga_x_pbe, "XC_GGA_C_PBE": gga_c_pbe,
+        "XC_GGA_X_PBE_SOL": gga_x_pbesol, "XC_GGA_C_PBE_SOL": gga_c_pbesol}
 _LDA_SPIN = {"XC_LDA_X": lda_x_spin, "XC_LDA_C_PZ": lda_c_pz_spin,
              "XC_LDA_C_PW": lda_c_pw_spin}
 
@@ -360,15 +375,17 @@ def evaluate_spin(names: list[str], ru: torch.Tensor, rd: torch.Tensor,
             eps = eps + e
             vu = vu + a
             vd = vd + b
-        elif name == "XC_GGA_X_PBE":
-            e, a, b, su, sd = gga_x_pbe_spin(ru, rd, s_uu, s_dd)
+        elif name in ("XC_GGA_X_PBE", "XC_GGA_X_PBE_SOL"):
+            mu = _PBE_MU if name == "XC_GGA_X_PBE" else _PBESOL_MU
+            e, a, b, su, sd = gga_x_pbe_spin(ru, rd, s_uu, s_dd, mu=mu)
             eps = eps + e
             vu = vu + a
             vd = vd + b
             vs_uu = vs_uu + su
             vs_dd = vs_dd + sd
-        elif name == "XC_GGA_C_PBE":
-            e, a, b, st = gga_c_pbe_spin(ru, rd, s_tot)
+        elif name in ("XC_GGA_C_PBE", "XC_GGA_C_PBE_SOL"):
+            bt = _PBE_BETA if name == "XC_GGA_C_PBE" else _PBESOL_BETA
+            e, a, b, st = gga_c_pbe_spin(ru, rd, s_tot, beta_pbe=bt)
             eps = eps + e
             vu = vu + a
             vd = vd + b

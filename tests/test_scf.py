@@ -54,8 +54,54 @@ def test06_fe_uspp_lda_fm():
     assert abs(res["magnetization"] - 6.760705375907565) < 1e-3
 
 
+full_suite = pytest.mark.skipif(
+    not os.environ.get("SIRIUS_AMD_FULL_TESTS"),
+    reason="heavy anchor; set SIRIUS_AMD_FULL_TESTS=1 to run")
+
+
 @requires_reference
 @pytest.mark.slow
+def test22_nio_uspp_hubbard_u_afm():
+    res, eref = run_case("test22", num_iter=45)
+    assert res["converged"]
+    assert abs(res["energy"]["total"] - eref) < 1e-5
+
+
+@requires_reference
+@full_suite
+def test26_nio_hubbard_full_ortho():
+    res, eref = run_case("test26", num_iter=50)
+    assert res["converged"]
+    assert abs(res["energy"]["total"] - eref) < 1e-5
+
+
+@requires_reference
+@full_suite
+def test27_licoo2_hubbard_u_plus_v():
+    res, eref = run_case("test27", num_iter=50)
+    assert res["converged"]
+    assert abs(res["energy"]["total"] - eref) < 1.1e-5
+
+
+@requires_reference
+@full_suite
+def test05_nio_uspp_lda_afm():
+    res, eref = run_case("test05", num_iter=45)
+    assert res["converged"]
+    assert abs(res["energy"]["total"] - eref) < 1e-5
+
+
+@requires_reference
+@pytest.mark.slow
+def test21_fesi_nc_pbe_fm():
+    res, eref = run_case("test21", num_iter=60)
+    assert res["converged"]
+    assert abs(res["energy"]["total"] - eref) < 1e-5
+    assert abs(res["magnetization"] - 1.2877521548061668) < 1e-3
+
+
+@requires_reference
+@full_suite
 def test01_srvo3_uspp_lda():
     res, eref = run_case("test01")
     assert res["converged"]
