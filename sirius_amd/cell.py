@@ -63,6 +63,11 @@ class AtomType:
         self.core_correction = False
         # PAW extras
         self.paw_core_energy = 0.0
+        self.paw_ae_wfs: list[np.ndarray] = []
+        self.paw_ps_wfs: list[np.ndarray] = []
+        self.paw_ae_core = np.zeros(0)
+        self.paw_wf_occ: list[float] = []
+        self.paw_cutoff_index = 0
 
     # -- parsing ----------------------------------------------------------
 
@@ -122,7 +127,23 @@ class AtomType:
 
         if at.is_paw:
             at.paw_core_energy = float(h.get("paw_core_energy", 0.0))
-            # full PAW data parsing added with the PAW feature
+            at.paw_cutoff_index = int(h.get("cutoff_radius_index", len(at.r)))
+            paw = pp.get("paw_data", {})
+            at.paw_ae_core = np.asarray(
+                paw.get("ae_core_charge_density", np.zeros_like(at.r)),
+                dtype=np.float64)
+            at.paw_wf_occ = [float(x) for x in paw.get("occupations", [])]
+            ncut = at.paw_cutoff_index
+            for w in paw.get("ae_wfc", []):
+                f = np.zeros_like(at.r)
+                v = np.asarray(w["radial_function"], dtype=np.float64)[:ncut]
+                f[:len(v)] = v
+                at.paw_ae_wfs.append(f)
+            for w in paw.get("ps_wfc", []):
+                f = np.zeros_like(at.r)
+                v = np.asarray(w["radial_function"], dtype=np.float64)[:ncut]
+                f[:len(v)] = v
+                at.paw_ps_wfs.append(f)
 
         return at
 
@@ -166,6 +187,45 @@ class AtomType:
             at.beta.append(BetaProjector(l=int(b.attrib["angular_momentum"]), j=None, f_r=f))
         if nbf:
             at.d_ion = vec(nl.find("PP_DIJ"), 0.5).reshape(nbf, nbf)  # Ry -> Ha
+        aug = nl.find("PP_AUGMENTATION")
+        if aug is not None:
+            for ch in aug:
+                if not ch.tag.startswith("PP_QIJL"):
+                    continue
+                # tag PP_QIJL.i.j.l (1-based i,j)
+                parts = ch.tag.split(".")
+                i, j, l = int(parts[1]) - 1, int(parts[2]) - 1, int(parts[3])
+                f = np.zeros_like(at.r)
+                v = vec(ch)
+                f[:min(len(v), len(f))] = v[:len(f)]
+                at.q_radial.append(QRadialFunction(i=i, j=j, l=l, f_r=f))
+
+        if at.is_paw:
+            at.paw_cutoff_index = int(h.get("cutoff_radius_index",
+                                            len(at.r)))
+            paw = root.find("PP_PAW")
+            if paw is not None:
+                at.paw_core_energy = 0.5 * float(paw.attrib.get("core_energy", 0.0))
+                nlcc_ae = paw.find("PP_AE_NLCC")
+                if nlcc_ae is not None:
+                    at.paw_ae_core = vec(nlcc_ae)
+                occ = paw.find("PP_OCCUPATIONS")
+                if occ is not None:
+                    at.paw_wf_occ = list(vec(occ))
+            full = root.find("PP_FULL_WFC")
+            ncut = at.paw_cutoff_index
+            if full is not None:
+                for i in range(nbf):
+                    for tag, dest in (("PP_AEWFC", at.paw_ae_wfs),
+                                      ("PP_PSWFC", at.paw_ps_wfs)):
+                        node = full.find(f"{tag}.{i + 1}")
+                        f = np.zeros_like(at.r)
+                        if node is not None:
+                            v = vec(node)[:ncut]
+                            f[:len(v)] = v
+                        dest.append(f)
+            if at.paw_ae_core.size == 0:
+                at.paw_ae_core = np.zeros_like(at.r)
 
         pswfc = root.find("PP_PSWFC")
         if pswfc is not None:

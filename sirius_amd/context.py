@@ -168,6 +168,13 @@ class SimulationContext:
 
         self.ri = RadialIntegralsCache(self)
 
+        # PAW on-site machinery (reference: src/potential/paw_potential.cpp)
+        self.paw = None
+        if any(at.is_paw for at in uc.atom_types.values()):
+            from .paw import PAWModule
+
+            self.paw = PAWModule(self)
+
         # DFT+U (reference: Hubbard class, src/hubbard/)
         self.hubbard = None
         if p.hubbard_correction:

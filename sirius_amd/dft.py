@@ -149,13 +149,14 @@ class DFTGroundState:
         d["vloc"] = self.potential.energy_vloc(self.density)
         d["ewald"] = self.potential.ewald
         d["entropy"] = self.kset.entropy_sum()
-        d["PAW_total_energy"] = 0.0
-        d["PAW_one_elec"] = 0.0
         d["scf_correction"] = self.scf_correction_energy
         d["fermi"] = self.kset.energy_fermi
         hub = self.ctx.hubbard
         d["hubbard_energy"] = hub.energy() if hub else 0.0
         d["hubbard_one_el"] = hub.one_electron_energy() if hub else 0.0
+        paw = self.ctx.paw
+        d["PAW_total_energy"] = paw.total_energy() if paw else 0.0
+        d["PAW_one_elec"] = paw.one_elec_energy(self.density) if paw else 0.0
         return d
 
     def total_energy(self) -> float:
@@ -173,6 +174,7 @@ class DFTGroundState:
         if self.ctx.hubbard is not None:
             self.ctx.hubbard.initial_occupation()
         self.potential.generate(self.density)
+        self.potential.generate_paw(self.density)
         h0 = Hamiltonian0(self.ctx, self.potential)
         for kp in self.kset:
             initialize_subspace(self.ctx, kp, h0(kp))
@@ -213,11 +215,22 @@ class DFTGroundState:
             itsol_tol = max(itso.min_tolerance, tol)
             itsol_converged = tol <= itso.min_tolerance
 
+            paw = ctx.paw
+            dm_fresh = {k: v.clone() for k, v in
+                        self.density.density_matrix.items()}                 if (paw and self.density.density_matrix) else None
+            if paw:
+                class _D:
+                    density_matrix = dm_fresh
+                e1 += paw.one_elec_energy(_D)
+
             self.potential.generate(self.density)
+            self.potential.generate_paw(self.density)
 
             e2 = self.ctx.integrate_rg_fine(rho1_r * self.potential.veff_r)
             if mag1_r is not None and self.potential.bz_r is not None:
                 e2 += self.ctx.integrate_rg_fine(mag1_r * self.potential.bz_r)
+            if paw:
+                e2 += paw.one_elec_energy(_D)
             self.scf_correction_energy = e2 - e1
 
             etot = self.total_energy()

@@ -156,19 +156,25 @@ class Hamiltonian0:
                 dt0 = dints(potential.veff_g)               # [nqlm, na]
                 dt1 = dints(potential.bz_g) if nsp == 2 else None
                 nbf = aug.nbf
-                iu, il = np.triu_indices(nbf)   # xi1 <= xi2 pairs (row=xi1)
+                paw = ctx.paw
                 for i, ia in enumerate(ia_list):
+                    # unpack packed Q·V integrals per component
+                    comps = [np.zeros((nbf, nbf)) for _ in range(nsp if nsp == 2 else 1)]
+                    d0 = np.zeros((nbf, nbf))
+                    d1 = np.zeros((nbf, nbf)) if nsp == 2 else None
+                    for xi2 in range(nbf):
+                        for xi1 in range(xi2 + 1):
+                            idx12 = xi2 * (xi2 + 1) // 2 + xi1
+                            d0[xi1, xi2] = d0[xi2, xi1] = dt0[idx12, i]
+                            if nsp == 2:
+                                d1[xi1, xi2] = d1[xi2, xi1] = dt1[idx12, i]
+                    if paw is not None and ia in paw.dij:
+                        d0 = d0 + paw.dij[ia][..., 0]
+                        if nsp == 2:
+                            d1 = d1 + paw.dij[ia][..., 1]
                     for ispn in range(nsp):
-                        dt = dt0[:, i] if nsp == 1 else (
-                            dt0[:, i] + dt1[:, i] if ispn == 0 else dt0[:, i] - dt1[:, i])
                         d = np.array(ions)
-                        k = 0
-                        for xi2 in range(nbf):
-                            for xi1 in range(xi2 + 1):
-                                idx12 = xi2 * (xi2 + 1) // 2 + xi1
-                                d[xi1, xi2] += dt[idx12]
-                                if xi1 != xi2:
-                                    d[xi2, xi1] += dt[idx12]
+                        d += d0 if nsp == 1 else (d0 + d1 if ispn == 0 else d0 - d1)
                         self.d_atom[ispn][ia] = d
             else:
                 for ia in ia_list:

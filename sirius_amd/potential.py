@@ -132,6 +132,11 @@ class Potential:
             ctx.hubbard.generate_potential()
         return self
 
+    def generate_paw(self, density):
+        """PAW on-site potentials + Dij (generate_PAW_effective_potential)."""
+        if self.ctx.paw is not None:
+            self.ctx.paw.generate(density)
+
     def energy_bxc(self, density) -> float:
         """∫ m_z·B_z (reference energy.cpp:85-93)."""
         if self.bz_r is None or density.mag_r is None:
