@@ -182,7 +182,13 @@ class AtomType:
             b = nl.find(f"PP_BETA.{i + 1}")
             f = np.zeros_like(at.r)
             fr = vec(b)
-            n = min(len(fr), len(at.r))
+            # truncate at the projector's own cutoff (atom_type.cpp:45-56:
+            # attribute, else the last |v| > 1e-80)
+            nr = int(float(b.attrib.get("cutoff_radius_index", 0)))
+            if nr == 0:
+                nz = np.nonzero(np.abs(fr) > 1e-80)[0]
+                nr = int(nz[-1]) + 1 if len(nz) else len(fr)
+            n = min(nr, len(at.r), len(fr))
             f[:n] = fr[:n]
             at.beta.append(BetaProjector(l=int(b.attrib["angular_momentum"]), j=None, f_r=f))
         if nbf:
@@ -201,8 +207,9 @@ class AtomType:
                 at.q_radial.append(QRadialFunction(i=i, j=j, l=l, f_r=f))
 
         if at.is_paw:
-            at.paw_cutoff_index = int(h.get("cutoff_radius_index",
-                                            len(at.r)))
+            aug_node = nl.find("PP_AUGMENTATION")
+            at.paw_cutoff_index = int(float(aug_node.attrib.get(
+                "cutoff_r_index", len(at.r)))) if aug_node is not None else len(at.r)
             paw = root.find("PP_PAW")
             if paw is not None:
                 at.paw_core_energy = 0.5 * float(paw.attrib.get("core_energy", 0.0))
