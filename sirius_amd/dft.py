@@ -26,6 +26,7 @@ from .hamiltonian import Hamiltonian0
 from .kpoint import KPointSet
 from .parallel import get_comm
 from .potential import Potential
+from .utils.profiler import profiler
 
 
 def atomic_orbitals(ctx, kp) -> torch.Tensor:
@@ -198,16 +199,22 @@ class DFTGroundState:
         t0 = time.time()
 
         for it in range(num_dft_iter):
-            h0 = Hamiltonian0(ctx, self.potential, self.density)
-            bands_converged = diagonalize(ctx, h0, self.kset, itsol_tol)
-            self.kset.find_band_occupancies()
-            self.density.generate(self.kset, h0)
+            with profiler("scf_iteration"):
+                with profiler("Hamiltonian0"):
+                    h0 = Hamiltonian0(ctx, self.potential, self.density)
+                with profiler("diagonalize"):
+                    bands_converged = diagonalize(ctx, h0, self.kset, itsol_tol)
+                with profiler("occupancies"):
+                    self.kset.find_band_occupancies()
+                with profiler("density"):
+                    self.density.generate(self.kset, h0)
 
             e1 = self.energy_potential(self.density.rho_r, self.density.mag_r)
             rho1_r = self.density.rho_r.clone()
             mag1_r = self.density.mag_r.clone() if self.density.mag_r is not None else None
 
-            rms = self.density.mix()
+            with profiler("mix"):
+                rms = self.density.mix()
 
             tol = rms
             tol = min(itso.tolerance_scale[0] * tol,
@@ -223,8 +230,9 @@ class DFTGroundState:
                     density_matrix = dm_fresh
                 e1 += paw.one_elec_energy(_D)
 
-            self.potential.generate(self.density)
-            self.potential.generate_paw(self.density)
+            with profiler("potential"):
+                self.potential.generate(self.density)
+                self.potential.generate_paw(self.density)
 
             e2 = self.ctx.integrate_rg_fine(rho1_r * self.potential.veff_r)
             if mag1_r is not None and self.potential.bz_r is not None:
@@ -247,6 +255,7 @@ class DFTGroundState:
             eold = etot
 
         out = {
+            "timers": profiler.to_dict(),
             "converged": num_iter >= 0,
             "num_scf_iterations": num_iter,
             "energy": self.total_energy_components() | {

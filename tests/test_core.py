@@ -143,3 +143,39 @@ def test_mixer_linear_convergence():
             mx.set_input({"x": step(out)})
             rms = mx.mix()
         assert torch.allclose(mx.get_output()["x"], target, atol=1e-6), cls
+
+
+def test_checkpoint_roundtrip(tmp_path):
+    """Save/load preserves the density and G-vector remap works
+    (reference tree: simulation_context.cpp:1153-1191)."""
+    from sirius_amd.models.synthetic import make_context
+    from sirius_amd.kpoint import KPointSet
+    from sirius_amd.dft import DFTGroundState
+    from sirius_amd import checkpoint
+    import torch
+
+    ctx = make_context(natoms=2, gk_cutoff=3.5, pw_cutoff=8.0, device="cpu")
+    kset = KPointSet(ctx)
+    dft = DFTGroundState(kset).initial_state()
+    dft.find(num_dft_iter=3)
+    rho_before = dft.density.rho_g.clone()
+    p = str(tmp_path / "sirius.npz")
+    checkpoint.save_state(p, dft)
+    dft.density.rho_g = torch.zeros_like(dft.density.rho_g)
+    checkpoint.load_state(p, dft)
+    assert torch.allclose(dft.density.rho_g, rho_before, atol=1e-14)
+
+
+def test_profiler_tree():
+    from sirius_amd.utils.profiler import Profiler
+
+    p = Profiler()
+    with p("outer"):
+        with p("inner"):
+            pass
+        with p("inner"):
+            pass
+    d = p.to_dict()
+    assert d["outer"]["count"] == 1
+    assert d["outer"]["sub"]["inner"]["count"] == 2
+    assert "outer" in p.report()
