@@ -126,6 +126,9 @@ class SimulationContext:
         self.num_mag_dims = int(p.num_mag_dims)
         self.num_spins = 2 if self.num_mag_dims > 0 else 1
         self.num_spinors = 2 if self.num_mag_dims == 3 else 1
+        self.nc_magnetism = self.num_mag_dims == 3
+        # diagonalization channels: one spinor problem for nc, else per spin
+        self.num_spin_steps = 1 if self.nc_magnetism else self.num_spins
         self.xc_names = list(p.xc_functionals)
         self.is_gga = xc_mod.is_gga(self.xc_names)
 
@@ -152,8 +155,10 @@ class SimulationContext:
         self.max_occupancy = 2.0 if self.num_mag_dims == 0 else 1.0
 
         # crystal symmetry (space group + IBZ; reference: Crystal_symmetry)
+        # (noncollinear case needs spin-rotation symmetrization — run the
+        # full k-mesh instead)
         self.symmetry = None
-        if p.use_symmetry:
+        if p.use_symmetry and not self.nc_magnetism:
             from .symmetry import CrystalSymmetry
 
             self.symmetry = CrystalSymmetry(self.unit_cell)
@@ -185,6 +190,17 @@ class SimulationContext:
                 # inter-site occupation symmetrization is not implemented;
                 # run the full k-mesh instead (physically equivalent)
                 self.symmetry = None
+
+    def so_fcoef(self, lab: str):
+        """Cached spin-orbit f-coefficients per type (generate_f_coefficients)."""
+        if not hasattr(self, "_so_fcoef"):
+            self._so_fcoef = {}
+        if lab not in self._so_fcoef:
+            from . import so as so_mod
+
+            self._so_fcoef[lab] = so_mod.f_coefficients(
+                self.unit_cell.atom_types[lab])
+        return self._so_fcoef[lab]
 
     # -- augmentation (USPP/PAW) ------------------------------------------
 

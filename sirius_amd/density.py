@@ -30,6 +30,9 @@ class Density:
         # collinear magnetization m_z (num_mag_dims == 1)
         self.mag_g = torch.zeros_like(self.rho_g) if ctx.num_mag_dims else None
         self.mag_r = torch.zeros_like(self.rho_r) if ctx.num_mag_dims else None
+        # noncollinear vector magnetization [mx, my, mz]
+        self.magv_g = [torch.zeros_like(self.rho_g) for _ in range(3)]             if ctx.nc_magnetism else None
+        self.magv_r = [torch.zeros_like(self.rho_r) for _ in range(3)]             if ctx.nc_magnetism else None
         self.rho_core_r = torch.zeros_like(self.rho_r)
         self.rho_core_g = None
         self.density_matrix = None
@@ -83,6 +86,7 @@ class Density:
             # starting moment (density.cpp:246-272, the default
             # !smooth_initial_mag branch; R from auto MT radii)
             mz = np.zeros(ctx.fft_fine.dims)
+            mxy = [np.zeros(ctx.fft_fine.dims), np.zeros(ctx.fft_fine.dims)]
             dims = ctx.fft_fine.dims
             frac_grid = np.stack(np.meshgrid(
                 np.arange(dims[0]) / dims[0], np.arange(dims[1]) / dims[1],
@@ -105,9 +109,19 @@ class Density:
                 s = w.sum() * uc.omega / np.prod(dims)
                 if s > 1e-12:
                     mz += v[2] * w / s
+                    if ctx.nc_magnetism:
+                        mxy[0] += v[0] * w / s
+                        mxy[1] += v[1] * w / s
             self.mag_r = torch.from_numpy(mz).to(device=ctx.device,
                                                  dtype=ctx.rdtype)
             self.mag_g = ctx.fft_fine.to_pw(self.mag_r.to(ctx.dtype))
+            if ctx.nc_magnetism:
+                self.magv_r = [
+                    torch.from_numpy(mxy[0]).to(device=ctx.device, dtype=ctx.rdtype),
+                    torch.from_numpy(mxy[1]).to(device=ctx.device, dtype=ctx.rdtype),
+                    self.mag_r]
+                self.magv_g = [ctx.fft_fine.to_pw(m.to(ctx.dtype))
+                               for m in self.magv_r]
         return self
 
     def init_density_matrix_for_paw(self):
@@ -153,18 +167,39 @@ class Density:
         comm = get_comm()
         dims = ctx.coarse_dims
         nsp = ctx.num_spins
+        ncomp = 4 if ctx.nc_magnetism else nsp
         rho_c = [torch.zeros(*dims, dtype=ctx.rdtype, device=ctx.device)
-                 for _ in range(nsp)]
+                 for _ in range(ncomp)]
+        mag_c = None
         min_occ = ctx.cfg.iterative_solver.min_occupancy
 
         for kp in kset:
-            for ispn in range(nsp):
-                occ = torch.from_numpy(kp.occ[ispn]).to(ctx.device)
+            if ctx.nc_magnetism:
+                ng = kp.num_gkvec
+                occ = torch.from_numpy(kp.occ[0]).to(ctx.device)
                 sel = torch.nonzero(occ.abs() > min_occ).reshape(-1)
                 if len(sel) == 0:
                     continue
                 w = ((kp.weight / ctx.unit_cell.omega) * occ[sel]).to(ctx.rdtype)
-                kp.fft.density_accumulate(kp.psi[ispn][sel], w, rho_c[ispn])
+                up_r = kp.fft.to_real(kp.psi[0][sel, :ng])
+                dn_r = kp.fft.to_real(kp.psi[0][sel, ng:])
+                # density channels [up², dn², 2Re(ψ↑ψ↓*), −2Im(ψ↑ψ↓*)]
+                # (add_k_point_contribution_rg, density.cpp:66-80)
+                rho_c[0] += torch.einsum("b,bxyz->xyz", w,
+                                         up_r.real**2 + up_r.imag**2)
+                rho_c[1] += torch.einsum("b,bxyz->xyz", w,
+                                         dn_r.real**2 + dn_r.imag**2)
+                z = up_r * dn_r.conj()
+                rho_c[2] += 2.0 * torch.einsum("b,bxyz->xyz", w, z.real)
+                rho_c[3] -= 2.0 * torch.einsum("b,bxyz->xyz", w, z.imag)
+            else:
+                for ispn in range(nsp):
+                    occ = torch.from_numpy(kp.occ[ispn]).to(ctx.device)
+                    sel = torch.nonzero(occ.abs() > min_occ).reshape(-1)
+                    if len(sel) == 0:
+                        continue
+                    w = ((kp.weight / ctx.unit_cell.omega) * occ[sel]).to(ctx.rdtype)
+                    kp.fft.density_accumulate(kp.psi[ispn][sel], w, rho_c[ispn])
 
         for t in rho_c:
             if comm.active:
@@ -178,7 +213,12 @@ class Density:
             out[ctx.coarse_to_fine] = cg
             return out
 
-        if nsp == 1:
+        if ctx.nc_magnetism:
+            self.rho_g = to_fine(rho_c[0] + rho_c[1])
+            self.magv_g = [to_fine(rho_c[2]), to_fine(rho_c[3]),
+                           to_fine(rho_c[0] - rho_c[1])]
+            self.mag_g = self.magv_g[2]
+        elif nsp == 1:
             self.rho_g = to_fine(rho_c[0])
         else:
             self.rho_g = to_fine(rho_c[0] + rho_c[1])
@@ -193,7 +233,12 @@ class Density:
             self.density_matrix = dm
             aug = self.generate_rho_aug(dm)
             self.rho_g = self.rho_g + aug[0]
-            if nsp == 2:
+            if ctx.nc_magnetism:
+                self.magv_g[2] = self.magv_g[2] + aug[1]
+                self.magv_g[0] = self.magv_g[0] + aug[2]
+                self.magv_g[1] = self.magv_g[1] + aug[3]
+                self.mag_g = self.magv_g[2]
+            elif nsp == 2:
                 self.mag_g = self.mag_g + aug[1]
         if ctx.hubbard is not None:
             ctx.hubbard.generate_occupation_matrix(kset, hamiltonian0)
@@ -206,7 +251,10 @@ class Density:
                 self.mag_g = symmetrize_rho_g(self.mag_g, ctx.gvec_fine,
                                               ctx.symmetry.ops)
         self.rho_r = ctx.fft_fine.to_real(self.rho_g).real
-        if nsp == 2:
+        if ctx.nc_magnetism:
+            self.magv_r = [ctx.fft_fine.to_real(m).real for m in self.magv_g]
+            self.mag_r = self.magv_r[2]
+        elif nsp == 2:
             self.mag_r = ctx.fft_fine.to_real(self.mag_g).real
         return self
 
@@ -218,17 +266,45 @@ class Density:
         uc = ctx.unit_cell
         from .hamiltonian import BetaProjectors
 
+        ndmc = 3 if ctx.nc_magnetism else ctx.num_spins
         dm = {}
         for lab, at in uc.atom_types.items():
             na = len(uc.atoms_of_type(lab))
             nbf = at.num_beta_lm
-            dm[lab] = torch.zeros(na, nbf, nbf, ctx.num_spins,
+            dm[lab] = torch.zeros(na, nbf, nbf, ndmc,
                                   dtype=ctx.dtype, device=ctx.device)
         min_occ = ctx.cfg.iterative_solver.min_occupancy
+
+        def type_rows(bp, lab, nbf):
+            ia_list = uc.atoms_of_type(lab)
+            offs = [bp.atom_offsets[ia] for ia in ia_list]
+            return ia_list, torch.tensor(
+                [o + x for o in offs for x in range(nbf)], device=ctx.device)
+
         for kp in kset:
             if kp.beta is None:
                 kp.beta = BetaProjectors(ctx, kp)
             bp = kp.beta
+            if ctx.nc_magnetism:
+                ng = kp.num_gkvec
+                occ = torch.from_numpy(kp.occ[0]).to(ctx.device)
+                sel = torch.nonzero(occ.abs() > min_occ).reshape(-1)
+                if len(sel) == 0:
+                    continue
+                bu = bp.inner(kp.psi[0][sel, :ng])
+                bd = bp.inner(kp.psi[0][sel, ng:])
+                wf = (kp.weight * occ[sel]).to(ctx.dtype)
+                for lab, at in uc.atom_types.items():
+                    nbf = at.num_beta_lm
+                    if nbf == 0:
+                        continue
+                    ia_list, rows = type_rows(bp, lab, nbf)
+                    xu = bu[rows].reshape(len(ia_list), nbf, -1)
+                    xd = bd[rows].reshape(len(ia_list), nbf, -1)
+                    dm[lab][..., 0] += torch.einsum("aib,b,ajb->aij", xu, wf, xu.conj())
+                    dm[lab][..., 1] += torch.einsum("aib,b,ajb->aij", xd, wf, xd.conj())
+                    dm[lab][..., 2] += torch.einsum("aib,b,ajb->aij", xu, wf, xd.conj())
+                continue
             for ispn in range(ctx.num_spins):
                 occ = torch.from_numpy(kp.occ[ispn]).to(ctx.device)
                 sel = torch.nonzero(occ.abs() > min_occ).reshape(-1)
@@ -240,11 +316,7 @@ class Density:
                     nbf = at.num_beta_lm
                     if nbf == 0:
                         continue
-                    ia_list = uc.atoms_of_type(lab)
-                    offs = [bp.atom_offsets[ia] for ia in ia_list]
-                    rows = torch.tensor(
-                        [o + x for o in offs for x in range(nbf)],
-                        device=ctx.device)
+                    ia_list, rows = type_rows(bp, lab, nbf)
                     x = bpsi[rows].reshape(len(ia_list), nbf, -1)  # [na,nbf,nocc]
                     dm[lab][..., ispn] += torch.einsum(
                         "aib,b,ajb->aij", x, wf, x.conj())
@@ -260,7 +332,7 @@ class Density:
         generate_dm_pw / sum_q_pw_dm_pw kernels)."""
         ctx = self.ctx
         uc = ctx.unit_cell
-        nch = 2 if ctx.num_spins == 2 else 1
+        nch = 4 if ctx.nc_magnetism else (2 if ctx.num_spins == 2 else 1)
         out = [torch.zeros(ctx.gvec_fine.num_gvec, dtype=ctx.dtype,
                            device=ctx.device) for _ in range(nch)]
         for lab, at in uc.atom_types.items():
@@ -276,10 +348,14 @@ class Density:
             d = dm[lab]                                    # [na, nbf, nbf, nspin]
             dpk = d[:, xi2_idx, xi1_idx, :]                # [na, nqlm, nspin]
             # channels (density_matrix_aux, density.cpp:1782-1810):
-            # ch0 = Re Σ_σ dm, ch1 = Re(dm↑ − dm↓)
-            chans = [dpk.sum(-1).real.T]
-            if nch == 2:
+            # ch0 = Re(d0+d1), ch1 = Re(d0−d1), nc: ch2 = 2Re d2, ch3 = −2Im d2
+            chans = [(dpk[..., 0] + dpk[..., 1]).real.T] \
+                if dpk.shape[-1] > 1 else [dpk[..., 0].real.T]
+            if nch >= 2:
                 chans.append((dpk[..., 0] - dpk[..., 1]).real.T)
+            if nch == 4:
+                chans.append(2.0 * dpk[..., 2].real.T)
+                chans.append(-2.0 * dpk[..., 2].imag.T)
             phases = ctx.phase_pos(lab).conj()             # e^{-iGτ} [na, nG]
             for ic, dm_aux in enumerate(chans):
                 dm_pw = dm_aux.to(ctx.dtype) @ phases      # [nqlm, nG]
@@ -305,7 +381,12 @@ class Density:
 
         comps = [Component("rho_g", inner=inner_pw, global_size=omega)]
         init = {"rho_g": self.rho_g}
-        if self.ctx.num_spins == 2:
+        if self.ctx.nc_magnetism:
+            for i in range(3):
+                comps.append(Component(f"magv{i}_g", inner=inner_pw,
+                                       global_size=omega))
+                init[f"magv{i}_g"] = self.magv_g[i]
+        elif self.ctx.num_spins == 2:
             comps.append(Component("mag_g", inner=inner_pw, global_size=omega))
             init["mag_g"] = self.mag_g
         if self.density_matrix is not None:
@@ -352,7 +433,10 @@ class Density:
 
     def mix(self) -> float:
         inp = {"rho_g": self.rho_g}
-        if self.ctx.num_spins == 2:
+        if self.ctx.nc_magnetism:
+            for i in range(3):
+                inp[f"magv{i}_g"] = self.magv_g[i]
+        elif self.ctx.num_spins == 2:
             inp["mag_g"] = self.mag_g
         if self.density_matrix is not None and "dm" in self.mixer.components:
             inp["dm"] = torch.cat([t.reshape(-1)
@@ -364,7 +448,12 @@ class Density:
         out = self.mixer.get_output()
         self.rho_g = out["rho_g"]
         self.rho_r = self.ctx.fft_fine.to_real(self.rho_g).real
-        if self.ctx.num_spins == 2:
+        if self.ctx.nc_magnetism:
+            self.magv_g = [out[f"magv{i}_g"] for i in range(3)]
+            self.magv_r = [self.ctx.fft_fine.to_real(m).real for m in self.magv_g]
+            self.mag_g = self.magv_g[2]
+            self.mag_r = self.magv_r[2]
+        elif self.ctx.num_spins == 2:
             self.mag_g = out["mag_g"]
             self.mag_r = self.ctx.fft_fine.to_real(self.mag_g).real
         if self.density_matrix is not None and "dm" in out:

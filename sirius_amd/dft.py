@@ -67,16 +67,23 @@ def initialize_subspace(ctx, kp, hk):
     init = ctx.cfg.iterative_solver.init_subspace
     phi = atomic_orbitals(ctx, kp) if init == "lcao" else \
         torch.zeros(0, kp.num_gkvec, dtype=ctx.dtype, device=ctx.device)
-    n_ao = phi.shape[0]
+    n_ao = phi.shape[0] * ctx.num_spinors
     if n_ao < nb:
         gen = torch.Generator(device="cpu").manual_seed(12345 + kp.gkvec.num_gvec)
-        rnd = torch.randn(nb - n_ao, kp.num_gkvec, 2, generator=gen,
-                          dtype=torch.float64)
+        rnd = torch.randn((nb - n_ao + 1) // ctx.num_spinors, kp.num_gkvec, 2,
+                          generator=gen, dtype=torch.float64)
         rnd = torch.view_as_complex(rnd.contiguous()).to(ctx.device)
         # damp high-G components for smoother start
         damp = 1.0 / (1.0 + kp.gkvec.gk2_t)
         rnd = rnd * damp
         phi = torch.cat([phi, rnd], dim=0)
+
+    if ctx.nc_magnetism:
+        # spinor trial: each scalar orbital as (φ,0) and (0,φ)
+        n0 = phi.shape[0]
+        z = torch.zeros_like(phi)
+        phi = torch.cat([torch.cat([phi, z], dim=1),
+                         torch.cat([z, phi], dim=1)], dim=0)
 
     from .davidson import _ortho_joint
     hphi, sphi = hk.apply_h_s(phi)
@@ -88,7 +95,7 @@ def initialize_subspace(ctx, kp, hk):
     psi = Z[:, :nsel].T @ phi
     if nsel < nb:
         raise RuntimeError("not enough trial orbitals for requested bands")
-    for ispn in range(ctx.num_spins):
+    for ispn in range(ctx.num_spin_steps):
         kp.psi[ispn] = psi
         kp.eigvals[ispn] = evals[:nb].real.cpu().numpy()
 
@@ -101,7 +108,7 @@ def diagonalize(ctx, h0: Hamiltonian0, kset: KPointSet, itsol_tol: float) -> boo
     for kp in kset:
         hk = h0(kp)
         o_diag = hk.o_diag()
-        for ispn in range(ctx.num_spins):
+        for ispn in range(ctx.num_spin_steps):
             h_diag = hk.h_diag(ispn)
             res = davidson(
                 lambda p, s=ispn: hk.apply_h_s(p, s),
@@ -132,7 +139,11 @@ class DFTGroundState:
         """∫ρ V_eff + ∫m·B (+ PAW/hubbard terms when present;
         reference energy.cpp:251-257)."""
         e = self.ctx.integrate_rg_fine(rho_r * self.potential.veff_r)
-        if mag_r is not None and self.potential.bz_r is not None:
+        if self.ctx.nc_magnetism and self.potential.bvec_r is not None:
+            for i in range(3):
+                e += self.ctx.integrate_rg_fine(
+                    self.density.magv_r[i] * self.potential.bvec_r[i])
+        elif mag_r is not None and self.potential.bz_r is not None:
             e += self.ctx.integrate_rg_fine(mag_r * self.potential.bz_r)
         # NOTE: the reference's energy_potential also adds hubbard_energy
         # (energy.cpp:251-257), but both SCF-correction evaluations use the

@@ -133,15 +133,24 @@ class Hamiltonian0:
             return ctx.fft_coarse.to_real(fg[ctx.coarse_to_fine]).real
 
         veff_c = to_coarse(potential.veff_g)
+        self.v_ud_coarse = None
         if nsp == 1:
             self.veff_r_coarse = [veff_c]
+        elif ctx.nc_magnetism:
+            bz_c = to_coarse(potential.bvec_g[2])
+            bx_c = to_coarse(potential.bvec_g[0])
+            by_c = to_coarse(potential.bvec_g[1])
+            self.veff_r_coarse = [veff_c + bz_c, veff_c - bz_c]
+            # off-diagonal spin block (Bx − iBy); conj is the dn-up block
+            self.v_ud_coarse = (bx_c - 1j * by_c).to(ctx.dtype)
         else:
             bz_c = to_coarse(potential.bz_g)
             # spin 0 = up: V + Bz; spin 1 = dn: V − Bz (non_local_operator.cpp:236-238)
             self.veff_r_coarse = [veff_c + bz_c, veff_c - bz_c]
         uc = ctx.unit_cell
-        # per-atom, per-spin D matrices [nbf, nbf] (numpy, real)
-        self.d_atom = [[None] * uc.num_atoms for _ in range(nsp)]
+        # per-atom, per-spin(-block) D matrices [nbf, nbf]
+        nblk = 4 if ctx.nc_magnetism else nsp
+        self.d_atom = [[None] * uc.num_atoms for _ in range(nblk)]
         for lab, at in uc.atom_types.items():
             ions = d_matrix_lm(at)
             ia_list = uc.atoms_of_type(lab)
@@ -154,32 +163,69 @@ class Hamiltonian0:
                     return ((aug.q_pw.conj() @ va.T).real * uc.omega).cpu().numpy()
 
                 dt0 = dints(potential.veff_g)               # [nqlm, na]
-                dt1 = dints(potential.bz_g) if nsp == 2 else None
+                if ctx.nc_magnetism:
+                    dtz = dints(potential.bvec_g[2])
+                    dtx = dints(potential.bvec_g[0])
+                    dty = dints(potential.bvec_g[1])
+                elif nsp == 2:
+                    dt1 = dints(potential.bz_g)
+                else:
+                    dt1 = None
                 nbf = aug.nbf
                 paw = ctx.paw
                 for i, ia in enumerate(ia_list):
-                    # unpack packed Q·V integrals per component
-                    comps = [np.zeros((nbf, nbf)) for _ in range(nsp if nsp == 2 else 1)]
-                    d0 = np.zeros((nbf, nbf))
-                    d1 = np.zeros((nbf, nbf)) if nsp == 2 else None
-                    for xi2 in range(nbf):
-                        for xi1 in range(xi2 + 1):
-                            idx12 = xi2 * (xi2 + 1) // 2 + xi1
-                            d0[xi1, xi2] = d0[xi2, xi1] = dt0[idx12, i]
-                            if nsp == 2:
-                                d1[xi1, xi2] = d1[xi2, xi1] = dt1[idx12, i]
+                    def unpack(dt):
+                        m = np.zeros((nbf, nbf))
+                        for xi2 in range(nbf):
+                            for xi1 in range(xi2 + 1):
+                                idx12 = xi2 * (xi2 + 1) // 2 + xi1
+                                m[xi1, xi2] = m[xi2, xi1] = dt[idx12, i]
+                        return m
+
+                    d0 = unpack(dt0)
                     if paw is not None and ia in paw.dij:
                         d0 = d0 + paw.dij[ia][..., 0]
-                        if nsp == 2:
+                    if ctx.nc_magnetism:
+                        dz = unpack(dtz)
+                        dx = unpack(dtx)
+                        dy = unpack(dty)
+                        if at.spin_orbit:
+                            from . import so as so_mod
+
+                            fc = ctx.so_fcoef(lab)
+                            blocks = so_mod.so_d_blocks(at, [d0, dz, dx, dy], fc)
+                            for ib in range(4):
+                                self.d_atom[ib][ia] = blocks[ib]
+                            continue
+                        self.d_atom[0][ia] = ions + d0 + dz        # up-up
+                        self.d_atom[1][ia] = ions + d0 - dz        # dn-dn
+                        self.d_atom[2][ia] = dx - 1j * dy          # up-dn
+                        self.d_atom[3][ia] = dx + 1j * dy          # dn-up
+                    elif nsp == 2:
+                        d1 = unpack(dt1)
+                        if paw is not None and ia in paw.dij:
                             d1 = d1 + paw.dij[ia][..., 1]
-                    for ispn in range(nsp):
-                        d = np.array(ions)
-                        d += d0 if nsp == 1 else (d0 + d1 if ispn == 0 else d0 - d1)
-                        self.d_atom[ispn][ia] = d
+                        self.d_atom[0][ia] = ions + d0 + d1
+                        self.d_atom[1][ia] = ions + d0 - d1
+                    else:
+                        self.d_atom[0][ia] = ions + d0
             else:
-                for ia in ia_list:
-                    for ispn in range(nsp):
-                        self.d_atom[ispn][ia] = ions
+                if ctx.nc_magnetism and at.spin_orbit and at.num_beta:
+                    from . import so as so_mod
+
+                    fc = ctx.so_fcoef(lab)
+                    zero = np.zeros_like(ions)
+                    blocks = so_mod.so_d_blocks(at, [zero] * 4, fc)
+                    for ia in ia_list:
+                        for ib in range(4):
+                            self.d_atom[ib][ia] = blocks[ib]
+                else:
+                    for ia in ia_list:
+                        for ib in range(nblk):
+                            if ib < 2 or not ctx.nc_magnetism:
+                                self.d_atom[ib][ia] = ions
+                            else:
+                                self.d_atom[ib][ia] = np.zeros_like(ions)
 
     def __call__(self, kp) -> "HamiltonianK":
         return HamiltonianK(self, kp)
@@ -197,41 +243,78 @@ class HamiltonianK:
             kp.beta = BetaProjectors(self.ctx, kp)
         self.bp = kp.beta
         ctx = self.ctx
+        nblk = 4 if ctx.nc_magnetism else ctx.num_spins
         self.D = [block_diag_matrix(ctx, self.bp,
                                     lambda ia, s=s: h0.d_atom[s][ia])
-                  for s in range(ctx.num_spins)]
+                  for s in range(nblk)]
         self.Q = None  # USPP overlap augmentation (else S = I)
+        self.Q_blocks = None  # nc/SO spin blocks [Q00, Q11, Q01, Q10]
         if ctx.has_aug:
             uc = ctx.unit_cell
+            has_so = any(at.spin_orbit and at.augment
+                         for at in uc.atom_types.values())
 
             def qmat(ia):
                 lab = uc.atoms[ia][0]
                 at = uc.atom_types[lab]
                 return ctx.aug_op(lab).q_mtrx if at.augment else None
 
-            self.Q = block_diag_matrix(ctx, self.bp, qmat)
+            if ctx.nc_magnetism and has_so:
+                from . import so as so_mod
+
+                def qblk(ia, ib):
+                    lab = uc.atoms[ia][0]
+                    at = uc.atom_types[lab]
+                    if not at.augment:
+                        return None
+                    q = ctx.aug_op(lab).q_mtrx
+                    if at.spin_orbit:
+                        return so_mod.so_q_blocks(at, q, ctx.so_fcoef(lab))[ib]
+                    return q if ib < 2 else np.zeros_like(q)
+
+                self.Q_blocks = [block_diag_matrix(ctx, self.bp,
+                                                   lambda ia, b=b: qblk(ia, b))
+                                 for b in range(4)]
+                self.Q = self.Q_blocks[0]
+            else:
+                self.Q = block_diag_matrix(ctx, self.bp, qmat)
         self.ekin = 0.5 * kp.gkvec.gk2_t  # [nGk] float64
 
     def h_diag(self, ispn: int = 0) -> torch.Tensor:
         """Diagonal of H for the Davidson preconditioner
         (reference get_h_o_diag_pw, hamiltonian_k.cpp:67)."""
-        v0 = float(self.h0.veff_r_coarse[ispn].mean())
-        d = self.ekin + v0
-        if self.bp.num_beta_total:
-            # sum_{ij} conj(β_i(G)) D_ij β_j(G) per G
-            bD = self.bp.beta @ self.D[ispn].T  # [nGk, nbf]
-            d = d + torch.einsum("gi,gi->g", bD, self.bp.beta.conj()).real
-        return d
+        def diag_for(s):
+            v0 = float(self.h0.veff_r_coarse[s].mean())
+            d = self.ekin + v0
+            if self.bp.num_beta_total:
+                bD = self.bp.beta @ self.D[s].T  # [nGk, nbf]
+                d = d + torch.einsum("gi,gi->g", bD, self.bp.beta.conj()).real
+            return d
+        if self.ctx.nc_magnetism:
+            return torch.cat([diag_for(0), diag_for(1)])
+        return diag_for(ispn)
 
     def o_diag(self) -> torch.Tensor:
+        if self.ctx.nc_magnetism and self.Q_blocks is not None:
+            outs = []
+            for b in (0, 1):
+                d = torch.ones_like(self.ekin)
+                bQ = self.bp.beta @ self.Q_blocks[b].T
+                d = d + torch.einsum("gi,gi->g", bQ, self.bp.beta.conj()).real
+                outs.append(d)
+            return torch.cat(outs)
         d = torch.ones_like(self.ekin)
         if self.Q is not None:
             bQ = self.bp.beta @ self.Q.T
             d = d + torch.einsum("gi,gi->g", bQ, self.bp.beta.conj()).real
+        if self.ctx.nc_magnetism:
+            return torch.cat([d, d])
         return d
 
     def apply_h_s(self, psi: torch.Tensor, ispn: int = 0):
-        """psi [nb, nGk] -> (hpsi, spsi). spsi is None for NC (S = I)."""
+        """psi [nb, nGk] (or [nb, 2·nGk] spinors) -> (hpsi, spsi)."""
+        if self.ctx.nc_magnetism:
+            return self._apply_h_s_nc(psi)
         kp = self.kp
         # fused local operator: FFT⁻¹ → ×V_eff(r) → FFT → +½|G+k|²ψ
         hpsi = kp.fft.apply_veff_kinetic(psi, self.h0.veff_r_coarse[ispn],
@@ -245,4 +328,36 @@ class HamiltonianK:
                 spsi = psi + (self.bp.beta @ (self.Q @ bphi)).T
         if self.ctx.hubbard is not None:
             self.ctx.hubbard.apply(kp, self, psi, hpsi, ispn)
+        return hpsi, spsi
+
+    def _apply_h_s_nc(self, psi: torch.Tensor):
+        """Spinor apply (hamiltonian.hpp:454-515 noncollinear branch):
+        h↑ = (T+V+Bz)ψ↑ + (Bx−iBy)ψ↓ + β[D⁰⁰⟨β|ψ↑⟩ + D⁰¹⟨β|ψ↓⟩], etc."""
+        kp = self.kp
+        ng = kp.num_gkvec
+        up, dn = psi[:, :ng], psi[:, ng:]
+        h0 = self.h0
+        up_r = kp.fft.to_real(up)
+        dn_r = kp.fft.to_real(dn)
+        vud = h0.v_ud_coarse
+        hup = kp.fft.to_pw(up_r * h0.veff_r_coarse[0] + dn_r * vud)
+        hdn = kp.fft.to_pw(dn_r * h0.veff_r_coarse[1] + up_r * vud.conj())
+        hup = hup + self.ekin * up
+        hdn = hdn + self.ekin * dn
+        sup = sdn = None
+        if self.bp.num_beta_total:
+            bu = self.bp.inner(up)
+            bd = self.bp.inner(dn)
+            hup = hup + (self.bp.beta @ (self.D[0] @ bu + self.D[2] @ bd)).T
+            hdn = hdn + (self.bp.beta @ (self.D[1] @ bd + self.D[3] @ bu)).T
+            if self.Q_blocks is not None:
+                sup = up + (self.bp.beta @ (self.Q_blocks[0] @ bu
+                                            + self.Q_blocks[2] @ bd)).T
+                sdn = dn + (self.bp.beta @ (self.Q_blocks[1] @ bd
+                                            + self.Q_blocks[3] @ bu)).T
+            elif self.Q is not None:
+                sup = up + (self.bp.beta @ (self.Q @ bu)).T
+                sdn = dn + (self.bp.beta @ (self.Q @ bd)).T
+        hpsi = torch.cat([hup, hdn], dim=1)
+        spsi = torch.cat([sup, sdn], dim=1) if sup is not None else None
         return hpsi, spsi

@@ -84,10 +84,12 @@ class KPoint:
         self.fft = SphericalFFT(self.gkvec)
         self.num_gkvec = self.gkvec.num_gvec
         nb = ctx.num_bands
-        self.psi = torch.zeros(ctx.num_spins, nb, self.num_gkvec,
-                               dtype=ctx.dtype, device=ctx.device)
-        self.eigvals = np.zeros((ctx.num_spins, nb))
-        self.occ = np.zeros((ctx.num_spins, nb))
+        nss = ctx.num_spin_steps
+        wf_len = self.num_gkvec * ctx.num_spinors   # spinor components stacked
+        self.psi = torch.zeros(nss, nb, wf_len, dtype=ctx.dtype,
+                               device=ctx.device)
+        self.eigvals = np.zeros((nss, nb))
+        self.occ = np.zeros((nss, nb))
         self.beta = None  # BetaProjectors, built lazily by Hamiltonian
 
 
@@ -147,7 +149,7 @@ class KPointSet:
         p = ctx.cfg.parameters
         self.sync_band()
         ne = ctx.unit_cell.num_electrons
-        eig = self._all_eig  # [nk, nspin, nb]
+        eig = self._all_eig  # [nk, num_spin_steps, nb]
         nk, nspin, nb = eig.shape
         eigf = eig.reshape(nk * nspin, nb)
         wf = np.repeat(self._all_w, nspin)

@@ -34,6 +34,8 @@ class Potential:
         self.veff_g = None
         self.bz_r = None              # collinear XC magnetic field B_z(r)
         self.bz_g = None
+        self.bvec_r = None            # noncollinear [Bx, By, Bz](r)
+        self.bvec_g = None
         self.energy_vha = 0.0
         self.ewald = None
         self._ig0 = g.index_of_zero()
@@ -91,7 +93,36 @@ class Potential:
         self.energy_vha = ctx.integrate_rg_fine(density.rho_r * vha_r)
 
         rho_xc = density.rho_r + density.rho_core_r
-        if ctx.num_spins == 1:
+        if ctx.nc_magnetism:
+            # noncollinear: locally collinear along m̂ (xc_rg_magnetic with
+            # rho_up/dn = (ρ+core ± |m|)/2; B = ½(vu−vd)·m̂)
+            mx, my, mz = (density.magv_r[0], density.magv_r[1], density.magv_r[2])
+            mlen = torch.sqrt(mx**2 + my**2 + mz**2).clamp(min=1e-30)
+            ru = 0.5 * (rho_xc + mlen)
+            rd = 0.5 * (rho_xc - mlen)
+            if ctx.is_gga:
+                gu = self._grad_r(ru)
+                gd = self._grad_r(rd)
+                s_uu = gu[0]**2 + gu[1]**2 + gu[2]**2
+                s_dd = gd[0]**2 + gd[1]**2 + gd[2]**2
+                gt = [gu[d] + gd[d] for d in range(3)]
+                s_tot = gt[0]**2 + gt[1]**2 + gt[2]**2
+                eps, vu, vd, vs_uu, vs_dd, vs_tot = xc_mod.evaluate_spin(
+                    ctx.xc_names, ru, rd, s_uu, s_dd, s_tot)
+                vu = vu - 2.0 * self._div_r([vs_uu * gu[d] for d in range(3)])                         - 2.0 * self._div_r([vs_tot * gt[d] for d in range(3)])
+                vd = vd - 2.0 * self._div_r([vs_dd * gd[d] for d in range(3)])                         - 2.0 * self._div_r([vs_tot * gt[d] for d in range(3)])
+            else:
+                eps, vu, vd, *_ = xc_mod.evaluate_spin(ctx.xc_names, ru, rd)
+            self.vxc_r = 0.5 * (vu + vd)
+            b_amp = 0.5 * (vu - vd)
+            self.bvec_r = [b_amp * mx / mlen, b_amp * my / mlen,
+                           b_amp * mz / mlen]
+            self.bvec_g = [ctx.fft_fine.to_pw(b.to(ctx.dtype))
+                           for b in self.bvec_r]
+            self.bz_r = self.bvec_r[2]
+            self.bz_g = self.bvec_g[2]
+            self.exc_r = eps
+        elif ctx.num_spins == 1:
             if ctx.is_gga:
                 grads = self._grad_r(rho_xc)
                 sigma = grads[0] ** 2 + grads[1] ** 2 + grads[2] ** 2
@@ -138,7 +169,10 @@ class Potential:
             self.ctx.paw.generate(density)
 
     def energy_bxc(self, density) -> float:
-        """∫ m_z·B_z (reference energy.cpp:85-93)."""
+        """∫ m·B (reference energy.cpp:85-93)."""
+        if self.ctx.nc_magnetism and self.bvec_r is not None:
+            return sum(self.ctx.integrate_rg_fine(density.magv_r[i] * self.bvec_r[i])
+                       for i in range(3))
         if self.bz_r is None or density.mag_r is None:
             return 0.0
         return self.ctx.integrate_rg_fine(density.mag_r * self.bz_r)
