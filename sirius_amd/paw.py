@@ -40,8 +40,8 @@ class AngularGrid:
         self.lmax = lmax
         self.lmmax = ylm_mod.lmmax(lmax)
         ltot = 3 * lmax + 2
-        nth = ltot // 2 + 2
-        nph = ltot + 2
+        nth = ltot + 4          # oversampled: XC is nonlinear on the sphere
+        nph = 2 * ltot + 4
         x, wx = np.polynomial.legendre.leggauss(nth)
         theta = np.arccos(x)
         phi = np.arange(nph) * 2 * math.pi / nph
