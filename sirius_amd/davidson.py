@@ -87,9 +87,26 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
     if occ is not None:
         tol = np.where(occ > min_occupancy, tol_occ, tol_occ + tol_empty)
 
+    # preallocated subspace buffers: blocks are appended in place instead of
+    # torch.cat (each cat re-copies the whole [N, nG] subspace — measured as
+    # ~5% pure copyBuffer traffic in profiles/r01_si64_1gpu_kernel_stats_v2)
+    dev, cdt = psi0.device, psi0.dtype
+    phi_buf = torch.empty(num_phi_max, nG, dtype=cdt, device=dev)
+    hphi_buf = torch.empty(num_phi_max, nG, dtype=cdt, device=dev)
+    sphi_buf = None
+
     phi = psi0.clone()
     hphi, sphi = apply_h_s(phi)
     phi, hphi, sphi = _ortho_joint(phi, hphi, sphi, None, None, None)
+    n0 = phi.shape[0]
+    phi_buf[:n0] = phi
+    hphi_buf[:n0] = hphi
+    phi = phi_buf[:n0]
+    hphi = hphi_buf[:n0]
+    if sphi is not None:
+        sphi_buf = torch.empty(num_phi_max, nG, dtype=cdt, device=dev)
+        sphi_buf[:n0] = sphi
+        sphi = sphi_buf[:n0]
     H = _inner(phi, hphi)
     H = 0.5 * (H + H.conj().T)
     evals, Z = la.eigh(H)
@@ -143,7 +160,13 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
             psi = Z[:, :nb].T @ phi
             hpsi_f = Z[:, :nb].T @ hphi
             spsi_f = Z[:, :nb].T @ sphi if sphi is not None else None
-            phi, hphi, sphi = psi, hpsi_f, spsi_f
+            phi_buf[:nb] = psi
+            hphi_buf[:nb] = hpsi_f
+            phi = phi_buf[:nb]
+            hphi = hphi_buf[:nb]
+            if spsi_f is not None:
+                sphi_buf[:nb] = spsi_f
+                sphi = sphi_buf[:nb]
             H = torch.diag(evals[:nb].to(H.dtype))
             evals = evals[:nb].clone()
             Z = torch.eye(nb, dtype=H.dtype, device=H.device)
@@ -161,11 +184,15 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
         if res.shape[0] == 0:
             converged = True
             break
-        # grow subspace
-        phi = torch.cat([phi, res], dim=0)
-        hphi = torch.cat([hphi, hnew], dim=0)
+        # grow subspace in place
+        nn = res.shape[0]
+        phi_buf[N:N + nn] = res
+        hphi_buf[N:N + nn] = hnew
+        phi = phi_buf[:N + nn]
+        hphi = hphi_buf[:N + nn]
         if sphi is not None:
-            sphi = torch.cat([sphi, snew], dim=0)
+            sphi_buf[N:N + nn] = snew
+            sphi = sphi_buf[:N + nn]
         Nn = phi.shape[0]
         Hn = torch.empty(Nn, Nn, dtype=H.dtype, device=H.device)
         Hn[:N, :N] = H

@@ -13,6 +13,7 @@ import os
 import torch
 
 _ext = None
+_ext_zgemm = None
 _tried = False
 
 _SRC_DIR = os.path.join(os.path.dirname(os.path.abspath(__file__)), "src")
@@ -20,9 +21,9 @@ _BUILD_DIR = os.path.join(os.path.dirname(os.path.abspath(__file__)), "_build")
 
 
 def build_extensions(verbose: bool = False):
-    """Compile the HIP extension for gfx950 (cross-compiles fine on CPU-only
+    """Compile the HIP extensions for gfx950 (cross-compiles fine on CPU-only
     boxes; hipcc needs no GPU present)."""
-    global _ext, _tried
+    global _ext, _ext_zgemm, _tried
     _tried = True
     os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
     os.makedirs(_BUILD_DIR, exist_ok=True)
@@ -34,6 +35,16 @@ def build_extensions(verbose: bool = False):
         extra_cflags=["-O3"],
         extra_cuda_cflags=["-O3"],
         build_directory=_BUILD_DIR,
+        verbose=verbose,
+    )
+    zdir = os.path.join(_BUILD_DIR, "zgemm")
+    os.makedirs(zdir, exist_ok=True)
+    _ext_zgemm = load(
+        name="sirius_amd_zgemm",
+        sources=[os.path.join(_SRC_DIR, "zgemm_gram.hip")],
+        extra_cflags=["-O3"],
+        extra_cuda_cflags=["-O3"],
+        build_directory=zdir,
         verbose=verbose,
     )
     return _ext
@@ -62,6 +73,12 @@ def get_ext(required: bool | None = None):
     if _ext is None and required:
         raise RuntimeError("sirius_amd HIP extension unavailable on GPU machine")
     return _ext
+
+
+def get_zgemm(required: bool | None = None):
+    """The MFMA fp64 Gram-GEMM extension (None on CPU)."""
+    get_ext(required)
+    return _ext_zgemm
 
 
 def available() -> bool:

@@ -1,0 +1,148 @@
+// Hand-written CDNA4 (gfx950) MFMA fp64 kernel for the Davidson subspace
+// Gram product:  C[M,N] = Σ_k conj(A[m,k])·B[n,k]
+//
+// This is the reference's SPLA pgemm_ssb seam (wf::inner,
+// wave_functions.hpp:1659-1723) — the ⟨φ|φ⟩ / ⟨φ|Hφ⟩ tall-skinny
+// contraction with K = num_gvec (10⁴-10⁵) and M,N = subspace size
+// (10²-10³). rocBLAS/Tensile reaches only ~13.5 TF/s fp64 on this shape
+// (profiles/r01_si64_1gpu_kernel_stats_v2.csv); this kernel uses
+// v_mfma_f64_16x16x4f64 tiles with LDS staging and split-K over the long
+// dimension, combining with global fp64 atomic adds.
+//
+// Geometry: 256 threads = 4 waves; each wave owns a 16×16 complex output
+// tile; the workgroup computes a 32×32 complex tile of C. blockIdx.z
+// split-K chunks stream disjoint K ranges. A and B rows are contiguous
+// in k (torch row-major [rows, K]), so LDS stages are fully coalesced.
+
+#include <hip/hip_runtime.h>
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+typedef double d4 __attribute__((ext_vector_type(4)));
+
+#define TILE 32              // block tile (M and N)
+#define KCH 16               // k-chunk staged in LDS per iteration
+
+__launch_bounds__(256, 2)
+__global__ void zgram_splitk_kernel(const double* __restrict__ A,  // [M, 2K] interleaved
+                                    const double* __restrict__ B,  // [N, 2K]
+                                    double* __restrict__ C,        // [M, 2N] interleaved
+                                    int M, int N, long K, long chunk) {
+    const long k_begin = (long)blockIdx.z * chunk;
+    const long k_end = min(K, k_begin + chunk);
+    // LDS: A tile [TILE rows][KCH][2], B tile same: 2 * 32*16*2*8 = 16 KiB
+    __shared__ double lA[TILE][2 * KCH + 2];
+    __shared__ double lB[TILE][2 * KCH + 2];
+
+    const int m0 = blockIdx.x * TILE;
+    const int n0 = blockIdx.y * TILE;
+    if (m0 >= M || n0 >= N) {
+        return;
+    }
+
+    const int tid = threadIdx.x;
+    const int wave = tid / 64;          // 0..3
+    const int lane = tid % 64;
+    // wave grid 2x2 over the 32x32 tile
+    const int wm = (wave / 2) * 16;     // wave row offset inside tile
+    const int wn = (wave % 2) * 16;     // wave col offset
+
+    // accumulators: C_re, C_im (16x16 f64 tile each, 4 values per lane)
+    d4 acc_re = {0, 0, 0, 0};
+    d4 acc_im = {0, 0, 0, 0};
+
+    // mfma_f64_16x16x4: A operand: lane l holds A[row = l%16, k = l/16]
+    // B operand: lane l holds B[k = l/16, col = l%16]
+    const int a_row = lane % 16;
+    const int a_k = lane / 16;          // 0..3
+
+    for (long kb = k_begin; kb < k_end; kb += KCH) {
+        const int kc = (int)min((long)KCH, k_end - kb);
+        // stage A/B rows: thread t loads row t/8, dword pair t%8 → strided
+        // over the chunk; 256 threads cover 32 rows × 8 slots of 2 complex
+        for (int s = tid; s < TILE * KCH; s += 256) {
+            int r = s / KCH;
+            int k = s % KCH;
+            double re = 0.0, im = 0.0;
+            if (k < kc && m0 + r < M) {
+                const double* src = A + ((long)(m0 + r)) * 2 * K + 2 * (kb + k);
+                re = src[0];
+                im = src[1];
+            }
+            lA[r][2 * k] = re;
+            lA[r][2 * k + 1] = im;
+            re = 0.0;
+            im = 0.0;
+            if (k < kc && n0 + r < N) {
+                const double* src = B + ((long)(n0 + r)) * 2 * K + 2 * (kb + k);
+                re = src[0];
+                im = src[1];
+            }
+            lB[r][2 * k] = re;
+            lB[r][2 * k + 1] = im;
+        }
+        __syncthreads();
+
+        // 4 MFMA k-steps of 4 over the 16-wide chunk
+        #pragma unroll
+        for (int ks = 0; ks < KCH; ks += 4) {
+            double ar = lA[wm + a_row][2 * (ks + a_k)];
+            double ai = lA[wm + a_row][2 * (ks + a_k) + 1];
+            double br = lB[wn + a_row][2 * (ks + a_k)];
+            double bi = lB[wn + a_row][2 * (ks + a_k) + 1];
+            // C = conj(a)·b: re += ar·br + ai·bi ; im += ar·bi − ai·br
+            acc_re = __builtin_amdgcn_mfma_f64_16x16x4f64(ar, br, acc_re, 0, 0, 0);
+            acc_re = __builtin_amdgcn_mfma_f64_16x16x4f64(ai, bi, acc_re, 0, 0, 0);
+            acc_im = __builtin_amdgcn_mfma_f64_16x16x4f64(ar, bi, acc_im, 0, 0, 0);
+            acc_im = __builtin_amdgcn_mfma_f64_16x16x4f64(-ai, br, acc_im, 0, 0, 0);
+        }
+        __syncthreads();
+    }
+
+    // accumulator layout of mfma_f64_16x16x4 (probed on gfx950 hardware):
+    // lane l, vreg v holds C[row = 4*v + l/16, col = l%16]
+    const int c_col = lane % 16;
+    const int c_rowg = lane / 16;
+    #pragma unroll
+    for (int v = 0; v < 4; v++) {
+        int m = m0 + wm + 4 * v + c_rowg;
+        int n = n0 + wn + c_col;
+        if (m < M && n < N) {
+            // NOTE: MFMA computes a*b accumulate; row/col mapping of the
+            // A operand puts OUR row index on the output row — C[m,n]
+            unsafeAtomicAdd(&C[(long)m * 2 * N + 2 * n], acc_re[v]);
+            unsafeAtomicAdd(&C[(long)m * 2 * N + 2 * n + 1], acc_im[v]);
+        }
+    }
+}
+
+torch::Tensor zgram(torch::Tensor A, torch::Tensor B, int split_k) {
+    TORCH_CHECK(A.is_cuda() && B.is_cuda(), "device tensors required");
+    TORCH_CHECK(A.is_contiguous() && B.is_contiguous(), "contiguous required");
+    const int M = A.size(0);
+    const int N = B.size(0);
+    const long K = A.size(1);
+    TORCH_CHECK(B.size(1) == K, "K mismatch");
+    auto C = torch::zeros({M, N}, A.options());
+    if (split_k <= 0) {
+        // heuristic: enough chunks to fill 256 CUs
+        long tiles = ((M + TILE - 1) / TILE) * ((N + TILE - 1) / TILE);
+        split_k = (int)std::min<long>(64, std::max<long>(1, 512 / std::max<long>(tiles, 1)));
+    }
+    long chunk = (K + split_k - 1) / split_k;
+    chunk = ((chunk + KCH - 1) / KCH) * KCH;   // multiple of the LDS stage
+    split_k = (int)((K + chunk - 1) / chunk);
+
+    dim3 grid((M + TILE - 1) / TILE, (N + TILE - 1) / TILE, split_k);
+    auto stream = at::hip::getCurrentHIPStream().stream();
+    hipLaunchKernelGGL(zgram_splitk_kernel, grid, dim3(256), 0, stream,
+                       (const double*)A.data_ptr(), (const double*)B.data_ptr(),
+                       (double*)C.data_ptr(), M, N, K, chunk);
+    return C;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("zgram", &zgram, "C[m,n] = sum_k conj(A[m,k]) B[n,k] (MFMA f64)",
+          py::arg("A"), py::arg("B"), py::arg("split_k") = 0);
+}
