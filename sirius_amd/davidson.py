@@ -49,11 +49,14 @@ def _ortho_joint(new, hnew, snew, phi, hphi, sphi):
     s_of_new = snew if snew is not None else new
     if phi is not None and phi.shape[0]:
         ov = _inner(sphi if sphi is not None else phi, new)   # [N, n]
-        new = new - ov.T @ phi
+        new = la.transform(ov, phi, out=new.contiguous(), alpha=-1.0,
+                           accumulate=True)
         if hnew is not None:
-            hnew = hnew - ov.T @ hphi
+            hnew = la.transform(ov, hphi, out=hnew.contiguous(), alpha=-1.0,
+                                accumulate=True)
         if snew is not None:
-            snew = snew - ov.T @ sphi
+            snew = la.transform(ov, sphi, out=snew.contiguous(), alpha=-1.0,
+                                accumulate=True)
         s_of_new = snew if snew is not None else new
     gram = _inner(new, s_of_new)
     gram = 0.5 * (gram + gram.conj().T)
@@ -64,9 +67,10 @@ def _ortho_joint(new, hnew, snew, phi, hphi, sphi):
         w, v = la.eigh(gram)
         keep = w > 1e-10
         t = (v[:, keep] / torch.sqrt(w[keep])).conj().T     # [nkeep, n]
-    new = t @ new
-    hnew = t @ hnew if hnew is not None else None
-    snew = t @ snew if snew is not None else None
+    tT = t.transpose(0, 1)
+    new = la.transform(tT, new)
+    hnew = la.transform(tT, hnew) if hnew is not None else None
+    snew = la.transform(tT, snew) if snew is not None else None
     return new, hnew, snew
 
 
@@ -128,8 +132,8 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
         idx = torch.from_numpy(unconv).to(psi0.device)
         Zs = Z[:, idx]                                     # [N, n]
         e = evals[idx].real
-        hpsi = Zs.T @ hphi                                  # [n, nG]
-        spsi = Zs.T @ (sphi if sphi is not None else phi)
+        hpsi = la.transform(Zs, hphi)                       # [n, nG]
+        spsi = la.transform(Zs, sphi if sphi is not None else phi)
         if hpsi.is_cuda:
             from . import ops
 
@@ -157,9 +161,10 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
         n_new = res.shape[0]
         if N + n_new > num_phi_max:
             # restart: collapse subspace to current Ritz vectors
-            psi = Z[:, :nb].T @ phi
-            hpsi_f = Z[:, :nb].T @ hphi
-            spsi_f = Z[:, :nb].T @ sphi if sphi is not None else None
+            Znb = Z[:, :nb]
+            psi = la.transform(Znb, phi)
+            hpsi_f = la.transform(Znb, hphi)
+            spsi_f = la.transform(Znb, sphi) if sphi is not None else None
             phi_buf[:nb] = psi
             hphi_buf[:nb] = hpsi_f
             phi = phi_buf[:nb]
@@ -202,6 +207,6 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
         H = 0.5 * (Hn + Hn.conj().T)
         evals, Z = la.eigh(H)
 
-    psi = Z[:, :nb].T @ phi
+    psi = la.transform(Z[:, :nb], phi)
     return DavidsonResult(eval=evals[:nb].real.cpu().numpy(), psi=psi,
                           niter=niter, converged=converged)

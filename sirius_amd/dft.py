@@ -92,7 +92,7 @@ def initialize_subspace(ctx, kp, hk):
     H = 0.5 * (H + H.conj().T)
     evals, Z = la.eigh(H)
     nsel = min(nb, phi.shape[0])
-    psi = Z[:, :nsel].T @ phi
+    psi = la.transform(Z[:, :nsel], phi)
     if nsel < nb:
         raise RuntimeError("not enough trial orbitals for requested bands")
     for ispn in range(ctx.num_spin_steps):

@@ -25,6 +25,7 @@ import math
 import numpy as np
 import torch
 
+from .core import la
 from .core import ylm as ylm_mod
 from .core.radial import RadialIntegrals
 
@@ -75,15 +76,19 @@ class BetaProjectors:
             phase = np.exp(-2j * math.pi * (mk @ tau))  # [nGk]
             blocks.append(cols_t[lab] * phase[:, None])
         self.num_beta_total = off
+        # stored TRANSPOSED [nbf_tot, nGk]: both ⟨β|ψ⟩ (Gram) and the
+        # β·(D⟨β|ψ⟩) apply (transform) then run on the MFMA zgemm kernels
+        # (ops/src/zgemm_gram.hip) with fully coalesced row-major access.
         if blocks:
-            self.beta = torch.from_numpy(np.concatenate(blocks, axis=1)).to(dev)
+            bt = np.ascontiguousarray(np.concatenate(blocks, axis=1).T)
+            self.beta_t = torch.from_numpy(bt).to(dev)
         else:
-            self.beta = torch.zeros(len(glen), 0, dtype=ctx.dtype, device=dev)
+            self.beta_t = torch.zeros(0, len(glen), dtype=ctx.dtype, device=dev)
 
     def inner(self, psi: torch.Tensor) -> torch.Tensor:
-        """⟨β|ψ⟩ = β^H ψ: [nbf_tot, nb] (zgemm; reference
-        non_local_operator_base.hpp:130-199 inner_prod_beta)."""
-        return self.beta.conj().T @ psi.T  # psi [nb, nGk] -> [nbf, nb]
+        """⟨β|ψ⟩ = β^H ψ: [nbf_tot, nb] (MFMA Gram kernel; reference
+        non_local_operator_base.hpp:130-199 inner_prod_beta → SPLA)."""
+        return la.inner(self.beta_t, psi)  # psi [nb, nGk] -> [nbf, nb]
 
 
 def block_diag_matrix(ctx, bp: BetaProjectors, per_atom_matrix):
@@ -287,8 +292,8 @@ class HamiltonianK:
             v0 = float(self.h0.veff_r_coarse[s].mean())
             d = self.ekin + v0
             if self.bp.num_beta_total:
-                bD = self.bp.beta @ self.D[s].T  # [nGk, nbf]
-                d = d + torch.einsum("gi,gi->g", bD, self.bp.beta.conj()).real
+                bD = self.D[s] @ self.bp.beta_t  # [nbf, nGk]
+                d = d + torch.einsum("ig,ig->g", bD, self.bp.beta_t.conj()).real
             return d
         if self.ctx.nc_magnetism:
             return torch.cat([diag_for(0), diag_for(1)])
@@ -299,14 +304,14 @@ class HamiltonianK:
             outs = []
             for b in (0, 1):
                 d = torch.ones_like(self.ekin)
-                bQ = self.bp.beta @ self.Q_blocks[b].T
-                d = d + torch.einsum("gi,gi->g", bQ, self.bp.beta.conj()).real
+                bQ = self.Q_blocks[b] @ self.bp.beta_t
+                d = d + torch.einsum("ig,ig->g", bQ, self.bp.beta_t.conj()).real
                 outs.append(d)
             return torch.cat(outs)
         d = torch.ones_like(self.ekin)
         if self.Q is not None:
-            bQ = self.bp.beta @ self.Q.T
-            d = d + torch.einsum("gi,gi->g", bQ, self.bp.beta.conj()).real
+            bQ = self.Q @ self.bp.beta_t
+            d = d + torch.einsum("ig,ig->g", bQ, self.bp.beta_t.conj()).real
         if self.ctx.nc_magnetism:
             return torch.cat([d, d])
         return d
@@ -323,9 +328,12 @@ class HamiltonianK:
         spsi = None
         if self.bp.num_beta_total:
             bphi = self.bp.inner(psi)           # [nbf, nb]
-            hpsi += (self.bp.beta @ (self.D[ispn] @ bphi)).T
+            la.transform(self.D[ispn] @ bphi, self.bp.beta_t,
+                         out=hpsi, accumulate=True)
             if self.Q is not None:
-                spsi = psi + (self.bp.beta @ (self.Q @ bphi)).T
+                spsi = psi.clone()
+                la.transform(self.Q @ bphi, self.bp.beta_t,
+                             out=spsi, accumulate=True)
         if self.ctx.hubbard is not None:
             self.ctx.hubbard.apply(kp, self, psi, hpsi, ispn)
         return hpsi, spsi
@@ -348,16 +356,23 @@ class HamiltonianK:
         if self.bp.num_beta_total:
             bu = self.bp.inner(up)
             bd = self.bp.inner(dn)
-            hup = hup + (self.bp.beta @ (self.D[0] @ bu + self.D[2] @ bd)).T
-            hdn = hdn + (self.bp.beta @ (self.D[1] @ bd + self.D[3] @ bu)).T
+            bt = self.bp.beta_t
+            la.transform(self.D[0] @ bu + self.D[2] @ bd, bt,
+                         out=hup, accumulate=True)
+            la.transform(self.D[1] @ bd + self.D[3] @ bu, bt,
+                         out=hdn, accumulate=True)
             if self.Q_blocks is not None:
-                sup = up + (self.bp.beta @ (self.Q_blocks[0] @ bu
-                                            + self.Q_blocks[2] @ bd)).T
-                sdn = dn + (self.bp.beta @ (self.Q_blocks[1] @ bd
-                                            + self.Q_blocks[3] @ bu)).T
+                sup = up.clone().contiguous()
+                sdn = dn.clone().contiguous()
+                la.transform(self.Q_blocks[0] @ bu + self.Q_blocks[2] @ bd,
+                             bt, out=sup, accumulate=True)
+                la.transform(self.Q_blocks[1] @ bd + self.Q_blocks[3] @ bu,
+                             bt, out=sdn, accumulate=True)
             elif self.Q is not None:
-                sup = up + (self.bp.beta @ (self.Q @ bu)).T
-                sdn = dn + (self.bp.beta @ (self.Q @ bd)).T
+                sup = up.clone().contiguous()
+                sdn = dn.clone().contiguous()
+                la.transform(self.Q @ bu, bt, out=sup, accumulate=True)
+                la.transform(self.Q @ bd, bt, out=sdn, accumulate=True)
         hpsi = torch.cat([hup, hdn], dim=1)
         spsi = torch.cat([sup, sdn], dim=1) if sup is not None else None
         return hpsi, spsi

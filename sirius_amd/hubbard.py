@@ -185,7 +185,7 @@ class HubbardModule:
             ov = 0.5 * (ov + ov.conj().T)
             w, v = la.eigh(ov)
             B = (v / torch.sqrt(w.clamp(min=1e-12))) @ v.conj().T  # S^{-1/2}
-            phi_all = B.T @ phi_all
+            phi_all = la.transform(B, phi_all)
             sphi_all = self._apply_S(phi_all, hk)
         # extract the hubbard channels in level order
         rows = []
@@ -218,7 +218,9 @@ class HubbardModule:
         if hk.Q is None or hk.bp.num_beta_total == 0:
             return phi
         bphi = hk.bp.inner(phi)
-        return phi + (hk.bp.beta @ (hk.Q @ bphi)).T
+        out = phi.clone().contiguous()
+        la.transform(hk.Q @ bphi, hk.bp.beta_t, out=out, accumulate=True)
+        return out
 
     # -- occupation matrix -------------------------------------------------
 
@@ -394,7 +396,7 @@ class HubbardModule:
         swf = self.hubbard_wf_S(kp, hk)
         dm = la.inner(swf, psi)                  # [nwf, nb]
         up = self.u_matrix_full(ispn, kp.k_frac) @ dm
-        hpsi += up.T @ swf
+        la.transform(up, swf, out=hpsi, accumulate=True)
 
     # -- energies ----------------------------------------------------------
 

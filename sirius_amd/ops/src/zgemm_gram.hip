@@ -117,6 +117,132 @@ __global__ void zgram_splitk_kernel(const double* __restrict__ A,  // [M, 2K] in
     }
 }
 
+// Subspace transform:  C[m,g] = alpha * Σ_k op(T[k,m]) · X[k,g]  (+ C[m,g])
+//
+// The other half of the Davidson seam (wf::transform, SPLA pgemm_sbs,
+// wave_functions.hpp:1441-1520): applying the small subspace matrix Z
+// ([K,M], K=subspace N ≈ 150-600) to the wavefunction stack X ([K, nG],
+// nG ≈ 10⁴-10⁵). rocBLAS runs this at ~2.2 TF/s (Cijk_Alik_* tiles,
+// profiles/r01 prof3). Here: each workgroup owns a 32m×32g output tile
+// (2×2 waves of 16×16 MFMA), k-loop over the short dimension, direct
+// coalesced writeback (no atomics; G/32 tiles fill the 256 CUs).
+__launch_bounds__(256, 2)
+__global__ void ztrans_kernel(const double* __restrict__ T,  // [K, 2M] interleaved
+                              const double* __restrict__ X,  // [K, 2G]
+                              double* __restrict__ C,        // [M, 2G]
+                              int M, long G, int K, int conj_t,
+                              double alpha, int accumulate) {
+    __shared__ double lT[KCH][2 * TILE + 2];
+    __shared__ double lX[KCH][2 * TILE + 2];
+
+    const int m0 = blockIdx.x * TILE;
+    const long g0 = (long)blockIdx.y * TILE;
+    if (m0 >= M || g0 >= G) {
+        return;
+    }
+    const int tid = threadIdx.x;
+    const int wave = tid / 64;
+    const int lane = tid % 64;
+    const int wm = (wave / 2) * 16;
+    const int wn = (wave % 2) * 16;
+
+    d4 acc_re = {0, 0, 0, 0};
+    d4 acc_im = {0, 0, 0, 0};
+    const int a_row = lane % 16;     // m within wave tile
+    const int a_k = lane / 16;       // k within 4-step
+    const double tsgn = conj_t ? -1.0 : 1.0;
+
+    for (int kb = 0; kb < K; kb += KCH) {
+        const int kc = min(KCH, K - kb);
+        // stage: thread s loads row k = s/TILE? — use s = tid..: row-major
+        // over [KCH][TILE]; for fixed k the TILE m (or g) slots are
+        // contiguous in global memory → coalesced.
+        for (int s = tid; s < KCH * TILE; s += 256) {
+            int k = s / TILE;
+            int c = s % TILE;
+            double re = 0.0, im = 0.0;
+            if (k < kc && m0 + c < M) {
+                const double* src = T + ((long)(kb + k)) * 2 * M + 2 * (m0 + c);
+                re = src[0];
+                im = src[1];
+            }
+            lT[k][2 * c] = re;
+            lT[k][2 * c + 1] = im;
+            re = 0.0;
+            im = 0.0;
+            if (k < kc && g0 + c < G) {
+                const double* src = X + ((long)(kb + k)) * 2 * G + 2 * (g0 + c);
+                re = src[0];
+                im = src[1];
+            }
+            lX[k][2 * c] = re;
+            lX[k][2 * c + 1] = im;
+        }
+        __syncthreads();
+
+        #pragma unroll
+        for (int ks = 0; ks < KCH; ks += 4) {
+            double tr = lT[ks + a_k][2 * (wm + a_row)];
+            double ti = tsgn * lT[ks + a_k][2 * (wm + a_row) + 1];
+            double xr = lX[ks + a_k][2 * (wn + a_row)];
+            double xi = lX[ks + a_k][2 * (wn + a_row) + 1];
+            // C += (tr + i ti)(xr + i xi)
+            acc_re = __builtin_amdgcn_mfma_f64_16x16x4f64(tr, xr, acc_re, 0, 0, 0);
+            acc_re = __builtin_amdgcn_mfma_f64_16x16x4f64(-ti, xi, acc_re, 0, 0, 0);
+            acc_im = __builtin_amdgcn_mfma_f64_16x16x4f64(tr, xi, acc_im, 0, 0, 0);
+            acc_im = __builtin_amdgcn_mfma_f64_16x16x4f64(ti, xr, acc_im, 0, 0, 0);
+        }
+        __syncthreads();
+    }
+
+    const int c_col = lane % 16;
+    const int c_rowg = lane / 16;
+    #pragma unroll
+    for (int v = 0; v < 4; v++) {
+        int m = m0 + wm + 4 * v + c_rowg;
+        long g = g0 + wn + c_col;
+        if (m < M && g < G) {
+            double* dst = C + (long)m * 2 * G + 2 * g;
+            double re = alpha * acc_re[v];
+            double im = alpha * acc_im[v];
+            if (accumulate) {
+                dst[0] += re;
+                dst[1] += im;
+            } else {
+                dst[0] = re;
+                dst[1] = im;
+            }
+        }
+    }
+}
+
+void ztrans(torch::Tensor T, torch::Tensor X, torch::Tensor C,
+            bool conj_t, double alpha, bool accumulate) {
+    TORCH_CHECK(T.is_cuda() && X.is_cuda() && C.is_cuda(), "device tensors required");
+    TORCH_CHECK(T.is_contiguous() && X.is_contiguous() && C.is_contiguous(),
+                "contiguous required");
+    const int K = T.size(0);
+    const int M = T.size(1);
+    const long G = X.size(1);
+    TORCH_CHECK(X.size(0) == K, "K mismatch");
+    TORCH_CHECK(C.size(0) == M && C.size(1) == G, "C shape mismatch");
+    if (M == 0 || G == 0) {
+        return;
+    }
+    if (K == 0) {
+        if (!accumulate) {
+            C.zero_();
+        }
+        return;
+    }
+    dim3 grid((M + TILE - 1) / TILE, (G + TILE - 1) / TILE, 1);
+    auto stream = at::hip::getCurrentHIPStream().stream();
+    hipLaunchKernelGGL(ztrans_kernel, grid, dim3(256), 0, stream,
+                       (const double*)T.data_ptr(), (const double*)X.data_ptr(),
+                       (double*)C.data_ptr(), M, G, K,
+                       conj_t ? 1 : 0, alpha, accumulate ? 1 : 0);
+}
+
 torch::Tensor zgram(torch::Tensor A, torch::Tensor B, int split_k) {
     TORCH_CHECK(A.is_cuda() && B.is_cuda(), "device tensors required");
     TORCH_CHECK(A.is_contiguous() && B.is_contiguous(), "contiguous required");
@@ -145,4 +271,8 @@ torch::Tensor zgram(torch::Tensor A, torch::Tensor B, int split_k) {
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("zgram", &zgram, "C[m,n] = sum_k conj(A[m,k]) B[n,k] (MFMA f64)",
           py::arg("A"), py::arg("B"), py::arg("split_k") = 0);
+    m.def("ztrans", &ztrans,
+          "C[m,g] (+)= alpha * sum_k op(T[k,m]) X[k,g] (MFMA f64)",
+          py::arg("T"), py::arg("X"), py::arg("C"), py::arg("conj_t") = false,
+          py::arg("alpha") = 1.0, py::arg("accumulate") = false);
 }

@@ -115,6 +115,38 @@ def test_zgram_mfma():
 
 
 @requires_gpu
+def test_ztrans_mfma():
+    """MFMA transform kernel vs torch: C (+)= alpha · Tᵀ X."""
+    from sirius_amd import ops
+
+    z = ops.get_zgemm(required=True)
+    torch.manual_seed(2)
+    for (K, M, G) in [(4, 4, 64), (306, 153, 18277), (37, 61, 4099)]:
+        T = torch.view_as_complex(
+            torch.randn(K, M, 2, dtype=torch.float64, device="cuda:0"))
+        X = torch.view_as_complex(
+            torch.randn(K, G, 2, dtype=torch.float64, device="cuda:0"))
+        C = torch.empty(M, G, dtype=torch.complex128, device="cuda:0")
+        z.ztrans(T, X, C, False, 1.0, False)
+        ref = T.transpose(0, 1) @ X
+        assert (C - ref).abs().max() / ref.abs().max() < 1e-12
+        # accumulate with alpha
+        C0 = torch.view_as_complex(
+            torch.randn(M, G, 2, dtype=torch.float64, device="cuda:0"))
+        C = C0.clone()
+        z.ztrans(T, X, C, False, -2.5, True)
+        assert torch.allclose(C, C0 - 2.5 * ref, atol=1e-10)
+        # conj(T)
+        z.ztrans(T, X, C, True, 1.0, False)
+        assert torch.allclose(C, T.conj().transpose(0, 1) @ X, atol=1e-10)
+    # la.transform wrapper routing
+    from sirius_amd.core import la
+
+    got = la.transform(T, X)
+    assert torch.allclose(got, T.transpose(0, 1) @ X, atol=1e-10)
+
+
+@requires_gpu
 def test_scf_gpu_matches_cpu():
     """Full SCF on device: total energy equals the CPU torch-reference run."""
     from sirius_amd.models.synthetic import make_context
