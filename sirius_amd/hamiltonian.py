@@ -292,8 +292,11 @@ class HamiltonianK:
             v0 = float(self.h0.veff_r_coarse[s].mean())
             d = self.ekin + v0
             if self.bp.num_beta_total:
-                bD = self.D[s] @ self.bp.beta_t  # [nbf, nGk]
-                d = d + torch.einsum("ig,ig->g", bD, self.bp.beta_t.conj()).real
+                bt = self.bp.beta_t
+                bD = la.transform(self.D[s].transpose(0, 1), bt)  # [nbf, nGk]
+                # Σ_i bD[i,g]·conj(bt[i,g]) without materializing conj(bt)
+                d = d + (torch.einsum("ig,ig->g", bD.real, bt.real)
+                         + torch.einsum("ig,ig->g", bD.imag, bt.imag))
             return d
         if self.ctx.nc_magnetism:
             return torch.cat([diag_for(0), diag_for(1)])
@@ -304,14 +307,18 @@ class HamiltonianK:
             outs = []
             for b in (0, 1):
                 d = torch.ones_like(self.ekin)
-                bQ = self.Q_blocks[b] @ self.bp.beta_t
-                d = d + torch.einsum("ig,ig->g", bQ, self.bp.beta_t.conj()).real
+                bt = self.bp.beta_t
+                bQ = la.transform(self.Q_blocks[b].transpose(0, 1), bt)
+                d = d + (torch.einsum("ig,ig->g", bQ.real, bt.real)
+                         + torch.einsum("ig,ig->g", bQ.imag, bt.imag))
                 outs.append(d)
             return torch.cat(outs)
         d = torch.ones_like(self.ekin)
         if self.Q is not None:
-            bQ = self.Q @ self.bp.beta_t
-            d = d + torch.einsum("ig,ig->g", bQ, self.bp.beta_t.conj()).real
+            bt = self.bp.beta_t
+            bQ = la.transform(self.Q.transpose(0, 1), bt)
+            d = d + (torch.einsum("ig,ig->g", bQ.real, bt.real)
+                     + torch.einsum("ig,ig->g", bQ.imag, bt.imag))
         if self.ctx.nc_magnetism:
             return torch.cat([d, d])
         return d
