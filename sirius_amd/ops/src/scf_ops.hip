@@ -47,8 +47,9 @@ __global__ void pack_sphere_kernel(const double2* __restrict__ coeff,
 }
 
 // ---------------------------------------------------------------------------
-// unpack + kinetic: out[b,g] = grid[b, idx[g]] + ekin[g]*psi[b,g]
-// (ekin = ½|G+k|² precomputed on host side)
+// unpack + kinetic: out[b,g] = scale*grid[b, idx[g]] + ekin[g]*psi[b,g]
+// (ekin = ½|G+k|² precomputed on host side; `scale` folds the 1/N FFT
+// normalization in — the unscaled fftn avoids a full-grid scale kernel)
 // (fusion of SpFFT unpack with add_to_hphi_pw, local_operator.cu:32-60)
 // ---------------------------------------------------------------------------
 __global__ void unpack_add_kinetic_kernel(const double2* __restrict__ grid,
@@ -56,7 +57,8 @@ __global__ void unpack_add_kinetic_kernel(const double2* __restrict__ grid,
                                           const double* __restrict__ ekin,
                                           const double2* __restrict__ psi,
                                           double2* __restrict__ out,
-                                          long ng, long grid_size, int nb) {
+                                          long ng, long grid_size, int nb,
+                                          double scale) {
     for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < ng * nb;
          i += (long)gridDim.x * blockDim.x) {
         int b = i / ng;
@@ -64,20 +66,22 @@ __global__ void unpack_add_kinetic_kernel(const double2* __restrict__ grid,
         double2 v = grid[(long)b * grid_size + idx[g]];
         double2 p = psi[i];
         double t = ekin[g];
-        out[i] = make_double2(v.x + t * p.x, v.y + t * p.y);
+        out[i] = make_double2(scale * v.x + t * p.x, scale * v.y + t * p.y);
     }
 }
 
-// plain unpack: out[b,g] = grid[b, idx[g]]
+// plain unpack: out[b,g] = scale*grid[b, idx[g]]
 __global__ void unpack_sphere_kernel(const double2* __restrict__ grid,
                                      const long* __restrict__ idx,
                                      double2* __restrict__ out,
-                                     long ng, long grid_size, int nb) {
+                                     long ng, long grid_size, int nb,
+                                     double scale) {
     for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < ng * nb;
          i += (long)gridDim.x * blockDim.x) {
         int b = i / ng;
         long g = i - (long)b * ng;
-        out[i] = grid[(long)b * grid_size + idx[g]];
+        double2 v = grid[(long)b * grid_size + idx[g]];
+        out[i] = make_double2(scale * v.x, scale * v.y);
     }
 }
 
@@ -214,7 +218,8 @@ void pack_sphere(torch::Tensor coeff, torch::Tensor idx, torch::Tensor grid) {
                        (double2*)grid.data_ptr(), ng, gs, nb);
 }
 
-void unpack_sphere(torch::Tensor grid, torch::Tensor idx, torch::Tensor out) {
+void unpack_sphere(torch::Tensor grid, torch::Tensor idx, torch::Tensor out,
+                   double scale) {
     CHECK_HIP(grid);
     long ng = out.size(-1);
     int nb = out.numel() / ng;
@@ -222,11 +227,11 @@ void unpack_sphere(torch::Tensor grid, torch::Tensor idx, torch::Tensor out) {
     hipLaunchKernelGGL(unpack_sphere_kernel, dim3(grid_1d(ng * nb, 256)), dim3(256), 0,
                        cur_stream(),
                        (const double2*)grid.data_ptr(), idx.data_ptr<long>(),
-                       (double2*)out.data_ptr(), ng, gs, nb);
+                       (double2*)out.data_ptr(), ng, gs, nb, scale);
 }
 
 void unpack_add_kinetic(torch::Tensor grid, torch::Tensor idx, torch::Tensor gk2,
-                        torch::Tensor psi, torch::Tensor out) {
+                        torch::Tensor psi, torch::Tensor out, double scale) {
     CHECK_HIP(grid);
     long ng = psi.size(-1);
     int nb = psi.numel() / ng;
@@ -235,7 +240,7 @@ void unpack_add_kinetic(torch::Tensor grid, torch::Tensor idx, torch::Tensor gk2
                        dim3(256), 0, cur_stream(),
                        (const double2*)grid.data_ptr(), idx.data_ptr<long>(),
                        gk2.data_ptr<double>(), (const double2*)psi.data_ptr(),
-                       (double2*)out.data_ptr(), ng, gs, nb);
+                       (double2*)out.data_ptr(), ng, gs, nb, scale);
 }
 
 void mul_veff(torch::Tensor grid, torch::Tensor veff) {
@@ -275,9 +280,12 @@ torch::Tensor residual_precond(torch::Tensor hpsi, torch::Tensor spsi,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("pack_sphere", &pack_sphere, "scatter sphere coeffs into dense FFT grid");
-    m.def("unpack_sphere", &unpack_sphere, "gather sphere coeffs from dense FFT grid");
+    m.def("unpack_sphere", &unpack_sphere, "gather sphere coeffs from dense FFT grid",
+          py::arg("grid"), py::arg("idx"), py::arg("out"), py::arg("scale") = 1.0);
     m.def("unpack_add_kinetic", &unpack_add_kinetic,
-          "gather + add 0.5|G+k|^2 psi (fused local-op epilogue)");
+          "gather + add 0.5|G+k|^2 psi (fused local-op epilogue)",
+          py::arg("grid"), py::arg("idx"), py::arg("gk2"), py::arg("psi"),
+          py::arg("out"), py::arg("scale") = 1.0);
     m.def("mul_veff", &mul_veff, "in-place psi(r) *= V(r)");
     m.def("density_acc", &density_acc, "rho(r) += sum_b w_b |psi_b(r)|^2");
     m.def("residual_precond", &residual_precond,
