@@ -349,23 +349,32 @@ class UnitCell:
         return float(sum(self.atom_types[lab].zn for lab, _ in self.atoms))
 
     def nearest_neighbours(self, r_cut: float):
-        """(i, j, distance, shift) pairs within r_cut, for the Ewald real-space
+        """(i, j, distance) pairs within r_cut, for the Ewald real-space
         sum (reference: unit_cell nearest-neighbour list used in energy.cpp:52-60)."""
+        return [(ia, ja, d) for ia, ja, d, _, _ in
+                self.nearest_neighbours_full(r_cut)]
+
+    def nearest_neighbours_full(self, r_cut: float):
+        """(ia, ja, distance, T_int [3], rc [3]) within r_cut; rc is the
+        Cartesian connecting vector pos(ja)+T·A−pos(ia) and T the integer
+        lattice translation (reference: Unit_cell::find_nearest_neighbours,
+        unit_cell.cpp:389-437 — nnd.translation and nnd.rc). Needed by the
+        Ewald force/stress real-space sums."""
         latt = self.lattice
         pos = self.atom_positions_cart()
-        # enough periodic images to cover r_cut
         inv_len = np.linalg.norm(np.linalg.inv(latt), axis=0)  # 1/interplanar dist
         nmax = np.ceil(r_cut * inv_len).astype(int) + 1
-        shifts = np.array([(i, j, k)
-                           for i in range(-nmax[0], nmax[0] + 1)
-                           for j in range(-nmax[1], nmax[1] + 1)
-                           for k in range(-nmax[2], nmax[2] + 1)], dtype=np.float64)
-        svec = shifts @ latt
+        tint = np.array([(i, j, k)
+                         for i in range(-nmax[0], nmax[0] + 1)
+                         for j in range(-nmax[1], nmax[1] + 1)
+                         for k in range(-nmax[2], nmax[2] + 1)], dtype=np.int64)
+        svec = tint.astype(np.float64) @ latt
         pairs = []
         for ia in range(self.num_atoms):
             for ja in range(self.num_atoms):
-                d = np.linalg.norm(pos[ja] + svec - pos[ia], axis=1)
-                sel = (d > 1e-8) & (d < r_cut)
-                for dd in d[sel]:
-                    pairs.append((ia, ja, dd))
+                rc = pos[ja] + svec - pos[ia]
+                d = np.linalg.norm(rc, axis=1)
+                sel = np.nonzero((d > 1e-8) & (d < r_cut))[0]
+                for s in sel:
+                    pairs.append((ia, ja, d[s], tint[s], rc[s]))
         return pairs

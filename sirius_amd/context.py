@@ -100,6 +100,96 @@ class RadialIntegralsCache:
             self._core[lab] = RITable(q, vals)
         return self._core[lab]
 
+    # -- derivative tables for forces/stress (reference: *_djl_ members,
+    # simulation_context.cpp:1055-1100) -----------------------------------
+
+    def _cached(self, store: str, lab: str, build):
+        d = getattr(self, store, None)
+        if d is None:
+            d = {}
+            setattr(self, store, d)
+        if lab not in d:
+            d[lab] = build()
+        return d[lab]
+
+    def rho_pseudo(self, lab):
+        """Pseudo-atom total-density form factor (Radial_integrals_rho_pseudo,
+        radial_integrals.cpp:133-158; np=20, simulation_context.cpp:1076)."""
+        from .core.radial import RITable, make_q_grid, RadialIntegrals
+
+        def build():
+            at = self.ctx.unit_cell.atom_types[lab]
+            q = make_q_grid(self.ctx.pw_cutoff, 20)
+            if at.rho_total_4pir2.size:
+                vals = RadialIntegrals.rho_q(at.r, at.rho_total_4pir2, q)
+            else:
+                vals = np.zeros_like(q)
+            return RITable(q, vals)
+        return self._cached("_ps_rho", lab, build)
+
+    def vloc_djl(self, lab):
+        from .core.radial import VlocDqTable
+
+        def build():
+            s = self.ctx.cfg.settings
+            return VlocDqTable(self.ctx.unit_cell.atom_types[lab],
+                               self.ctx.pw_cutoff, s.nprii_vloc,
+                               r_cut=s.pseudo_grid_cutoff)
+        return self._cached("_vloc_djl", lab, build)
+
+    def rho_core_djl(self, lab):
+        from .core.radial import RITable, make_q_grid, RadialIntegrals
+
+        def build():
+            at = self.ctx.unit_cell.atom_types[lab]
+            q = make_q_grid(self.ctx.pw_cutoff, self.ctx.cfg.settings.nprii_rho_core)
+            vals = RadialIntegrals.sbessel_dq_transform(0, at.r, at.rho_core_r,
+                                                        q, rpow=2)
+            return RITable(q, vals)
+        return self._cached("_core_djl", lab, build)
+
+    def beta_djl(self, lab):
+        """d/dq of the beta form factors [n_beta_radial, nq]."""
+        from .core.radial import RITable, make_q_grid, RadialIntegrals
+
+        def build():
+            at = self.ctx.unit_cell.atom_types[lab]
+            q = make_q_grid(self.ctx.gk_cutoff, self.ctx.cfg.settings.nprii_beta)
+            vals = np.stack([
+                RadialIntegrals.sbessel_dq_transform(b.l, at.r, b.f_r, q, rpow=1)
+                for b in at.beta]) if at.num_beta else np.zeros((0, len(q)))
+            return RITable(q, vals)
+        return self._cached("_beta_djl", lab, build)
+
+    def aug_djl(self, lab):
+        """d/dq of the augmentation form factors [n_rf_pairs, 2lmax+1, nq]."""
+        from .core.radial import RITable, make_q_grid, RadialIntegrals
+
+        def build():
+            at = self.ctx.unit_cell.atom_types[lab]
+            q = make_q_grid(self.ctx.pw_cutoff, self.ctx.cfg.settings.nprii_aug)
+            qmap = {}
+            for qq in at.q_radial:
+                i, j = min(qq.i, qq.j), max(qq.i, qq.j)
+                qmap[(i, j, qq.l)] = qq.f_r
+            nbrf = at.num_beta
+            lmax3 = 2 * max((b.l for b in at.beta), default=0)
+            vals = np.zeros((nbrf * (nbrf + 1) // 2, lmax3 + 1, len(q)))
+            for j in range(nbrf):
+                lj = at.beta[j].l
+                for i in range(j + 1):
+                    li = at.beta[i].l
+                    pair = j * (j + 1) // 2 + i
+                    for l3 in range(abs(li - lj), min(li + lj, lmax3) + 1):
+                        if (li + lj + l3) % 2 != 0:
+                            continue
+                        f = qmap.get((i, j, l3))
+                        if f is not None:
+                            vals[pair, l3] = RadialIntegrals.sbessel_dq_transform(
+                                l3, at.r, f, q, rpow=0)
+            return RITable(q, vals)
+        return self._cached("_aug_djl", lab, build)
+
 
 class SimulationContext:
     def __init__(self, cfg: Config, unit_cell: UnitCell | None = None,

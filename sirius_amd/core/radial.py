@@ -137,6 +137,18 @@ class RadialIntegrals:
         return spline_integrate(r, integrand, axis=0)
 
     @staticmethod
+    def sbessel_dq_transform(l: int, r: np.ndarray, fr: np.ndarray,
+                             q: np.ndarray, rpow: int = 2) -> np.ndarray:
+        """∫ fr(r)·[d j_l(q r)/dq]·r^rpow dr = ∫ fr·r·j_l'(qr)·r^rpow dr
+        (jl_deriv=true branch of the radial integral generators,
+        radial_integrals.cpp:51-56/104/182/224). Used by stress."""
+        q = np.asarray(q, dtype=np.float64)
+        x = np.outer(r, q)
+        djl = sbessel_dx(l, x)
+        integrand = djl * (fr * r ** (rpow + 1))[:, None]
+        return spline_integrate(r, integrand, axis=0)
+
+    @staticmethod
     def vloc_q(r: np.ndarray, vloc_r: np.ndarray, zn: float, q: np.ndarray,
                r_cut: float = 10.0) -> np.ndarray:
         """Local-potential form factor (without 4π/Ω and without structure phase).
@@ -190,3 +202,51 @@ class RadialIntegrals:
         if (~nz).any():
             out[~nz] = spline_integrate(r, rho4pir2)
         return out / (4 * np.pi)
+
+
+def sbessel_dx(l: int, x: np.ndarray) -> np.ndarray:
+    """d j_l(x)/dx, stable at x→0 (j0'(0)=0, j1'(0)=1/3, else 0).
+
+    Reference: Spherical_Bessel_functions::deriv_q (src/core/sf/
+    sbessel.cpp:57-77) builds d j_l(q r)/dq = r·j_l'(q r)."""
+    from scipy.special import spherical_jn
+
+    x = np.asarray(x, dtype=np.float64)
+    small = np.abs(x) < 1e-9
+    xs = np.where(small, 1.0, x)
+    out = spherical_jn(l, xs, derivative=True)
+    out[small] = (1.0 / 3.0) if l == 1 else 0.0
+    return out
+
+
+class VlocDqTable:
+    """d/dq of the local-potential form factor (jl_deriv branch of
+    Radial_integrals_vloc, radial_integrals.cpp:275-283 and value()
+    radial_integrals.hpp:393-407):
+      raw(q)  = ∫ (r·V + Z·erf r)(sin(qr) − qr·cos(qr)) dr
+      value   = raw/q³ − Z·e^{−q²/4}·(4+q²)/(2q⁴);   0 at q=0.
+    Used by the stress tensor (vloc term)."""
+
+    def __init__(self, at, qmax: float, points_per_au: int, r_cut: float = 10.0):
+        from scipy.special import erf
+
+        q = make_q_grid(qmax, points_per_au)
+        n = max(int(np.searchsorted(at.r, r_cut, side="right")) - 1, 2)
+        rr, vv = at.r[:n], at.vloc_r[:n]
+        raw = np.zeros_like(q)
+        qr = np.outer(rr, q[1:])
+        ker = np.sin(qr) - qr * np.cos(qr)
+        integrand = (rr * vv + at.zn * erf(rr))[:, None] * ker
+        raw[1:] = spline_integrate(rr, integrand, axis=0)
+        self.table = RITable(q, raw)
+        self.zn = at.zn
+
+    def __call__(self, q: np.ndarray) -> np.ndarray:
+        q = np.asarray(q, dtype=np.float64)
+        out = np.zeros_like(q)
+        nz = q > 1e-12
+        qn = q[nz]
+        q2 = qn * qn
+        out[nz] = (self.table(qn) / (q2 * qn)
+                   - self.zn * np.exp(-q2 / 4) * (4 + q2) / (2 * q2 * q2))
+        return out

@@ -213,6 +213,7 @@ class DFTGroundState:
             with profiler("scf_iteration"):
                 with profiler("Hamiltonian0"):
                     h0 = Hamiltonian0(ctx, self.potential, self.density)
+                    self.h0 = h0  # kept for post-SCF forces (D at diag-time V)
                 with profiler("diagonalize"):
                     bands_converged = diagonalize(ctx, h0, self.kset, itsol_tol)
                 with profiler("occupancies"):
@@ -281,3 +282,14 @@ class DFTGroundState:
             if ctx.num_mag_dims else 0.0,
         }
         return out
+
+    def forces(self, add_scf_corr: bool = True) -> dict:
+        """Post-SCF atomic forces (reference: Force::calc_forces_total;
+        the non-local term uses the D matrices of the LAST diagonalization
+        while vloc/us/core/scf_corr use the final potential — matching the
+        reference's sequencing, where scf_corr compensates exactly that
+        potential lag)."""
+        from .forces import Forces
+
+        f = Forces(self.ctx, self.density, self.potential, self.kset, self.h0)
+        return f.calc_forces_total(add_scf_corr=add_scf_corr)

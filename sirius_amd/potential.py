@@ -83,6 +83,8 @@ class Potential:
         """
         ctx = self.ctx
         g = ctx.gvec_fine
+        # previous V_eff(G) for the SCF force correction (potential.cpp:240-316)
+        veff_old = self.veff_g
 
         # Hartree
         g2 = g.gk2_t.clamp(min=1e-30)
@@ -159,6 +161,8 @@ class Potential:
 
         self.veff_r = self.vloc_r + vha_r + self.vxc_r
         self.veff_g = ctx.fft_fine.to_pw(self.veff_r.to(ctx.dtype))
+        self.dveff_g = (self.veff_g - veff_old if veff_old is not None
+                        else self.veff_g.clone())
         if ctx.hubbard is not None:
             ctx.hubbard.generate_potential()
         return self
@@ -193,10 +197,9 @@ class Potential:
         return self.ctx.integrate_rg_fine(density.rho_r * self.veff_r)
 
 
-def ewald_energy(ctx) -> float:
-    """Ewald energy of point charges Z_a (reference: energy.cpp:18-63)."""
+def ewald_lambda(ctx) -> float:
+    """Adaptive Ewald screening parameter (simulation_context.cpp:130-151)."""
     uc = ctx.unit_cell
-    # adaptive lambda (simulation_context.cpp:130-151)
     lam = 1.0
     gmax = ctx.pw_cutoff
     q = uc.num_electrons
@@ -205,7 +208,14 @@ def ewald_energy(ctx) -> float:
         ub = q * q * math.sqrt(2.0 * lam / (2 * math.pi)) * math.erfc(gmax * math.sqrt(1.0 / (4.0 * lam)))
         if ub >= 1e-8:
             break
-    alpha = lam
+    return lam
+
+
+def ewald_energy(ctx) -> float:
+    """Ewald energy of point charges Z_a (reference: energy.cpp:18-63)."""
+    uc = ctx.unit_cell
+    alpha = ewald_lambda(ctx)
+    q = uc.num_electrons
 
     g = ctx.gvec_fine
     zn = np.array([uc.atom_types[lab].zn for lab, _ in uc.atoms], dtype=np.float64)
