@@ -30,9 +30,15 @@ def packed_index(xi1: int, xi2: int) -> int:
 
 
 class AugmentationOperator:
-    """Per-atom-type Q_ij(G) on the fine G sphere (device tensor)."""
+    """Per-atom-type Q_ij(G) on the fine G sphere (device tensor).
 
-    def __init__(self, ctx, at):
+    With gvec_deriv = ν ∈ {0,1,2} builds instead |G|·dQ_ij(G)/dG_ν with
+    the SAME 4π/Ω prefactor as the base operator
+    (generate_pw_coeffs_gvec_deriv, augmentation_operator.cpp:125-180 —
+    there rlm_dq is the r-scaled gradient, divide_by_r=false; the
+    reference's deriv drops the 1/Ω and restores it in the us stress)."""
+
+    def __init__(self, ctx, at, gvec_deriv: int | None = None):
         self.ctx = ctx
         self.at = at
         uc = ctx.unit_cell
@@ -51,8 +57,14 @@ class AugmentationOperator:
         ri = ctx.ri.aug(at.label)(shells)   # [n_rf_pairs, lmax3+1, nshell]
 
         # R_lm(G-hat) for l<=lmax3, and Gaunt table
-        _, theta, phi = ylm_mod.spherical_coords(g.g_cart)
-        rlm3 = ylm_mod.rlm(lmax3, theta, phi)            # [nG, lmmax3]
+        if gvec_deriv is None:
+            _, theta, phi = ylm_mod.spherical_coords(g.g_cart)
+            rlm3 = ylm_mod.rlm(lmax3, theta, phi)        # [nG, lmmax3]
+        else:
+            rlm3, rlm3_dg = ylm_mod.rlm_and_cart_grad(lmax3, g.g_cart)
+            glen_g = g.gk_len
+            ri_dq = ctx.ri.aug_djl(at.label)(shells)
+            nu = gvec_deriv
         lmax_b = lmax_beta
         gc = gaunt_rrr(lmax_b, lmax_b, lmax3)            # [lmmax1, lmmax2, lmmax3]
 
@@ -76,7 +88,15 @@ class AugmentationOperator:
                     if abs(gcv) < 1e-14:
                         continue
                     l3 = int(l_by_lm[lm3])
-                    acc += ((-1j) ** l3 * gcv) * rlm3[:, lm3] * ri[pair, l3][shell_idx]
+                    if gvec_deriv is None:
+                        acc += ((-1j) ** l3 * gcv) * rlm3[:, lm3] \
+                            * ri[pair, l3][shell_idx]
+                    else:
+                        # v = conj(i^l3)·(|G|·dRlm/dG_ν·f + Rlm·f'·G_ν)
+                        acc += ((-1j) ** l3 * gcv) * (
+                            glen_g * rlm3_dg[:, nu, lm3] * ri[pair, l3][shell_idx]
+                            + rlm3[:, lm3] * ri_dq[pair, l3][shell_idx]
+                            * g.g_cart[:, nu])
                 q_pw[idx12] = pref * acc
         self.q_pw = torch.from_numpy(q_pw).to(ctx.device)
 

@@ -75,3 +75,28 @@ def ylm(lmax: int, theta: np.ndarray, phi: np.ndarray) -> np.ndarray:
             if m > 0:
                 out[:, lm_index(l, -m)] = (-1) ** m * np.conj(out[:, lm_index(l, m)])
     return out
+
+
+def rlm_and_cart_grad(lmax: int, gc):
+    """R_lm(Ĝ) [nG, lmmax] and the true Cartesian gradient
+    ∂R_lm(Ĝ)/∂G_x [nG, 3, lmmax] by central differences
+    (sf::dRlm_dr analogue, divide_by_r=true semantics; rows with
+    |G| < 1e-10 get zero gradient)."""
+    import numpy as _np
+
+    r = _np.linalg.norm(gc, axis=1)
+    _, th, ph = spherical_coords(gc)
+    rl = rlm(lmax, th, ph)
+    grad = _np.zeros((len(gc), 3, rl.shape[1]))
+    h = _np.maximum(1e-6 * r, 1e-9)
+    ok = r > 1e-10
+    for x in range(3):
+        gp = gc.copy()
+        gp[:, x] += h
+        gm = gc.copy()
+        gm[:, x] -= h
+        _, tp, pp = spherical_coords(gp)
+        _, tm, pm = spherical_coords(gm)
+        d = (rlm(lmax, tp, pp) - rlm(lmax, tm, pm)) / (2.0 * h)[:, None]
+        grad[ok, x, :] = d[ok]
+    return rl, grad

@@ -293,3 +293,11 @@ class DFTGroundState:
 
         f = Forces(self.ctx, self.density, self.potential, self.kset, self.h0)
         return f.calc_forces_total(add_scf_corr=add_scf_corr)
+
+    def stress(self) -> dict:
+        """Post-SCF stress tensor (reference: Stress::calc_stress_total)."""
+        from .stress import Stress
+
+        st = Stress(self.ctx, self.density, self.potential, self.kset,
+                    self.h0)
+        return st.calc_stress_total()

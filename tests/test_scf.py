@@ -141,9 +141,10 @@ def test_synthetic_si2_runs():
 
 @requires_reference
 @pytest.mark.slow
-def test15_forces_anchor():
-    """Atomic forces vs the reference's own output (force.cpp parity):
-    LiF PAW with a displaced atom — nonzero vloc/us/nonloc/core/ewald."""
+def test15_forces_stress_anchor():
+    """Forces AND stress vs the reference's own outputs (force.cpp /
+    stress.cpp parity): LiF PAW with a displaced atom — every PP term
+    (vloc/us/nonloc/core/ewald/kin/har/xc) is nonzero here."""
     base = os.path.join(REFERENCE, "verification", "test15")
     cfg = Config.from_json(os.path.join(base, "sirius.json"))
     ctx = SimulationContext(cfg, base_dir=base, device="cpu")
@@ -155,6 +156,25 @@ def test15_forces_anchor():
     fref = np.array(ref["ground_state"]["forces"])
     f = dft.forces()
     assert np.abs(f["total"] - fref).max() < 1e-6, f["total"] - fref
+    sref = np.array(ref["ground_state"]["stress"])
+    st = dft.stress()
+    assert np.abs(st["total"].T - sref).max() < 1e-7
+
+
+@requires_reference
+@pytest.mark.slow
+def test08_stress_anchor():
+    """Stress on Si USPP LDA (diagonal by symmetry)."""
+    base = os.path.join(REFERENCE, "verification", "test08")
+    cfg = Config.from_json(os.path.join(base, "sirius.json"))
+    ctx = SimulationContext(cfg, base_dir=base, device="cpu")
+    kset = KPointSet(ctx)
+    dft = DFTGroundState(kset).initial_state()
+    res = dft.find()
+    ref = json.load(open(os.path.join(base, "output_ref.json")))
+    sref = np.array(ref["ground_state"]["stress"])
+    st = dft.stress()
+    assert np.abs(st["total"].T - sref).max() < 1e-7
 
 
 @pytest.mark.skipif(not os.environ.get("SIRIUS_AMD_FULL_TESTS"),
