@@ -164,6 +164,30 @@ def test_scf_gpu_matches_cpu():
 
 
 @requires_gpu
+def test_forces_stress_gpu_match_cpu():
+    """Forces and stress computed on device equal the CPU torch reference."""
+    import numpy as np
+    from sirius_amd.models.synthetic import make_context
+    from sirius_amd.kpoint import KPointSet
+    from sirius_amd.dft import DFTGroundState
+
+    res = {}
+    for dev in ("cuda:0", "cpu"):
+        ctx = make_context(natoms=2, gk_cutoff=4.0, pw_cutoff=10.0, device=dev)
+        ctx.unit_cell.atoms[1] = (ctx.unit_cell.atoms[1][0],
+                                  ctx.unit_cell.atoms[1][1]
+                                  + np.array([0.02, 0.0, 0.0]))
+        ctx._phase_pos = {}
+        ctx.symmetry = None
+        kset = KPointSet(ctx)
+        dft = DFTGroundState(kset).initial_state()
+        dft.find(num_dft_iter=12)
+        res[dev] = (dft.forces()["total"], dft.stress()["total"])
+    assert np.abs(res["cuda:0"][0] - res["cpu"][0]).max() < 1e-7
+    assert np.abs(res["cuda:0"][1] - res["cpu"][1]).max() < 1e-7
+
+
+@requires_gpu
 def test_native_ops_loaded():
     """The in-tree HIP extension (not a fallback) is what runs on GPU."""
     from sirius_amd import ops
