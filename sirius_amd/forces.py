@@ -267,12 +267,54 @@ class Forces:
         return _sym_forces(ctx, out)
 
     def calc_forces_hubbard(self) -> np.ndarray:
+        """−Re Σ_k [V_loc·dn + Σ_pairs e^{−2πik·T}·conj(dn)·V_nl]
+        (calc_forces_hubbard + hubbard_force_add_k_contribution_collinear,
+        force.cpp:270-303, :591-653)."""
         ctx = self.ctx
-        if ctx.hubbard is not None:
-            raise NotImplementedError(
-                "Hubbard forces (compute_occupancies_derivatives) not yet "
-                "implemented")
-        return np.zeros((ctx.unit_cell.num_atoms, 3))
+        hub = ctx.hubbard
+        uc = ctx.unit_cell
+        out = np.zeros((uc.num_atoms, 3))
+        if hub is None or hub.num_wf == 0:
+            return out
+        if ctx.nc_magnetism:
+            raise NotImplementedError("Hubbard forces: collinear only "
+                                      "(matches reference)")
+        from .hamiltonian import HamiltonianK
+        from .hubbard import compute_occupancies_derivatives
+
+        nsp = ctx.num_spins
+        for kp in self.kset:
+            hk = HamiltonianK(self.h0, kp)
+            dn = compute_occupancies_derivatives(hub, kp, hk)
+            d = torch.zeros(3, uc.num_atoms, dtype=dn.dtype, device=dn.device)
+            for il, (ia1, o) in enumerate(hub.levels):
+                if not hub.level_use[il]:
+                    continue
+                off = hub.offsets[il]
+                mm = 2 * o.l + 1
+                for ispn in range(nsp):
+                    d += torch.einsum(
+                        "ab,baxj->xj", hub.um[il][..., ispn],
+                        dn[off:off + mm, off:off + mm, ispn])
+            for i, p in enumerate(hub.nonlocal_pairs):
+                z1 = complex(np.exp(-2j * math.pi
+                                    * float(np.dot(p.T, kp.k_frac))))
+                a1 = hub._find_level(p.ia, p.n1, p.il)
+                a2 = hub._find_level(p.ja, p.n2, p.jl)
+                o1, o2 = hub.offsets[a1], hub.offsets[a2]
+                mi, mj = 2 * p.il + 1, 2 * p.jl + 1
+                for ispn in range(nsp):
+                    d += (z1 * torch.einsum(
+                        "ab,baxj->xj", hub.um_nl[i][..., ispn],
+                        dn[o2:o2 + mj, o1:o1 + mi, ispn].conj())).real \
+                        .to(d.dtype)
+            out -= d.real.cpu().numpy().T
+        comm = get_comm()
+        if comm.active:
+            t = torch.from_numpy(out)
+            comm.allreduce_(t)
+            out = t.numpy()
+        return _sym_forces(ctx, out)
 
     # -- totals ------------------------------------------------------------
 

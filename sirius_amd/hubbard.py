@@ -467,3 +467,284 @@ class HubbardModule:
         if self.ctx.num_spins == 1:
             t *= 2.0
         return t
+
+
+# -- occupation-matrix derivatives (forces & stress) -----------------------
+#
+# Reference: Hubbard::compute_occupancies_derivatives and
+# compute_occupancies_stress_derivatives
+# (hubbard_occupancies_derivatives.cpp:119-567), based on PRB 84, 161102(R)
+# and PRB 102, 235159. Collinear only, like the reference.
+
+def _inv_sqrt_deriv(Op: torch.Tensor, v: torch.Tensor, w: torch.Tensor):
+    """d(O^{-1/2}) from dO (compute_inv_sqrt_O_deriv,
+    hubbard_occupancies_derivatives.cpp:104-117):
+    Õ' = v^H O' v;  Õ'(j,i) /= −(λ_i √λ_j + λ_j √λ_i);  back-transform."""
+    t = v.conj().T @ Op @ v
+    lam = w.real.clamp(min=1e-12)
+    sq = torch.sqrt(lam)
+    denom = -(lam[None, :] * sq[:, None] + lam[:, None] * sq[None, :])
+    t = t / denom.to(t.dtype)
+    return v @ t @ v.conj().T
+
+
+class _HubDerivSetup:
+    """Per-(k, hk) common quantities for both derivative kinds."""
+
+    def __init__(self, hub, kp, hk):
+        ctx = hub.ctx
+        from .dft import atomic_orbitals
+
+        self.hub = hub
+        self.ctx = ctx
+        self.kp = kp
+        self.hk = hk
+        self.phi = atomic_orbitals(ctx, kp)            # [nawf, nGk]
+        self.sphi = hub._apply_S(self.phi, hk)
+        self.nawf = self.phi.shape[0]
+        self.ortho = hub.subspace_method == "full_orthogonalization"
+        if self.ortho:
+            O = la.inner(self.phi, self.sphi)
+            O = 0.5 * (O + O.conj().T)
+            self.w_O, self.v_O = la.eigh(O)
+            wc = self.w_O.real.clamp(min=1e-12)
+            self.inv_sqrt_O = (self.v_O / torch.sqrt(wc)) @ self.v_O.conj().T
+        self.swf_hub = hub.hubbard_wf_S(kp, hk)        # [nhwf, nGk]
+        min_occ = ctx.cfg.iterative_solver.min_occupancy
+        self.psi, self.occw, self.psi_s_phi_hub = [], [], []
+        self.phi_s_psi, self.b_psi = [], []
+        for ispn in range(ctx.num_spins):
+            occ = kp.occ[ispn]
+            sel = np.nonzero(occ > min_occ)[0]
+            psi = kp.psi[ispn][torch.from_numpy(sel).to(ctx.device)].contiguous()
+            self.psi.append(psi)
+            self.occw.append(torch.from_numpy(occ[sel]).to(ctx.device))
+            # ⟨ψ|S|φ_hub⟩ [nocc, nhwf]
+            self.psi_s_phi_hub.append(la.inner(psi, self.swf_hub))
+            if self.ortho:
+                self.phi_s_psi.append(la.inner(self.sphi, psi))
+            self.b_psi.append(hk.bp.inner(psi) if hk.bp.num_beta_total
+                              else None)
+        self.b_phi = hk.bp.inner(self.phi) if hk.bp.num_beta_total else None
+
+    def hub_rows(self):
+        """(hub_offset, atomic_row_start, mm) per level (row map between the
+        hubbard subspace and the full atomic-wf set)."""
+        hub = self.hub
+        uc = self.ctx.unit_cell
+        ao_offset = {}
+        off = 0
+        for ia, (lab, _) in enumerate(uc.atoms):
+            at = uc.atom_types[lab]
+            ao_offset[ia] = off
+            off += sum(2 * w.l + 1 for w in at.atomic_wfs)
+        out = []
+        for il, (ia, o) in enumerate(hub.levels):
+            at = uc.atom_types[uc.atoms[ia][0]]
+            woff = ao_offset[ia]
+            for w in at.atomic_wfs:
+                if w.l == o.l and (o.n < 0 or w.n < 0 or w.n == o.n):
+                    out.append((hub.offsets[il], woff, 2 * o.l + 1))
+                    break
+                woff += 2 * w.l + 1
+        self.ao_offset = ao_offset
+        return out
+
+    def atom_rows(self, ia):
+        """(start, count) of atom ia's rows in the atomic-wf set."""
+        uc = self.ctx.unit_cell
+        at = uc.atom_types[uc.atoms[ia][0]]
+        return self.ao_offset[ia], sum(2 * w.l + 1 for w in at.atomic_wfs)
+
+    def build_deriv(self, ispn, Op, pds_psi, hub_rows):
+        """d⟨φ_hub S|ψ⟩ (build_phi_hub_s_psi_deriv,
+        hubbard_occupancies_derivatives.cpp:45-100); Op holds d(O^{-1/2})
+        when orthogonalizing."""
+        nocc = self.psi[ispn].shape[0]
+        nhwf = self.hub.num_wf
+        out = torch.zeros(nhwf, nocc, dtype=self.phi.dtype,
+                          device=self.phi.device)
+        for (ho, ao, mm) in hub_rows:
+            if self.ortho:
+                out[ho:ho + mm] += (Op[:, ao:ao + mm].conj().T
+                                    @ self.phi_s_psi[ispn])
+                out[ho:ho + mm] += (self.inv_sqrt_O[:, ao:ao + mm].conj().T
+                                    @ pds_psi)
+            else:
+                out[ho:ho + mm] += pds_psi[ao:ao + mm]
+        return out
+
+    def accumulate(self, ispn, deriv, dn_slice):
+        """dn += w_k·(D·⟨ψ|S|φ_hub⟩ + h.c.), D = occ-scaled deriv
+        (update_density_matrix_deriv)."""
+        D = deriv * self.occw[ispn][None, :].to(deriv.dtype)
+        blk = self.kp.weight * (D @ self.psi_s_phi_hub[ispn])
+        dn_slice += blk + blk.conj().T
+
+
+def compute_occupancies_derivatives(hub, kp, hk) -> torch.Tensor:
+    """dn[m1, m2, ispn, x, ja] — occupation derivative w.r.t. displacing
+    atom ja along Cartesian x (hubbard_occupancies_derivatives.cpp:119-360,
+    this k-point's contribution, weight included)."""
+    ctx = hub.ctx
+    uc = ctx.unit_cell
+    st = _HubDerivSetup(hub, kp, hk)
+    hub_rows = st.hub_rows()
+    nsp = ctx.num_spins
+    dn = torch.zeros(hub.num_wf, hub.num_wf, nsp, 3, uc.num_atoms,
+                     dtype=ctx.dtype, device=ctx.device)
+    gkc = torch.from_numpy(kp.gkvec.gkvec_cart).to(ctx.device)
+    bp = hk.bp
+
+    # gradients of the atomic orbitals: dφ = −i(G+k)_x φ
+    db_phi, sdphi_s_phi, sdphi_s_psi = [], [], []
+    db_psi = [[None] * nsp for _ in range(3)]
+    for x in range(3):
+        fac = ((-1j) * gkc[:, x].to(ctx.dtype))[None, :]
+        dphi = (st.phi * fac).contiguous()
+        sdphi = hub._apply_S(dphi, hk)
+        db_phi.append(la.inner(bp.beta_t, dphi) if bp.num_beta_total else None)
+        sdphi_s_phi.append(la.inner(sdphi, st.phi) if st.ortho else None)
+        sdphi_s_psi.append([la.inner(sdphi, st.psi[ispn])
+                            for ispn in range(nsp)])
+        if bp.num_beta_total:
+            for ispn in range(nsp):
+                db_psi[x][ispn] = la.inner(
+                    bp.beta_t * fac, st.psi[ispn])
+
+    for ja in range(uc.num_atoms):
+        at = uc.atom_types[uc.atoms[ja][0]]
+        if at.num_beta_lm == 0 and st.atom_rows(ja)[1] == 0:
+            continue
+        o, nb = bp.atom_offsets[ja], bp.atom_nbf[ja]
+        Qj = None
+        if hk.Q is not None and at.augment and nb:
+            Qj = hk.Q[o:o + nb, o:o + nb]
+        arow, acnt = st.atom_rows(ja)
+        for x in range(3):
+            if st.ortho:
+                if Qj is not None:
+                    # ⟨φ|dS/dr_ja|φ⟩ = ⟨φ|β⟩Q⟨dβ|φ⟩ + ⟨φ|dβ⟩Q⟨β|φ⟩
+                    A = st.b_phi[o:o + nb]
+                    dA = db_phi[x][o:o + nb]
+                    Op = (A.conj().T @ (Qj @ dA)
+                          + dA.conj().T @ (Qj @ A))
+                else:
+                    Op = torch.zeros(st.nawf, st.nawf, dtype=ctx.dtype,
+                                     device=ctx.device)
+                g = sdphi_s_phi[x][arow:arow + acnt]   # ⟨dφ_ja|S|φ⟩
+                Op[arow:arow + acnt, :] += g
+                Op[:, arow:arow + acnt] += g.conj().T
+                Op = _inv_sqrt_deriv(Op, st.v_O, st.w_O)
+            else:
+                Op = None
+            for ispn in range(nsp):
+                if Qj is not None:
+                    A = st.b_phi[o:o + nb]
+                    dA = db_phi[x][o:o + nb]
+                    pds_psi = (A.conj().T @ (Qj @ db_psi[x][ispn][o:o + nb])
+                               + dA.conj().T @ (Qj @ st.b_psi[ispn][o:o + nb]))
+                else:
+                    pds_psi = torch.zeros(st.nawf, st.psi[ispn].shape[0],
+                                          dtype=ctx.dtype, device=ctx.device)
+                pds_psi[arow:arow + acnt] += \
+                    sdphi_s_psi[x][ispn][arow:arow + acnt]
+                deriv = st.build_deriv(ispn, Op, pds_psi, hub_rows)
+                st.accumulate(ispn, deriv, dn[:, :, ispn, x, ja])
+    return dn
+
+
+def compute_occupancies_stress_derivatives(hub, kp, hk) -> torch.Tensor:
+    """dn[m1, m2, ispn, 3ν+μ] — occupation derivative w.r.t. strain ε_{μν}
+    (hubbard_occupancies_derivatives.cpp:371-560; this k-point's
+    contribution). Uses the strain-derivative atomic orbitals
+    (wavefunction_strain_deriv.hpp:20-85) and strain beta projectors."""
+    from .stress import BetaProjectorsStrain
+    from .core import ylm as ylm_mod
+    from .core.radial import RadialIntegrals
+
+    ctx = hub.ctx
+    uc = ctx.unit_cell
+    st = _HubDerivSetup(hub, kp, hk)
+    hub_rows = st.hub_rows()
+    st.hub_rows()  # ensure ao_offset
+    nsp = ctx.num_spins
+    dn = torch.zeros(hub.num_wf, hub.num_wf, nsp, 9, dtype=ctx.dtype,
+                     device=ctx.device)
+    g = kp.gkvec
+    gc = g.gkvec_cart
+    glen = g.gk_len
+    lmax = max((w.l for at in uc.atom_types.values() for w in at.atomic_wfs),
+               default=0)
+    rl, rl_dg = ylm_mod.rlm_and_cart_grad(lmax, gc)
+    inv_len = np.where(glen > 1e-10, 1.0 / np.maximum(glen, 1e-300), 0.0)
+    bps = BetaProjectorsStrain(ctx, kp) if hk.bp.num_beta_total else None
+
+    # per type: radial wf form factors and q-derivatives at |G+k|
+    f0t, f1t = {}, {}
+    for lab, at in uc.atom_types.items():
+        f0t[lab] = [RadialIntegrals.sbessel_transform(w.l, at.r, w.f_r, glen,
+                                                      rpow=1)
+                    for w in at.atomic_wfs]
+        f1t[lab] = [RadialIntegrals.sbessel_dq_transform(w.l, at.r, w.f_r,
+                                                         glen, rpow=1)
+                    for w in at.atomic_wfs]
+    mk = (g.miller + g.k_frac).astype(np.float64)
+
+    import math as _math
+
+    for nu in range(3):
+        for mu in range(3):
+            x = 3 * nu + mu
+            p = 0.5 if mu == nu else 0.0
+            # strain derivative of all atomic orbitals
+            blocks = []
+            for ia, (lab, tau) in enumerate(uc.atoms):
+                at = uc.atom_types[lab]
+                if not at.atomic_wfs:
+                    continue
+                phase = np.exp(-2j * _math.pi * (mk @ tau))
+                cols = []
+                for iw, w in enumerate(at.atomic_wfs):
+                    z = (-1j) ** w.l * (4 * _math.pi
+                                        / _math.sqrt(uc.omega))
+                    for m in range(-w.l, w.l + 1):
+                        lm = ylm_mod.lm_index(w.l, m)
+                        d1 = f0t[lab][iw] * (gc[:, mu] * rl_dg[:, nu, lm]
+                                             + p * rl[:, lm])
+                        d2 = f1t[lab][iw] * rl[:, lm] * (gc[:, mu]
+                                                         * gc[:, nu] * inv_len)
+                        cols.append(-z * (d1 + d2) * phase)
+                blocks.append(np.stack(cols, axis=0))
+            dphi = torch.from_numpy(np.concatenate(blocks, axis=0)) \
+                .to(ctx.device) if blocks else \
+                torch.zeros(0, g.num_gvec, dtype=ctx.dtype, device=ctx.device)
+            sdphi = hub._apply_S(dphi.contiguous(), hk)
+            # dS(ε)|φ⟩ projections (apply_S_operator_strain_deriv):
+            # ⟨φ|dS|ψ⟩ = ⟨φ|β⟩Q⟨dβ(ε)|ψ⟩ + ⟨φ|dβ(ε)⟩Q⟨β|ψ⟩ over ALL atoms
+            if bps is not None and hk.Q is not None:
+                dbt = bps.beta_t[x].contiguous()
+                db_phi = la.inner(dbt, st.phi)
+                QdB_phi = hk.Q @ db_phi
+                QB_phi = hk.Q @ st.b_phi
+            if st.ortho:
+                Op = torch.zeros(st.nawf, st.nawf, dtype=ctx.dtype,
+                                 device=ctx.device)
+                if bps is not None and hk.Q is not None:
+                    Op += st.b_phi.conj().T @ QdB_phi
+                    Op += db_phi.conj().T @ QB_phi
+                t = la.inner(sdphi, st.phi)     # ⟨dφ|S|φ⟩
+                Op += t + t.conj().T
+                Op = _inv_sqrt_deriv(Op, st.v_O, st.w_O)
+            else:
+                Op = None
+            for ispn in range(nsp):
+                pds_psi = la.inner(sdphi, st.psi[ispn])
+                if bps is not None and hk.Q is not None:
+                    db_psi = la.inner(dbt, st.psi[ispn])
+                    pds_psi = pds_psi + st.b_phi.conj().T @ (hk.Q @ db_psi)
+                    pds_psi = pds_psi + db_phi.conj().T @ (
+                        hk.Q @ st.b_psi[ispn])
+                deriv = st.build_deriv(ispn, Op, pds_psi, hub_rows)
+                st.accumulate(ispn, deriv, dn[:, :, ispn, x])
+    return dn

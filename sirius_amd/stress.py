@@ -412,9 +412,55 @@ class Stress:
         return _sym_stress(ctx, s)
 
     def calc_stress_hubbard(self) -> np.ndarray:
-        if self.ctx.hubbard is not None:
-            raise NotImplementedError("Hubbard stress not yet implemented")
-        return np.zeros((3, 3))
+        """calc_stress_hubbard (stress.cpp:103-203): σ(μ,ν) −= Re[V·dn]/Ω
+        per k, then a global sign flip."""
+        ctx = self.ctx
+        hub = ctx.hubbard
+        uc = ctx.unit_cell
+        s = np.zeros((3, 3))
+        if hub is None or hub.num_wf == 0:
+            return s
+        if ctx.nc_magnetism:
+            raise NotImplementedError("Hubbard stress: collinear only")
+        from .hamiltonian import HamiltonianK
+        from .hubbard import compute_occupancies_stress_derivatives
+
+        nsp = ctx.num_spins
+        for kp in self.kset:
+            hk = HamiltonianK(self.h0, kp)
+            dn = compute_occupancies_stress_derivatives(hub, kp, hk)
+            for d1 in range(3):
+                for d2 in range(3):
+                    x = d1 + 3 * d2
+                    acc = 0.0
+                    for il, (ia1, o) in enumerate(hub.levels):
+                        if not hub.level_use[il]:
+                            continue
+                        off = hub.offsets[il]
+                        mm = 2 * o.l + 1
+                        for ispn in range(nsp):
+                            acc += float(torch.einsum(
+                                "ba,abx->x", hub.um[il][..., ispn],
+                                dn[off:off + mm, off:off + mm, ispn])[x].real)
+                    for i, p in enumerate(hub.nonlocal_pairs):
+                        z1 = complex(np.exp(-2j * math.pi
+                                            * float(np.dot(p.T, kp.k_frac))))
+                        a1 = hub._find_level(p.ia, p.n1, p.il)
+                        a2 = hub._find_level(p.ja, p.n2, p.jl)
+                        o1, o2 = hub.offsets[a1], hub.offsets[a2]
+                        mi, mj = 2 * p.il + 1, 2 * p.jl + 1
+                        for ispn in range(nsp):
+                            acc += float((z1 * torch.einsum(
+                                "ab,bax->x", hub.um_nl[i][..., ispn],
+                                dn[o2:o2 + mj, o1:o1 + mi, ispn].conj())[x])
+                                .real)
+                    s[d1, d2] -= acc / uc.omega
+        comm = get_comm()
+        if comm.active:
+            t = torch.from_numpy(s)
+            comm.allreduce_(t)
+            s = t.numpy()
+        return -1.0 * _sym_stress(ctx, s)
 
     # -- total -------------------------------------------------------------
 

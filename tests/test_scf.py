@@ -203,3 +203,41 @@ def test_forces_finite_difference():
     dE = (ep - em) / (2 * h)
     a1 = ctx.unit_cell.lattice[0]
     assert abs(dE + f["total"][1] @ a1) < 5e-5
+
+
+@requires_reference
+@pytest.mark.skipif(not os.environ.get("SIRIUS_AMD_FULL_TESTS"),
+                    reason="set SIRIUS_AMD_FULL_TESTS=1")
+def test25_hubbard_forces_anchor():
+    """Hubbard-force parity (compute_occupancies_derivatives): NiO +U full
+    Liechtenstein with Löwdin orthogonalization, displaced O atoms."""
+    base = os.path.join(REFERENCE, "verification", "test25")
+    cfg = Config.from_json(os.path.join(base, "sirius.json"))
+    ctx = SimulationContext(cfg, base_dir=base, device="cpu")
+    kset = KPointSet(ctx)
+    dft = DFTGroundState(kset).initial_state()
+    dft.find()
+    ref = json.load(open(os.path.join(base, "output_ref.json")))
+    fref = np.array(ref["ground_state"]["forces"])
+    f = dft.forces()
+    assert np.abs(f["total"] - fref).max() < 1e-5
+
+
+@requires_reference
+@pytest.mark.skipif(not os.environ.get("SIRIUS_AMD_FULL_TESTS"),
+                    reason="set SIRIUS_AMD_FULL_TESTS=1")
+def test22_hubbard_stress_anchor():
+    """Hubbard-stress parity (compute_occupancies_stress_derivatives):
+    NiO +U Dudarev AFM."""
+    base = os.path.join(REFERENCE, "verification", "test22")
+    cfg = Config.from_json(os.path.join(base, "sirius.json"))
+    ctx = SimulationContext(cfg, base_dir=base, device="cpu")
+    kset = KPointSet(ctx)
+    dft = DFTGroundState(kset).initial_state()
+    dft.find()
+    ref = json.load(open(os.path.join(base, "output_ref.json")))
+    st = dft.stress()
+    sref = np.array(ref["ground_state"]["stress"])
+    assert np.abs(st["total"].T - sref).max() < 1e-7
+    f = dft.forces()
+    assert np.abs(f["total"] - np.array(ref["ground_state"]["forces"])).max() < 1e-7
