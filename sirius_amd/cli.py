@@ -8,6 +8,7 @@ self-checks |ΔE_tot| ≤ 1e-5 Ha (sirius.scf.cpp:309-341).
 Usage:
     python -m sirius_amd.cli sirius.json [--device cpu|cuda]
         [--num-iter N] [--test-against output_ref.json] [--output out.json]
+        [--forces] [--stress] [--relax | --vc-relax]
 """
 
 from __future__ import annotations
@@ -28,6 +29,14 @@ def main(argv=None):
     ap.add_argument("--output", default=None)
     ap.add_argument("--override", action="append", default=[],
                     help="section.key=json_value config overrides")
+    ap.add_argument("--forces", action="store_true",
+                    help="compute atomic forces after the SCF")
+    ap.add_argument("--stress", action="store_true",
+                    help="compute the stress tensor after the SCF")
+    ap.add_argument("--relax", action="store_true",
+                    help="relax atomic positions (fixed cell)")
+    ap.add_argument("--vc-relax", action="store_true",
+                    help="relax positions and cell")
     args = ap.parse_args(argv)
 
     from . import Config, SimulationContext, KPointSet, DFTGroundState
@@ -38,6 +47,18 @@ def main(argv=None):
         key, val = ov.split("=", 1)
         cfg.override(key, json.loads(val))
     base = os.path.dirname(os.path.abspath(args.input))
+    if args.relax or args.vc_relax:
+        from .relax import LatticeRelaxation
+
+        rlx = LatticeRelaxation(cfg, base_dir=base, device=args.device,
+                                variable_cell=args.vc_relax)
+        rres = rlx.run()
+        print(json.dumps(rres["history"][-1], indent=2, default=float))
+        path = args.output or f"relax_{int(time.time())}.json"
+        with open(path, "w") as f:
+            json.dump(rres, f, indent=2, default=float)
+        return 0 if rres["converged"] else 1
+
     ctx = SimulationContext(cfg, base_dir=base, device=args.device)
     kset = KPointSet(ctx)
     dft = DFTGroundState(kset).initial_state()
@@ -63,6 +84,12 @@ def main(argv=None):
             "device": str(ctx.device),
         },
     }
+    if args.forces:
+        f = dft.forces()
+        out["ground_state"]["forces"] = f["total"].tolist()
+    if args.stress:
+        st = dft.stress()
+        out["ground_state"]["stress"] = st["total"].T.tolist()
     path = args.output or f"output_{int(time.time())}.json"
     with open(path, "w") as f:
         json.dump(out, f, indent=2)
@@ -76,6 +103,21 @@ def main(argv=None):
         de = abs(etot - eref)
         ok = de < 1e-5
         print(f"test_against: |dE| = {de:.3e} Ha -> {'OK' if ok else 'FAIL'}")
+        import numpy as np
+        if args.forces and "forces" in ref["ground_state"]:
+            df = np.abs(np.array(out["ground_state"]["forces"])
+                        - np.array(ref["ground_state"]["forces"])).max()
+            okf = df < 1e-5
+            ok = ok and okf
+            print(f"test_against forces: max|dF| = {df:.3e} -> "
+                  f"{'OK' if okf else 'FAIL'}")
+        if args.stress and "stress" in ref["ground_state"]:
+            ds = np.abs(np.array(out["ground_state"]["stress"])
+                        - np.array(ref["ground_state"]["stress"])).max()
+            oks = ds < 1e-6
+            ok = ok and oks
+            print(f"test_against stress: max|ds| = {ds:.3e} -> "
+                  f"{'OK' if oks else 'FAIL'}")
         return 0 if ok else 1
     return 0
 
