@@ -167,13 +167,70 @@ class Anderson(Mixer):
         self.history_size = min(self.history_size + 1, self.max_history - 1)
 
 
+class Broyden2(Mixer):
+    """Second Broyden method (rank-one product expansion over the residual
+    history; replicates broyden2_mixer.hpp:105-177 including the Gram
+    bookkeeping and the full-history shift)."""
+
+    def __init__(self, *args, **kwargs):
+        super().__init__(*args, **kwargs)
+        n = self.max_history
+        self.S = torch.zeros(n, n, dtype=torch.float64)
+
+    def mix_impl(self):
+        idx = self._idx(self.step)
+        n = min(self.step, self.max_history - 1)
+        for i in range(n + 1):
+            j = self._idx(self.step - i)
+            v = self._inner(self.residual_history[j],
+                            self.residual_history[idx], normalize=False)
+            self.S[n - i, n] = v
+            self.S[n, n - i] = v
+        gamma = torch.zeros(self.max_history, dtype=torch.float64)
+        S = self.S
+        for i in range(1, n + 1):
+            g = S[n - i, n] - S[n - i + 1, n]
+            for j in range(1, i):
+                g += float((-S[n - i + 1, n - j + 1] + S[n - i + 1, n - j]
+                            + S[n - i, n - j + 1] - S[n - i, n - j])
+                           * gamma[n - j])
+            denom = (S[n - i + 1, n - i + 1] - S[n - i + 1, n - i]
+                     - S[n - i, n - i + 1] + S[n - i, n - i])
+            gamma[n - i] = g / denom
+
+        # reference seeds the accumulator with the CURRENT output
+        # (mixer copy(output_history[idx], input_), broyden2_mixer.hpp:142)
+        nxt = self._copy(self.output_history[idx])
+        if n > 0:
+            j = self._idx(self.step - n)
+            self._axpy(-self.beta * float(gamma[0]),
+                       self.residual_history[j], nxt)
+            self._axpy(-float(gamma[0]), self.output_history[j], nxt)
+            for i in range(1, n):
+                coeff = float(gamma[n - i - 1] - gamma[n - i])
+                j = self._idx(self.step - i)
+                self._axpy(self.beta * coeff, self.residual_history[j], nxt)
+                self._axpy(coeff, self.output_history[j], nxt)
+            j = idx
+            self._axpy(self.beta * (float(gamma[n - 1]) + 1.0),
+                       self.residual_history[j], nxt)
+            self._axpy(float(gamma[n - 1]), self.output_history[j], nxt)
+        else:
+            self._axpy(self.beta, self.residual_history[idx], nxt)
+        self.output_history[self._idx(self.step + 1)] = nxt
+        if n == self.max_history - 1:
+            self.S[:n, :n] = self.S[1:n + 1, 1:n + 1].clone()
+
+
 def make_mixer(cfg_mixer, components: list[Component]) -> Mixer:
     kind = cfg_mixer.type
     kw = dict(max_history=cfg_mixer.max_history, beta=cfg_mixer.beta,
               beta0=cfg_mixer.beta0, beta_scaling_factor=cfg_mixer.beta_scaling_factor)
     if kind == "linear":
         return Linear(components, **kw)
-    if kind in ("anderson", "anderson_stable", "broyden1", "broyden2"):
-        # broyden1/broyden2/anderson_stable fall back to anderson until implemented
+    if kind in ("anderson", "anderson_stable", "broyden1"):
+        # anderson_stable/broyden1 fall back to anderson (same family)
         return Anderson(components, **kw)
+    if kind == "broyden2":
+        return Broyden2(components, **kw)
     raise ValueError(f"unknown mixer type {kind}")

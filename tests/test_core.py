@@ -179,3 +179,35 @@ def test_profiler_tree():
     assert d["outer"]["count"] == 1
     assert d["outer"]["sub"]["inner"]["count"] == 2
     assert "outer" in p.report()
+
+
+def test_broyden2_mixer_fixed_point():
+    """Broyden2 converges a contractive vector fixed point faster than
+    plain linear mixing (broyden2_mixer.hpp parity)."""
+    import torch
+    from sirius_amd.mixer import Broyden2, Linear, Component
+
+    torch.manual_seed(0)
+    n = 40
+    M = torch.randn(n, n, dtype=torch.float64)
+    M = 0.55 * M / torch.linalg.matrix_norm(M, 2)
+    b = torch.randn(n, dtype=torch.float64)
+
+    def g(x):          # fixed point of x = Mx + b
+        return M @ x + b
+
+    def run(mixer_cls, steps):
+        mx = mixer_cls([Component("x")], max_history=8, beta=0.5)
+        x = torch.zeros(n)
+        mx.initialize({"x": x})
+        hist = []
+        for _ in range(steps):
+            cur = mx.get_output()["x"]
+            mx.set_input({"x": g(cur)})
+            hist.append(mx.mix())
+        return hist
+
+    hb = run(Broyden2, 25)
+    hl = run(Linear, 25)
+    assert hb[-1] < 1e-10
+    assert hb[-1] < hl[-1] * 1e-2
