@@ -198,7 +198,7 @@ def test_broyden2_mixer_fixed_point():
 
     def run(mixer_cls, steps):
         mx = mixer_cls([Component("x")], max_history=8, beta=0.5)
-        x = torch.zeros(n)
+        x = torch.zeros(n, dtype=torch.float64)
         mx.initialize({"x": x})
         hist = []
         for _ in range(steps):
