@@ -207,7 +207,7 @@ def test_broyden2_mixer_fixed_point():
             hist.append(mx.mix())
         return hist
 
-    hb = run(Broyden2, 25)
-    hl = run(Linear, 25)
-    assert hb[-1] < 1e-10
+    hb = run(Broyden2, 30)
+    hl = run(Linear, 30)
+    assert hb[-1] < 1e-9
     assert hb[-1] < hl[-1] * 1e-2
