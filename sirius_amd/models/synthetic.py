@@ -98,9 +98,15 @@ def make_synthetic_config(natoms: int = 8, gk_cutoff: float = 5.0,
     return cfg, None
 
 
-def make_context(natoms: int = 8, device: str | None = None, **kwargs):
+def make_context(natoms: int = 8, device: str | None = None,
+                 lattice_scale: float = 1.0, **kwargs):
     from ..context import SimulationContext
 
     cfg, _ = make_synthetic_config(natoms=natoms, **kwargs)
     uc = make_synthetic_cell(natoms)
+    if lattice_scale != 1.0:
+        from ..cell import UnitCell
+
+        uc = UnitCell(uc.lattice * float(lattice_scale), uc.atom_types,
+                      [(lab, pos) for lab, pos in uc.atoms])
     return SimulationContext(cfg, unit_cell=uc, device=device)
