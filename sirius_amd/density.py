@@ -301,6 +301,33 @@ class Density:
                     ia_list, rows = type_rows(bp, lab, nbf)
                     xu = bu[rows].reshape(len(ia_list), nbf, -1)
                     xd = bd[rows].reshape(len(ia_list), nbf, -1)
+                    if at.spin_orbit:
+                        # SO atoms: rotate ⟨β|ψ⟩ with the (same-radial-
+                        # function-masked) f coefficients before forming dm
+                        # (add_k_point_contribution_dm_pwpp_noncollinear,
+                        # density.cpp:945-1005)
+                        from . import so as so_mod
+
+                        fc = ctx.so_fcoef(lab)
+                        idxb = at.beta_lm_index()
+                        irf = np.array([i_ for i_, _, _ in idxb])
+                        m = (irf[:, None] == irf[None, :]).astype(np.float64)
+                        fm = torch.from_numpy(fc * m[:, :, None, None]) \
+                            .to(ctx.device).to(ctx.dtype)
+                        b2u0 = wf * xu.conj()
+                        b2d0 = wf * xd.conj()
+                        b1u = (torch.einsum("ip,apb->aib", fm[:, :, 0, 0], xu)
+                               + torch.einsum("ip,apb->aib", fm[:, :, 0, 1], xd))
+                        b1d = (torch.einsum("ip,apb->aib", fm[:, :, 1, 0], xu)
+                               + torch.einsum("ip,apb->aib", fm[:, :, 1, 1], xd))
+                        b2u = (torch.einsum("pi,apb->aib", fm[:, :, 0, 0], b2u0)
+                               + torch.einsum("pi,apb->aib", fm[:, :, 1, 0], b2d0))
+                        b2d = (torch.einsum("pi,apb->aib", fm[:, :, 0, 1], b2u0)
+                               + torch.einsum("pi,apb->aib", fm[:, :, 1, 1], b2d0))
+                        dm[lab][..., 0] += torch.einsum("aib,ajb->aij", b1u, b2u)
+                        dm[lab][..., 1] += torch.einsum("aib,ajb->aij", b1d, b2d)
+                        dm[lab][..., 2] += torch.einsum("aib,ajb->aij", b1u, b2d)
+                        continue
                     dm[lab][..., 0] += torch.einsum("aib,b,ajb->aij", xu, wf, xu.conj())
                     dm[lab][..., 1] += torch.einsum("aib,b,ajb->aij", xd, wf, xd.conj())
                     dm[lab][..., 2] += torch.einsum("aib,b,ajb->aij", xu, wf, xd.conj())

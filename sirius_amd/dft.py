@@ -224,6 +224,8 @@ class DFTGroundState:
             e1 = self.energy_potential(self.density.rho_r, self.density.mag_r)
             rho1_r = self.density.rho_r.clone()
             mag1_r = self.density.mag_r.clone() if self.density.mag_r is not None else None
+            magv1_r = [m.clone() for m in self.density.magv_r] \
+                if ctx.nc_magnetism else None
 
             with profiler("mix"):
                 rms = self.density.mix()
@@ -247,7 +249,14 @@ class DFTGroundState:
                 self.potential.generate_paw(self.density)
 
             e2 = self.ctx.integrate_rg_fine(rho1_r * self.potential.veff_r)
-            if mag1_r is not None and self.potential.bz_r is not None:
+            if magv1_r is not None and self.potential.bvec_r is not None:
+                # noncollinear: the full vector m·B enters both potential
+                # energies (energy.cpp:251-257) — z alone leaves a spurious
+                # residual in e2−e1 at self-consistency
+                for i in range(3):
+                    e2 += self.ctx.integrate_rg_fine(
+                        magv1_r[i] * self.potential.bvec_r[i])
+            elif mag1_r is not None and self.potential.bz_r is not None:
                 e2 += self.ctx.integrate_rg_fine(mag1_r * self.potential.bz_r)
             if paw:
                 e2 += paw.one_elec_energy(_D)

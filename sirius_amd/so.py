@@ -159,7 +159,14 @@ def so_d_blocks(at, d_alpha: list[np.ndarray], fcoef: np.ndarray) -> list[np.nda
     out = [np.zeros((nbf, nbf), dtype=np.complex128) for _ in range(4)]
     # result(xi1,xi2,σσ') = Σ_{x1p,x2p,α,σ1σ2} d[α,x1p,x2p] P[α,σ1,σ2]
     #                       f[xi1,x1p,σ,σ1] f[x2p,xi2,σ2,σ']
-    # f already vanishes across different (l,j) channels.
+    # The x1p (x2p) sums are restricted to the SAME radial function as
+    # xi1 (xi2) — compare_index_beta_functions demands equal idxrf, not
+    # just equal (l, j) (non_local_operator.cpp:139-144, atom_type.hpp:1188)
+    # — so cross-radial f elements (which the f table does contain for
+    # equal (l,j)) must NOT enter the rotation.
+    irf = np.array([irf_ for irf_, _, _ in idx])
+    m_irf = (irf[:, None] == irf[None, :]).astype(np.float64)
+    fm = fcoef * m_irf[:, :, None, None]
     for s in range(2):
         for sp in range(2):
             acc = np.zeros((nbf, nbf), dtype=np.complex128)
@@ -169,7 +176,7 @@ def so_d_blocks(at, d_alpha: list[np.ndarray], fcoef: np.ndarray) -> list[np.nda
                         p = PAULI[a, s1, s2]
                         if p == 0:
                             continue
-                        acc += p * (fcoef[:, :, s, s1] @ d[a] @ fcoef[:, :, s2, sp])
+                        acc += p * (fm[:, :, s, s1] @ d[a] @ fm[:, :, s2, sp])
             out[S_IDX[s][sp]] += acc
 
     # ionic term: dion over same-am channels with single f factor
@@ -189,12 +196,21 @@ def so_q_blocks(at, q_mtrx: np.ndarray, fcoef: np.ndarray) -> list[np.ndarray]:
     """Rotate the augmentation overlap charges into spin blocks (Eq.18;
     Q_operator::initialize, non_local_operator.cpp:307-340).
 
-    Returns [Q00, Q11, Q01, Q10]."""
+    Returns [Q00, Q11, Q01, Q10]. The primed sums are restricted to the
+    same radial function (compare_index_beta_functions,
+    non_local_operator.cpp:314-325), like the D rotation."""
+    idx = at.beta_lm_index()
     nbf = q_mtrx.shape[0]
+    irf = np.array([irf_ for irf_, _, _ in idx])
+    m_irf = (irf[:, None] == irf[None, :]).astype(np.float64)
+    fm = fcoef * m_irf[:, :, None, None]
     out = [np.zeros((nbf, nbf), dtype=np.complex128) for _ in range(4)]
     for si in range(2):
         for sj in range(2):
-            acc = (fcoef[:, :, sj, 0] @ q_mtrx @ fcoef[:, :, 0, si]
-                   + fcoef[:, :, sj, 1] @ q_mtrx @ fcoef[:, :, 1, si])
-            out[S_IDX[sj][si]] = acc
+            acc = (fm[:, :, sj, 0] @ q_mtrx @ fm[:, :, 0, si]
+                   + fm[:, :, sj, 1] @ q_mtrx @ fm[:, :, 1, si])
+            # ind = si when diagonal else sj+2 == s_idx[si][sj]
+            # (non_local_operator.cpp:330-338; verified against a
+            # brute-force replica of the reference loops)
+            out[S_IDX[si][sj]] = acc
     return out
