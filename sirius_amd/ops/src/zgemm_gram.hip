@@ -57,33 +57,52 @@ __global__ void zgram_splitk_kernel(const double* __restrict__ A,  // [M, 2K] in
     const int a_row = lane % 16;
     const int a_k = lane / 16;          // 0..3
 
-    for (long kb = k_begin; kb < k_end; kb += KCH) {
-        const int kc = (int)min((long)KCH, k_end - kb);
-        // stage A/B rows: thread t loads row t/8, dword pair t%8 → strided
-        // over the chunk; 256 threads cover 32 rows × 8 slots of 2 complex
-        for (int s = tid; s < TILE * KCH; s += 256) {
-            int r = s / KCH;
-            int k = s % KCH;
-            double re = 0.0, im = 0.0;
-            if (k < kc && m0 + r < M) {
-                const double* src = A + ((long)(m0 + r)) * 2 * K + 2 * (kb + k);
-                re = src[0];
-                im = src[1];
-            }
-            lA[r][2 * k] = re;
-            lA[r][2 * k + 1] = im;
-            re = 0.0;
-            im = 0.0;
-            if (k < kc && n0 + r < N) {
-                const double* src = B + ((long)(n0 + r)) * 2 * K + 2 * (kb + k);
-                re = src[0];
-                im = src[1];
-            }
-            lB[r][2 * k] = re;
-            lB[r][2 * k + 1] = im;
-        }
-        __syncthreads();
+    // software pipeline: each thread owns two fixed (row, k) staging slots;
+    // the NEXT chunk is prefetched into registers while MFMAs consume the
+    // LDS-resident one (hides the HBM latency behind the matrix math).
+    const int r0 = tid / KCH, k0 = tid % KCH;
+    const int r1 = (tid + 256) / KCH, k1 = (tid + 256) % KCH;
+    double pa[2][2], pb[2][2];
 
+    auto prefetch = [&](long kb) {
+        const int rr[2] = {r0, r1};
+        const int kk[2] = {k0, k1};
+        const int kc = (int)min((long)KCH, k_end - kb);
+        #pragma unroll
+        for (int j = 0; j < 2; j++) {
+            pa[j][0] = pa[j][1] = pb[j][0] = pb[j][1] = 0.0;
+            if (kk[j] < kc) {
+                if (m0 + rr[j] < M) {
+                    const double* src = A + ((long)(m0 + rr[j])) * 2 * K + 2 * (kb + kk[j]);
+                    pa[j][0] = src[0];
+                    pa[j][1] = src[1];
+                }
+                if (n0 + rr[j] < N) {
+                    const double* src = B + ((long)(n0 + rr[j])) * 2 * K + 2 * (kb + kk[j]);
+                    pb[j][0] = src[0];
+                    pb[j][1] = src[1];
+                }
+            }
+        }
+    };
+    auto commit = [&]() {
+        lA[r0][2 * k0] = pa[0][0];
+        lA[r0][2 * k0 + 1] = pa[0][1];
+        lB[r0][2 * k0] = pb[0][0];
+        lB[r0][2 * k0 + 1] = pb[0][1];
+        lA[r1][2 * k1] = pa[1][0];
+        lA[r1][2 * k1 + 1] = pa[1][1];
+        lB[r1][2 * k1] = pb[1][0];
+        lB[r1][2 * k1 + 1] = pb[1][1];
+    };
+
+    prefetch(k_begin);
+    for (long kb = k_begin; kb < k_end; kb += KCH) {
+        commit();
+        __syncthreads();
+        if (kb + KCH < k_end) {
+            prefetch(kb + KCH);
+        }
         // 4 MFMA k-steps of 4 over the 16-wide chunk
         #pragma unroll
         for (int ks = 0; ks < KCH; ks += 4) {
@@ -109,8 +128,6 @@ __global__ void zgram_splitk_kernel(const double* __restrict__ A,  // [M, 2K] in
         int m = m0 + wm + 4 * v + c_rowg;
         int n = n0 + wn + c_col;
         if (m < M && n < N) {
-            // NOTE: MFMA computes a*b accumulate; row/col mapping of the
-            // A operand puts OUR row index on the output row — C[m,n]
             unsafeAtomicAdd(&C[(long)m * 2 * N + 2 * n], acc_re[v]);
             unsafeAtomicAdd(&C[(long)m * 2 * N + 2 * n + 1], acc_im[v]);
         }
@@ -152,34 +169,51 @@ __global__ void ztrans_kernel(const double* __restrict__ T,  // [K, 2M] interlea
     const int a_k = lane / 16;       // k within 4-step
     const double tsgn = conj_t ? -1.0 : 1.0;
 
-    for (int kb = 0; kb < K; kb += KCH) {
-        const int kc = min(KCH, K - kb);
-        // stage: thread s loads row k = s/TILE? — use s = tid..: row-major
-        // over [KCH][TILE]; for fixed k the TILE m (or g) slots are
-        // contiguous in global memory → coalesced.
-        for (int s = tid; s < KCH * TILE; s += 256) {
-            int k = s / TILE;
-            int c = s % TILE;
-            double re = 0.0, im = 0.0;
-            if (k < kc && m0 + c < M) {
-                const double* src = T + ((long)(kb + k)) * 2 * M + 2 * (m0 + c);
-                re = src[0];
-                im = src[1];
-            }
-            lT[k][2 * c] = re;
-            lT[k][2 * c + 1] = im;
-            re = 0.0;
-            im = 0.0;
-            if (k < kc && g0 + c < G) {
-                const double* src = X + ((long)(kb + k)) * 2 * G + 2 * (g0 + c);
-                re = src[0];
-                im = src[1];
-            }
-            lX[k][2 * c] = re;
-            lX[k][2 * c + 1] = im;
-        }
-        __syncthreads();
+    // software pipeline (same scheme as zgram): 2 fixed staging slots per
+    // thread, next chunk prefetched into registers during the MFMAs.
+    const int k0s = tid / TILE, c0s = tid % TILE;
+    const int k1s = (tid + 256) / TILE, c1s = (tid + 256) % TILE;
+    double pt[2][2], px[2][2];
 
+    auto prefetch = [&](int kb) {
+        const int kks[2] = {k0s, k1s};
+        const int ccs[2] = {c0s, c1s};
+        const int kc = min(KCH, K - kb);
+        #pragma unroll
+        for (int j = 0; j < 2; j++) {
+            pt[j][0] = pt[j][1] = px[j][0] = px[j][1] = 0.0;
+            if (kks[j] < kc) {
+                if (m0 + ccs[j] < M) {
+                    const double* src = T + ((long)(kb + kks[j])) * 2 * M + 2 * (m0 + ccs[j]);
+                    pt[j][0] = src[0];
+                    pt[j][1] = src[1];
+                }
+                if (g0 + ccs[j] < G) {
+                    const double* src = X + ((long)(kb + kks[j])) * 2 * G + 2 * (g0 + ccs[j]);
+                    px[j][0] = src[0];
+                    px[j][1] = src[1];
+                }
+            }
+        }
+    };
+    auto commit = [&]() {
+        lT[k0s][2 * c0s] = pt[0][0];
+        lT[k0s][2 * c0s + 1] = pt[0][1];
+        lX[k0s][2 * c0s] = px[0][0];
+        lX[k0s][2 * c0s + 1] = px[0][1];
+        lT[k1s][2 * c1s] = pt[1][0];
+        lT[k1s][2 * c1s + 1] = pt[1][1];
+        lX[k1s][2 * c1s] = px[1][0];
+        lX[k1s][2 * c1s + 1] = px[1][1];
+    };
+
+    prefetch(0);
+    for (int kb = 0; kb < K; kb += KCH) {
+        commit();
+        __syncthreads();
+        if (kb + KCH < K) {
+            prefetch(kb + KCH);
+        }
         #pragma unroll
         for (int ks = 0; ks < KCH; ks += 4) {
             double tr = lT[ks + a_k][2 * (wm + a_row)];
