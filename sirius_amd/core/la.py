@@ -32,13 +32,26 @@ class _host_threads:
         torch.set_num_threads(self.saved)
 
 
+# NOTE on staging: pinned bounce buffers for these host-solve round trips
+# were tried and measured SLOWER end-to-end (0.265 -> 0.284 s/SCF-iter):
+# the extra host-side copies and the blocking sync on the pinned D2H sit
+# on the critical path, while the pageable copies overlap with queued GPU
+# work. Plain .cpu()/.to() stays.
+def to_host(t: torch.Tensor) -> torch.Tensor:
+    return t.cpu() if t.is_cuda else t
+
+
+def to_dev(t: torch.Tensor, device) -> torch.Tensor:
+    return t if str(device) == "cpu" else t.to(device)
+
+
 def eigh(H: torch.Tensor):
     """Hermitian eigensolve returning (evals, evecs) on H's device."""
     n = H.shape[-1]
     if H.is_cuda and n < EIGH_GPU_MIN:
         with _host_threads():
-            w, v = torch.linalg.eigh(H.cpu())
-        return w.to(H.device, non_blocking=True), v.to(H.device)
+            w, v = torch.linalg.eigh(to_host(H))
+        return to_dev(w, H.device), to_dev(v, H.device)
     return torch.linalg.eigh(H)
 
 
@@ -46,7 +59,7 @@ def cholesky(S: torch.Tensor):
     n = S.shape[-1]
     if S.is_cuda and n < EIGH_GPU_MIN:
         with _host_threads():
-            return torch.linalg.cholesky(S.cpu()).to(S.device)
+            return to_dev(torch.linalg.cholesky(to_host(S)), S.device)
     return torch.linalg.cholesky(S)
 
 
@@ -55,8 +68,9 @@ def inv_lower(L: torch.Tensor) -> torch.Tensor:
     eye = torch.eye(L.shape[-1], dtype=L.dtype, device=L.device)
     if L.is_cuda and L.shape[-1] < EIGH_GPU_MIN:
         with _host_threads():
-            out = torch.linalg.solve_triangular(L.cpu(), eye.cpu(), upper=False)
-        return out.to(L.device)
+            out = torch.linalg.solve_triangular(to_host(L), eye.cpu(),
+                                                upper=False)
+        return to_dev(out, L.device)
     return torch.linalg.solve_triangular(L, eye, upper=False)
 
 
@@ -67,11 +81,11 @@ def ortho_factor(gram: torch.Tensor) -> torch.Tensor:
     n = gram.shape[-1]
     if gram.is_cuda and n < EIGH_GPU_MIN:
         with _host_threads():
-            g = gram.cpu()
+            g = to_host(gram)
             L = torch.linalg.cholesky(g)
             t = torch.linalg.solve_triangular(
                 L, torch.eye(n, dtype=g.dtype), upper=False).conj()
-        return t.to(gram.device)
+        return to_dev(t.resolve_conj(), gram.device)
     L = torch.linalg.cholesky(gram)
     eye = torch.eye(n, dtype=gram.dtype, device=gram.device)
     return torch.linalg.solve_triangular(L, eye, upper=False).conj()

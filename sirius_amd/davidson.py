@@ -38,13 +38,15 @@ def _inner(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
     return la.inner(a, b)
 
 
-def _ortho_joint(new, hnew, snew, phi, hphi, sphi):
+def _ortho_joint(new, hnew, snew, phi, hphi, sphi, outs=None):
     """Project the existing S-orthonormal subspace out of `new` and
     S-orthonormalize it, applying identical transforms to hnew/snew
     (mirrors wf::orthogonalize, wave_functions.hpp:1781-2051).
 
     snew is None ⇔ S = I. Returns (new, hnew, snew) with possibly fewer
-    rows when rank-deficient.
+    rows when rank-deficient. With `outs` = (phi_buf, h_buf, s_buf) row
+    slices, the final orthonormalizing transform writes DIRECTLY into the
+    caller's subspace buffers (no extra device copy).
     """
     s_of_new = snew if snew is not None else new
     if phi is not None and phi.shape[0]:
@@ -68,6 +70,15 @@ def _ortho_joint(new, hnew, snew, phi, hphi, sphi):
         keep = w > 1e-10
         t = (v[:, keep] / torch.sqrt(w[keep])).conj().T     # [nkeep, n]
     tT = t.transpose(0, 1)
+    nkeep = t.shape[0]
+    if outs is not None:
+        o_phi, o_h, o_s = outs
+        new = la.transform(tT, new, out=o_phi[:nkeep])
+        hnew = la.transform(tT, hnew, out=o_h[:nkeep]) \
+            if hnew is not None else None
+        snew = la.transform(tT, snew, out=o_s[:nkeep]) \
+            if snew is not None else None
+        return new, hnew, snew
     new = la.transform(tT, new)
     hnew = la.transform(tT, hnew) if hnew is not None else None
     snew = la.transform(tT, snew) if snew is not None else None
@@ -183,20 +194,20 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
                     break
 
         hnew, snew = apply_h_s(res)
-        res, hnew, snew = _ortho_joint(res, hnew, snew, phi, hphi, sphi)
+        outs = (phi_buf[N:], hphi_buf[N:],
+                sphi_buf[N:] if sphi is not None else None)
         if extra_ortho:
             res, hnew, snew = _ortho_joint(res, hnew, snew, phi, hphi, sphi)
+        res, hnew, snew = _ortho_joint(res, hnew, snew, phi, hphi, sphi,
+                                       outs=outs)
         if res.shape[0] == 0:
             converged = True
             break
-        # grow subspace in place
+        # subspace grown in place by _ortho_joint(outs=...)
         nn = res.shape[0]
-        phi_buf[N:N + nn] = res
-        hphi_buf[N:N + nn] = hnew
         phi = phi_buf[:N + nn]
         hphi = hphi_buf[:N + nn]
         if sphi is not None:
-            sphi_buf[N:N + nn] = snew
             sphi = sphi_buf[:N + nn]
         Nn = phi.shape[0]
         Hn = torch.empty(Nn, Nn, dtype=H.dtype, device=H.device)
