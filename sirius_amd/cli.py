@@ -76,6 +76,7 @@ def main(argv=None):
             "efermi": res["efermi"],
             "magnetisation": {"total": [0.0, 0.0, res["magnetization"]]},
         },
+        "counters": res.get("counters", {}),
         "context": {
             "num_bands": ctx.num_bands,
             "num_kpoints": kset.num_kpoints,

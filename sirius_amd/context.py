@@ -244,6 +244,14 @@ class SimulationContext:
             int(p.num_bands) if p.num_bands > 0 else nbnd)
         self.max_occupancy = 2.0 if self.num_mag_dims == 0 else 1.0
 
+        # observability counters (reference: Simulation_context
+        # num_loc_op_applied/evp_work_count/num_itsol_steps,
+        # simulation_context.hpp:277-280; emitted in the output JSON like
+        # apps/mini_app/sirius.scf.cpp:233-235)
+        self.counters = {"local_operator_num_applied": 0,
+                         "band_evp_work_count": 0.0,
+                         "num_itsol_steps": 0}
+
         # crystal symmetry (space group + IBZ; reference: Crystal_symmetry)
         # (noncollinear case needs spin-rotation symmetrization — run the
         # full k-mesh instead)

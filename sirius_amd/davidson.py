@@ -31,6 +31,7 @@ class DavidsonResult:
     psi: torch.Tensor          # [nb, nG]
     niter: int = 0
     converged: bool = True
+    evp_work: float = 0.0      # Σ (N_subspace / num_bands)³ over eigensolves
 
 
 def _inner(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
@@ -125,6 +126,7 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
     H = _inner(phi, hphi)
     H = 0.5 * (H + H.conj().T)
     evals, Z = la.eigh(H)
+    evp_work = (phi.shape[0] / nb) ** 3
     eval_old = np.full(nb, 1e10)
     niter = 0
     converged = False
@@ -217,7 +219,9 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
         Hn[N:, :N] = blk[:N].conj().T
         H = 0.5 * (Hn + Hn.conj().T)
         evals, Z = la.eigh(H)
+        evp_work += (Nn / nb) ** 3
 
     psi = la.transform(Z[:, :nb], phi)
     return DavidsonResult(eval=evals[:nb].real.cpu().numpy(), psi=psi,
-                          niter=niter, converged=converged)
+                          niter=niter, converged=converged,
+                          evp_work=evp_work)

@@ -118,6 +118,8 @@ def diagonalize(ctx, h0: Hamiltonian0, kset: KPointSet, itsol_tol: float) -> boo
                 min_occupancy=itso.min_occupancy, extra_ortho=itso.extra_ortho)
             kp.psi[ispn] = res.psi
             kp.eigvals[ispn] = res.eval
+            ctx.counters["num_itsol_steps"] += res.niter
+            ctx.counters["band_evp_work_count"] += res.evp_work
             all_conv = all_conv and res.converged
     comm = get_comm()
     if comm.active:
@@ -289,6 +291,7 @@ class DFTGroundState:
             "efermi": self.kset.energy_fermi,
             "magnetization": self.density.total_magnetization()
             if ctx.num_mag_dims else 0.0,
+            "counters": dict(ctx.counters),
         }
         return out
 

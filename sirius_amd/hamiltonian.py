@@ -328,6 +328,7 @@ class HamiltonianK:
         if self.ctx.nc_magnetism:
             return self._apply_h_s_nc(psi)
         kp = self.kp
+        self.ctx.counters["local_operator_num_applied"] += psi.shape[0]
         # fused local operator: FFT⁻¹ → ×V_eff(r) → FFT → +½|G+k|²ψ
         hpsi = kp.fft.apply_veff_kinetic(psi, self.h0.veff_r_coarse[ispn],
                                          self.ekin)
@@ -349,6 +350,7 @@ class HamiltonianK:
         """Spinor apply (hamiltonian.hpp:454-515 noncollinear branch):
         h↑ = (T+V+Bz)ψ↑ + (Bx−iBy)ψ↓ + β[D⁰⁰⟨β|ψ↑⟩ + D⁰¹⟨β|ψ↓⟩], etc."""
         kp = self.kp
+        self.ctx.counters["local_operator_num_applied"] += 2 * psi.shape[0]
         ng = kp.num_gkvec
         up, dn = psi[:, :ng], psi[:, ng:]
         h0 = self.h0
