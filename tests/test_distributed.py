@@ -28,10 +28,16 @@ ctx = make_context(natoms=2, gk_cutoff=4.0, pw_cutoff=10.0, ngridk=(2, 2, 2),
 kset = KPointSet(ctx)
 dft = DFTGroundState(kset).initial_state()
 res = dft.find(num_dft_iter=6)
+f = dft.forces()
+st = dft.stress()
 if comm.rank == 0:
     print("RESULT " + json.dumps({"etot": res["energy"]["total"],
                                   "nk_local": len(kset.kpoints),
-                                  "nk": kset.num_kpoints}))
+                                  "nk": kset.num_kpoints,
+                                  "fmax": float(abs(f["total"]).max()),
+                                  "f00": float(f["total"][0][0]),
+                                  "s00": float(st["total"][0][0]),
+                                  "skin": float(st["kin"][0][0])}))
 """
 
 
@@ -54,8 +60,13 @@ def _run_world(n: int) -> dict:
 
 
 def test_kpoint_parallel_matches_serial():
+    """Energy, forces and stress identical between world sizes 1 and 2
+    (k-split + allreduce seams of density/forces/stress)."""
     r1 = _run_world(1)
     r2 = _run_world(2)
     assert r2["nk"] == r1["nk"]
     assert r2["nk_local"] < r1["nk_local"] or r1["nk"] == 1
+    assert abs(r2["etot"] - r1["etot"]) < 1e-8
+    for k in ("fmax", "f00", "s00", "skin"):
+        assert abs(r2[k] - r1[k]) < 1e-9, (k, r1[k], r2[k])
     assert abs(r1["etot"] - r2["etot"]) < 1e-8, (r1, r2)
