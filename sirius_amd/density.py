@@ -399,14 +399,32 @@ class Density:
 
     def mixer_init(self, cfg_mixer):
         """Register mixed quantities (density.cpp:1834; inner products per
-        mixer_functions.cpp — default metric: real-space L2, normalized by Ω)."""
+        mixer_functions.cpp — default metric: real-space L2, normalized by Ω;
+        mixer.use_hartree switches ρ to the Coulomb metric
+        Σ_{G≠0} 4π conj(x)y/G² (periodic_function_property_modified,
+        mixer_functions.cpp:85-110, density.cpp:1863)."""
         omega = self.ctx.unit_cell.omega
 
         def inner_pw(x, y):
             # ∫ x y dΩ = Ω Σ_G conj(x_G) y_G for real fields
             return omega * float(torch.vdot(x, y).real)
 
-        comps = [Component("rho_g", inner=inner_pw, global_size=omega)]
+        if getattr(cfg_mixer, "use_hartree", False):
+            g2 = self.ctx.gvec_fine.gk2_t.clamp(min=1e-30)
+            ig0 = self.ctx.gvec_fine.index_of_zero()
+            mask = torch.ones_like(g2)
+            if ig0 >= 0:
+                mask[ig0] = 0.0
+            hw = (4 * math.pi) * mask / g2
+
+            def inner_rho(x, y):
+                return float((torch.conj(x) * y * hw).sum().real)
+
+            rho_comp = Component("rho_g", inner=inner_rho,
+                                 global_size=1.0 / omega)
+        else:
+            rho_comp = Component("rho_g", inner=inner_pw, global_size=omega)
+        comps = [rho_comp]
         init = {"rho_g": self.rho_g}
         if self.ctx.nc_magnetism:
             for i in range(3):
