@@ -37,6 +37,10 @@ def main(argv=None):
                     help="relax atomic positions (fixed cell)")
     ap.add_argument("--vc-relax", action="store_true",
                     help="relax positions and cell")
+    ap.add_argument("--save-state", default=None, metavar="FILE.npz",
+                    help="write the converged density/potential state")
+    ap.add_argument("--restart", default=None, metavar="FILE.npz",
+                    help="start the SCF from a saved state")
     args = ap.parse_args(argv)
 
     from . import Config, SimulationContext, KPointSet, DFTGroundState
@@ -62,8 +66,16 @@ def main(argv=None):
     ctx = SimulationContext(cfg, base_dir=base, device=args.device)
     kset = KPointSet(ctx)
     dft = DFTGroundState(kset).initial_state()
+    if args.restart:
+        from .checkpoint import load_state
+
+        load_state(args.restart, dft)
     res = dft.find(num_dft_iter=args.num_iter)
     res["setup_and_scf_time"] = time.time() - t0
+    if args.save_state:
+        from .checkpoint import save_state
+
+        save_state(args.save_state, dft)
 
     out = {
         "ground_state": {
