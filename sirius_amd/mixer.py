@@ -222,15 +222,123 @@ class Broyden2(Mixer):
             self.S[:n, :n] = self.S[1:n + 1, 1:n + 1].clone()
 
 
+class AndersonStable(Mixer):
+    """Anderson with an orthonormal (QR) residual-difference history
+    (anderson_stable_mixer.hpp:50-222): modified Gram-Schmidt twice, the
+    Anderson solve through the triangular R, and Givens-rotation eviction
+    of the oldest column when the history is full."""
+
+    def __init__(self, *args, **kwargs):
+        super().__init__(*args, **kwargs)
+        self.history_size = 0
+        n = self.max_history - 1
+        self.R = torch.zeros(n, n, dtype=torch.float64)
+
+    def _rotate(self, c: float, sn: float, x: dict, y: dict):
+        for k in x:
+            xi = x[k].clone()
+            x[k].mul_(c).add_(sn * y[k])
+            y[k].mul_(c).sub_(sn * xi)
+
+    def mix_impl(self):
+        idx = self._idx(self.step)
+        idx_prev = self._idx(self.step - 1)
+        hs = self.history_size
+
+        nxt = self._copy(self.output_history[idx])
+        self._axpy(self.beta, self.residual_history[idx], nxt)
+
+        if hs > 0:
+            # Δf in the previous slot (f_n − f_{n−1}); same for Δx
+            self._scale(-1.0, self.residual_history[idx_prev])
+            self._axpy(1.0, self.residual_history[idx],
+                       self.residual_history[idx_prev])
+            self._scale(-1.0, self.output_history[idx_prev])
+            self._axpy(1.0, self.output_history[idx],
+                       self.output_history[idx_prev])
+            # modified Gram-Schmidt against the existing Q columns, twice
+            for _pass in range(2):
+                for i in range(1, hs):
+                    j = self._idx(self.step - i - 1)
+                    sz = self._inner(self.residual_history[j],
+                                     self.residual_history[idx_prev],
+                                     normalize=False)
+                    if _pass == 0:
+                        self.R[hs - 1 - i, hs - 1] = sz
+                    else:
+                        self.R[hs - 1 - i, hs - 1] += sz
+                    self._axpy(-sz, self.residual_history[j],
+                               self.residual_history[idx_prev])
+            nrm2 = self._inner(self.residual_history[idx_prev],
+                               self.residual_history[idx_prev],
+                               normalize=False)
+            if nrm2 > 0:
+                sz = nrm2 ** 0.5
+                self.R[hs - 1, hs - 1] = sz
+                self._scale(1.0 / sz, self.residual_history[idx_prev])
+                # h = Qᵀ f_n ; k = R⁻¹ h
+                h = torch.zeros(hs, dtype=torch.float64)
+                for i in range(1, hs + 1):
+                    j = self._idx(self.step - i)
+                    h[hs - i] = self._inner(self.residual_history[j],
+                                            self.residual_history[idx],
+                                            normalize=False)
+                k = h.clone()
+                for j in range(hs - 1, -1, -1):
+                    k[j] /= self.R[j, j]
+                    for i in range(j - 1, -1, -1):
+                        k[i] -= self.R[i, j] * k[j]
+                for i in range(1, hs + 1):
+                    j = self._idx(self.step - i)
+                    self._axpy(-self.beta * float(h[hs - i]),
+                               self.residual_history[j], nxt)
+                    self._axpy(-float(k[hs - i]),
+                               self.output_history[j], nxt)
+            else:
+                self.history_size = 0
+                hs = 0
+
+        if self.history_size == self.max_history - 1:
+            # Givens eviction of the oldest column
+            for row in range(1, hs):
+                a, b = float(self.R[row - 1, row]), float(self.R[row, row])
+                nrm = (a * a + b * b) ** 0.5
+                c = a / nrm if nrm > 0 else 1.0
+                sn = b / nrm if nrm > 0 else 0.0
+                self.R[row - 1, row] = nrm
+                self.R[row, row] = 0.0
+                for col in range(row + 1, hs):
+                    r1, r2 = float(self.R[row - 1, col]), float(self.R[row, col])
+                    self.R[row - 1, col] = c * r1 + sn * r2
+                    self.R[row, col] = -sn * r1 + c * r2
+                i1 = self._idx(self.step - hs + row - 1)
+                i2 = self._idx(self.step - hs + row)
+                self._rotate(c, sn, self.residual_history[i1],
+                             self.residual_history[i2])
+            for i in range(1, hs):
+                i1 = self._idx(self.step - i - 1)
+                i2 = self._idx(self.step - i)
+                self.residual_history[i1], self.residual_history[i2] = \
+                    self.residual_history[i2], self.residual_history[i1]
+            for col in range(hs - 1):
+                for row in range(col + 1):
+                    self.R[row, col] = self.R[row, col + 1]
+
+        self.output_history[self._idx(self.step + 1)] = nxt
+        self.history_size = min(self.history_size + 1, self.max_history - 1)
+
+
 def make_mixer(cfg_mixer, components: list[Component]) -> Mixer:
     kind = cfg_mixer.type
     kw = dict(max_history=cfg_mixer.max_history, beta=cfg_mixer.beta,
               beta0=cfg_mixer.beta0, beta_scaling_factor=cfg_mixer.beta_scaling_factor)
     if kind == "linear":
         return Linear(components, **kw)
-    if kind in ("anderson", "anderson_stable", "broyden1"):
-        # anderson_stable/broyden1 fall back to anderson (same family)
+    if kind in ("anderson", "broyden1"):
+        # broyden1 falls back to anderson (same family)
         return Anderson(components, **kw)
+    if kind == "anderson_stable":
+        return AndersonStable(components, **kw)
     if kind == "broyden2":
         return Broyden2(components, **kw)
     raise ValueError(f"unknown mixer type {kind}")

@@ -211,3 +211,25 @@ def test_broyden2_mixer_fixed_point():
     hl = run(Linear, 30)
     assert hb[-1] < 1e-9
     assert hb[-1] < hl[-1] * 1e-2
+
+
+def test_anderson_stable_mixer_fixed_point():
+    """AndersonStable (QR residual history + Givens eviction) converges
+    the same fixed point as Anderson, incl. past full history depth."""
+    import torch
+    from sirius_amd.mixer import AndersonStable, Component
+
+    torch.manual_seed(1)
+    n = 40
+    M = torch.randn(n, n, dtype=torch.float64)
+    M = 0.6 * M / torch.linalg.matrix_norm(M, 2)
+    b = torch.randn(n, dtype=torch.float64)
+    mx = AndersonStable([Component("x")], max_history=5, beta=0.4)
+    x = torch.zeros(n, dtype=torch.float64)
+    mx.initialize({"x": x})
+    hist = []
+    for _ in range(40):     # > max_history → exercises the eviction path
+        cur = mx.get_output()["x"]
+        mx.set_input({"x": M @ cur + b})
+        hist.append(mx.mix())
+    assert hist[-1] < 1e-10, hist[-5:]
