@@ -22,7 +22,8 @@
 typedef double d4 __attribute__((ext_vector_type(4)));
 
 #define TILE 32              // block tile (M and N)
-#define KCH 16               // k-chunk staged in LDS per iteration
+#define KCH 16               // k-chunk staged in LDS per iteration (32 tried: slower end-to-end)
+#define NSLOT ((TILE * KCH) / 256)   // staging slots per thread
 
 __launch_bounds__(256, 2)
 __global__ void zgram_splitk_kernel(const double* __restrict__ A,  // [M, 2K] interleaved
@@ -57,19 +58,21 @@ __global__ void zgram_splitk_kernel(const double* __restrict__ A,  // [M, 2K] in
     const int a_row = lane % 16;
     const int a_k = lane / 16;          // 0..3
 
-    // software pipeline: each thread owns two fixed (row, k) staging slots;
-    // the NEXT chunk is prefetched into registers while MFMAs consume the
-    // LDS-resident one (hides the HBM latency behind the matrix math).
-    const int r0 = tid / KCH, k0 = tid % KCH;
-    const int r1 = (tid + 256) / KCH, k1 = (tid + 256) % KCH;
-    double pa[2][2], pb[2][2];
+    // software pipeline: each thread owns NSLOT fixed (row, k) staging
+    // slots; the NEXT chunk is prefetched into registers while MFMAs
+    // consume the LDS-resident one.
+    int rr[NSLOT], kk[NSLOT];
+    #pragma unroll
+    for (int j = 0; j < NSLOT; j++) {
+        rr[j] = (tid + 256 * j) / KCH;
+        kk[j] = (tid + 256 * j) % KCH;
+    }
+    double pa[NSLOT][2], pb[NSLOT][2];
 
     auto prefetch = [&](long kb) {
-        const int rr[2] = {r0, r1};
-        const int kk[2] = {k0, k1};
         const int kc = (int)min((long)KCH, k_end - kb);
         #pragma unroll
-        for (int j = 0; j < 2; j++) {
+        for (int j = 0; j < NSLOT; j++) {
             pa[j][0] = pa[j][1] = pb[j][0] = pb[j][1] = 0.0;
             if (kk[j] < kc) {
                 if (m0 + rr[j] < M) {
@@ -86,14 +89,13 @@ __global__ void zgram_splitk_kernel(const double* __restrict__ A,  // [M, 2K] in
         }
     };
     auto commit = [&]() {
-        lA[r0][2 * k0] = pa[0][0];
-        lA[r0][2 * k0 + 1] = pa[0][1];
-        lB[r0][2 * k0] = pb[0][0];
-        lB[r0][2 * k0 + 1] = pb[0][1];
-        lA[r1][2 * k1] = pa[1][0];
-        lA[r1][2 * k1 + 1] = pa[1][1];
-        lB[r1][2 * k1] = pb[1][0];
-        lB[r1][2 * k1 + 1] = pb[1][1];
+        #pragma unroll
+        for (int j = 0; j < NSLOT; j++) {
+            lA[rr[j]][2 * kk[j]] = pa[j][0];
+            lA[rr[j]][2 * kk[j] + 1] = pa[j][1];
+            lB[rr[j]][2 * kk[j]] = pb[j][0];
+            lB[rr[j]][2 * kk[j] + 1] = pb[j][1];
+        }
     };
 
     prefetch(k_begin);
@@ -169,18 +171,20 @@ __global__ void ztrans_kernel(const double* __restrict__ T,  // [K, 2M] interlea
     const int a_k = lane / 16;       // k within 4-step
     const double tsgn = conj_t ? -1.0 : 1.0;
 
-    // software pipeline (same scheme as zgram): 2 fixed staging slots per
+    // software pipeline (same scheme as zgram): NSLOT staging slots per
     // thread, next chunk prefetched into registers during the MFMAs.
-    const int k0s = tid / TILE, c0s = tid % TILE;
-    const int k1s = (tid + 256) / TILE, c1s = (tid + 256) % TILE;
-    double pt[2][2], px[2][2];
+    int kks[NSLOT], ccs[NSLOT];
+    #pragma unroll
+    for (int j = 0; j < NSLOT; j++) {
+        kks[j] = (tid + 256 * j) / TILE;
+        ccs[j] = (tid + 256 * j) % TILE;
+    }
+    double pt[NSLOT][2], px[NSLOT][2];
 
     auto prefetch = [&](int kb) {
-        const int kks[2] = {k0s, k1s};
-        const int ccs[2] = {c0s, c1s};
         const int kc = min(KCH, K - kb);
         #pragma unroll
-        for (int j = 0; j < 2; j++) {
+        for (int j = 0; j < NSLOT; j++) {
             pt[j][0] = pt[j][1] = px[j][0] = px[j][1] = 0.0;
             if (kks[j] < kc) {
                 if (m0 + ccs[j] < M) {
@@ -197,14 +201,13 @@ __global__ void ztrans_kernel(const double* __restrict__ T,  // [K, 2M] interlea
         }
     };
     auto commit = [&]() {
-        lT[k0s][2 * c0s] = pt[0][0];
-        lT[k0s][2 * c0s + 1] = pt[0][1];
-        lX[k0s][2 * c0s] = px[0][0];
-        lX[k0s][2 * c0s + 1] = px[0][1];
-        lT[k1s][2 * c1s] = pt[1][0];
-        lT[k1s][2 * c1s + 1] = pt[1][1];
-        lX[k1s][2 * c1s] = px[1][0];
-        lX[k1s][2 * c1s + 1] = px[1][1];
+        #pragma unroll
+        for (int j = 0; j < NSLOT; j++) {
+            lT[kks[j]][2 * ccs[j]] = pt[j][0];
+            lT[kks[j]][2 * ccs[j] + 1] = pt[j][1];
+            lX[kks[j]][2 * ccs[j]] = px[j][0];
+            lX[kks[j]][2 * ccs[j] + 1] = px[j][1];
+        }
     };
 
     prefetch(0);
