@@ -113,7 +113,7 @@ def inner(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
             and a.is_contiguous() and b.is_contiguous()):
         from .. import ops
 
-        z = ops.get_zgemm(required=False)
+        z = ops.get_zgemm()     # raises on GPU if the extension is missing
         if z is not None:
             return z.zgram(a, b, 0)
     return (a @ b.conj().T).conj()
@@ -137,7 +137,7 @@ def transform(T: torch.Tensor, X: torch.Tensor,
             and (out is None or out.is_contiguous())):
         from .. import ops
 
-        z = ops.get_zgemm(required=False)
+        z = ops.get_zgemm()     # raises on GPU if the extension is missing
         if z is not None:
             Tc = T.resolve_conj().contiguous()
             if out is None:

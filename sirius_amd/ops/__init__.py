@@ -76,8 +76,17 @@ def get_ext(required: bool | None = None):
 
 
 def get_zgemm(required: bool | None = None):
-    """The MFMA fp64 Gram-GEMM extension (None on CPU)."""
+    """The MFMA fp64 zgemm extension (None on CPU).
+
+    required defaults to torch.cuda.is_available(): on a GPU box the MFMA
+    kernels are the compute path — a missing extension raises instead of
+    silently falling back to rocBLAS."""
+    if required is None:
+        required = torch.cuda.is_available()
     get_ext(required)
+    if _ext_zgemm is None and required:
+        raise RuntimeError(
+            "sirius_amd MFMA zgemm extension unavailable on GPU machine")
     return _ext_zgemm
 
 
