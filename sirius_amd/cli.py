@@ -41,6 +41,7 @@ def main(argv=None):
                     help="write the converged density/potential state")
     ap.add_argument("--restart", default=None, metavar="FILE.npz",
                     help="start the SCF from a saved state")
+    ap.add_argument("--verbosity", "-v", type=int, default=1)
     args = ap.parse_args(argv)
 
     from . import Config, SimulationContext, KPointSet, DFTGroundState
@@ -70,7 +71,12 @@ def main(argv=None):
         from .checkpoint import load_state
 
         load_state(args.restart, dft)
-    res = dft.find(num_dft_iter=args.num_iter)
+    cb = None
+    if args.verbosity >= 1:
+        def cb(it, etot, rms):
+            print(f"iter {it:3d}  Etot {etot:+.10f} Ha  rms {rms:.3e}",
+                  flush=True)
+    res = dft.find(num_dft_iter=args.num_iter, callback=cb)
     res["setup_and_scf_time"] = time.time() - t0
     if args.save_state:
         from .checkpoint import save_state
@@ -87,6 +93,7 @@ def main(argv=None):
             "scf_time": res["scf_time"],
             "efermi": res["efermi"],
             "magnetisation": {"total": [0.0, 0.0, res["magnetization"]]},
+            "band_gap": res.get("band_gap", 0.0),
         },
         "counters": res.get("counters", {}),
         "context": {

@@ -292,6 +292,7 @@ class DFTGroundState:
             "magnetization": self.density.total_magnetization()
             if ctx.num_mag_dims else 0.0,
             "counters": dict(ctx.counters),
+            "band_gap": self.kset.band_gap,
         }
         return out
 

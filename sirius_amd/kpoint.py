@@ -161,10 +161,19 @@ class KPointSet:
         for i, kp in enumerate(self.kpoints):
             kp.occ = occ_all[self.local_range[0] + i]
         self._all_occ = occ_all
-        # band gap (valence max / conduction min when gapped)
-        nve = int(round(ne / ctx.max_occupancy))
-        srt = np.sort(eigf, axis=0)
+        # band gap via per-band (min, max) ranges over the k set, sorted,
+        # at integer filling (K_point_set::find_band_occupancies,
+        # k_point_set.cpp:416-449)
         self.band_gap = 0.0
+        nve = int(ne + 1e-12)
+        if ctx.num_spins == 2 or (abs(nve - ne) < 1e-12 and nve % 2 == 0):
+            emin = self._all_eig.min(axis=0).reshape(-1)   # per (spin, band)
+            emax = self._all_eig.max(axis=0).reshape(-1)
+            order = np.lexsort((emax, emin))
+            emin, emax = emin[order], emax[order]
+            ist = nve if ctx.num_spins == 2 else nve // 2
+            if ist < len(emin) and emin[ist] > emax[ist - 1]:
+                self.band_gap = float(emin[ist] - emax[ist - 1])
 
     def valence_eval_sum(self) -> float:
         """Σ_k w_k Σ_n f_nk ε_nk (over ALL k; every rank returns the total)."""
