@@ -116,6 +116,12 @@ def main(argv=None):
     etot = res["energy"]["total"]
     print(f"total energy: {etot:.10f} Ha  (converged: {res['converged']}, "
           f"{res['num_scf_iterations']} iterations)")
+    if args.verbosity >= 2 or os.environ.get("SIRIUS_AMD_PRINT_PERFORMANCE"):
+        # rt_graph-style timer tree (reference: control.print_timers /
+        # SIRIUS_PRINT_PERFORMANCE env toggles)
+        from .utils.profiler import profiler
+
+        print(profiler.report())
 
     if args.test_against:
         ref = json.load(open(os.path.join(base, args.test_against)))
