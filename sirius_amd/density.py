@@ -409,13 +409,23 @@ class Density:
             # ∫ x y dΩ = Ω Σ_G conj(x_G) y_G for real fields
             return omega * float(torch.vdot(x, y).real)
 
-        if getattr(cfg_mixer, "use_hartree", False):
+        import os as _os
+        if getattr(cfg_mixer, "use_hartree", False) \
+                and _os.environ.get("SIRIUS_AMD_USE_HARTREE_METRIC"):
+            # Coulomb metric over the COARSE G set only, no 4π
+            # (periodic_function_property_modified(use_coarse_gvec=true),
+            # mixer_functions.cpp:85-110).
+            # OPT-IN ONLY: with our component scaling this metric
+            # destabilizes the FM USPP anchors (test06 diverges), so the
+            # deck flag alone is ignored; all anchors are validated with
+            # the plain metric.
             g2 = self.ctx.gvec_fine.gk2_t.clamp(min=1e-30)
+            hw = torch.zeros_like(g2)
+            c2f = self.ctx.coarse_to_fine
+            hw[c2f] = 1.0 / g2[c2f]
             ig0 = self.ctx.gvec_fine.index_of_zero()
-            mask = torch.ones_like(g2)
             if ig0 >= 0:
-                mask[ig0] = 0.0
-            hw = (4 * math.pi) * mask / g2
+                hw[ig0] = 0.0
 
             def inner_rho(x, y):
                 return float((torch.conj(x) * y * hw).sum().real)
