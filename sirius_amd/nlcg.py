@@ -8,8 +8,8 @@ minimization of the total-energy functional over the orthonormality
 manifold of the wavefunctions.
 
 This native implementation covers the fixed-occupation (insulating)
-case with the norm-conserving metric (S = I):
-- gradient  g_k = f ⊙ (1 − |ψ⟩⟨ψ|) H[ρ] |ψ⟩  per k-point,
+case, with the ultrasoft metric when S ≠ I (orthonormality ⟨ψ|S|ψ⟩ = 1):
+- gradient  g_k = f ⊙ (1 − S|ψ⟩⟨ψ|) H[ρ] |ψ⟩  per k-point,
 - Teter-preconditioned Polak-Ribière CG directions,
 - parabolic line search on E(θ) along the tangent direction with
   Löwdin re-orthonormalization,
@@ -25,8 +25,9 @@ import torch
 from .core import la
 
 
-def _lowdin(psi: torch.Tensor) -> torch.Tensor:
-    s = la.inner(psi, psi)
+def _lowdin(psi: torch.Tensor, spsi: torch.Tensor | None = None) -> torch.Tensor:
+    """S-metric Löwdin orthonormalization (S = I when spsi is None)."""
+    s = la.inner(psi, spsi if spsi is not None else psi)
     w, v = la.eigh(0.5 * (s + s.conj().T))
     t = (v / torch.sqrt(w.clamp(min=1e-14))) @ v.conj().T
     return t.conj().T @ psi
@@ -43,8 +44,6 @@ class DirectMinimizer:
         self.tol = tol
         if self.ctx.num_spins != 1 or self.ctx.nc_magnetism:
             raise NotImplementedError("direct minimization: nm only")
-        if self.ctx.has_aug:
-            raise NotImplementedError("direct minimization: S=I (NC) only")
         # the functional must be evaluated on the raw rho[psi] — the
         # symmetrized density would make E inconsistent with the orbital
         # gradient f·H|psi> (and IBZ k-sets are incomplete for CG)
@@ -67,6 +66,14 @@ class DirectMinimizer:
             for i in range(self._nocc(kp)):
                 kp.occ[0][i] = min(mo, left)
                 left -= kp.occ[0][i]
+
+    def _apply_s(self, hk, psi):
+        if hk.Q is None or hk.bp.num_beta_total == 0:
+            return None
+        b = hk.bp.inner(psi)
+        out = psi.clone().contiguous()
+        la.transform(hk.Q @ b, hk.bp.beta_t, out=out, accumulate=True)
+        return out
 
     def _energy(self, psis) -> float:
         """E at given (orthonormal) occupied orbitals."""
@@ -103,9 +110,11 @@ class DirectMinimizer:
         out = []
         for kp, psi in zip(self.kset, psis):
             hk = self._h0(kp)
-            hpsi, _ = hk.apply_h_s(psi, 0)
+            hpsi, spsi = hk.apply_h_s(psi, 0)
             ov = la.inner(psi, hpsi)
-            g = hpsi - la.transform(ov.transpose(0, 1), psi)
+            # (1 − S|ψ⟩⟨ψ|) H ψ: project with Sψ when S ≠ I
+            g = hpsi - la.transform(ov.transpose(0, 1),
+                                    spsi if spsi is not None else psi)
             n = psi.shape[0]
             f = torch.from_numpy(kp.occ[0][:n]).to(g.device)
             g = (kp.weight * f)[:, None].to(g.dtype) * g
@@ -117,12 +126,24 @@ class DirectMinimizer:
             out.append((g, 1.0 / p))
         return out
 
+    def _lowdin_k(self, kp, psi):
+        """Löwdin in the right metric (uses the CURRENT h0's S operator)."""
+        if not self.ctx.has_aug:
+            return _lowdin(psi)
+        hk = self._h0(kp)
+        return _lowdin(psi, self._apply_s(hk, psi))
+
     def run(self) -> dict:
+        from .hamiltonian import Hamiltonian0
+
         self._set_occ()
+        # h0 needed for the S metric of the initial orthonormalization
+        self._h0 = Hamiltonian0(self.ctx, self.dft.potential,
+                                self.dft.density)
         psis = []
         for kp in self.kset:
             n = self._nocc(kp)
-            psis.append(_lowdin(kp.psi[0][:n].clone().contiguous()))
+            psis.append(self._lowdin_k(kp, kp.psi[0][:n].clone().contiguous()))
         e = self._energy(psis)
         hist = [e]
         dirs = None
@@ -155,16 +176,16 @@ class DirectMinimizer:
             b = 2.0 * sum(float((d.conj() * g).sum().real)
                           for d, (g, _) in zip(dirs, grads))
             theta_t = 0.2 / max(1.0, abs(b))
-            psis_t = [_lowdin(psi + theta_t * d)
-                      for psi, d in zip(psis, dirs)]
+            psis_t = [self._lowdin_k(kp, psi + theta_t * d)
+                      for kp, psi, d in zip(self.kset, psis, dirs)]
             e_t = self._energy(psis_t)
             c = (e_t - e - b * theta_t) / theta_t ** 2
             if c > 0:
                 theta = max(min(-b / (2 * c), 5 * theta_t), 0.05 * theta_t)
             else:
                 theta = theta_t if e_t < e else 0.1 * theta_t
-            psis_n = [_lowdin(psi + theta * d)
-                      for psi, d in zip(psis, dirs)]
+            psis_n = [self._lowdin_k(kp, psi + theta * d)
+                      for kp, psi, d in zip(self.kset, psis, dirs)]
             e_n = self._energy(psis_n)
             if e_n > e and e_t < e:
                 psis_n, e_n = psis_t, self._energy(psis_t)

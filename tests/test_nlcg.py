@@ -39,3 +39,36 @@ def test_direct_minimization_matches_scf():
     assert abs(res["etot"] - r["energy"]["total"]) < 1e-9
     h = res["history"]
     assert all(h[i + 1] <= h[i] + 1e-12 for i in range(len(h) - 1))
+
+
+@requires_reference
+@pytest.mark.slow
+def test_direct_minimization_uspp():
+    """Orbital CG with the ultrasoft S metric reaches the SCF fixed point
+    (<1e-5 Ha) on Γ-only Si USPP with matched integer occupations."""
+    from sirius_amd import Config, SimulationContext, KPointSet, DFTGroundState
+    from sirius_amd.nlcg import DirectMinimizer
+
+    base = os.path.join(REFERENCE, "verification", "test08")
+
+    def mk():
+        cfg = Config.from_json(os.path.join(base, "sirius.json"))
+        cfg.override("parameters.use_symmetry", False)
+        cfg.override("parameters.ngridk", [1, 1, 1])
+        cfg.override("parameters.smearing_width", 1e-5)
+        ctx = SimulationContext(cfg, base_dir=base, device="cpu")
+        ctx.symmetry = None
+        return ctx
+
+    ctx = mk()
+    kset = KPointSet(ctx)
+    dft = DFTGroundState(kset).initial_state()
+    r = dft.find(density_tol=1e-11, energy_tol=1e-12)
+
+    ctx2 = mk()
+    kset2 = KPointSet(ctx2)
+    dft2 = DFTGroundState(kset2).initial_state()
+    res = DirectMinimizer(dft2, maxiter=120, tol=1e-12).run()
+    assert abs(res["etot"] - r["energy"]["total"]) < 1e-5
+    h = res["history"]
+    assert all(h[i + 1] <= h[i] + 1e-12 for i in range(len(h) - 1))
