@@ -121,3 +121,20 @@ class LinearResponseOperator:
         proj = la.transform(ov, self.evq)          # [nact, nG]
         sproj = self._apply_s(proj)
         return out + self.alpha_pv * sproj
+
+
+def linear_solver(hk, eigvals, evq, B, alpha_pv: float = 1.0,
+                  ispn: int = 0, maxiters: int = 300, tol: float = 1e-9):
+    """Sternheimer solve (H − ε_j S + α_pv·S Q Q^H S) X = B for a block of
+    right-hand sides — the `sirius_linear_solver` API seam used by
+    QE-DFPT (reference: sirius_api.cpp sirius_linear_solver →
+    multi_cg.hpp lr::multi_cg).
+
+    eigvals [n] (per-rhs ε_j), evq [nocc, nG] occupied eigenvectors,
+    B [n, nG]. Returns (X, niter, residual_histories).
+    """
+    A = LinearResponseOperator(hk, eigvals, evq, alpha_pv, ispn)
+    P = SmoothedDiagonalPreconditioner(hk.h_diag(ispn), hk.o_diag(), eigvals)
+    X = torch.zeros_like(B)
+    return multi_cg(A, P, X, B.clone(), maxiters=maxiters, tol=tol,
+                    initial_guess_is_zero=True)

@@ -188,6 +188,40 @@ def test_forces_stress_gpu_match_cpu():
 
 
 @requires_gpu
+def test_sternheimer_gpu():
+    """Block-CG linear response runs on device and satisfies the residual."""
+    import torch
+    from sirius_amd.models.synthetic import make_context
+    from sirius_amd.kpoint import KPointSet
+    from sirius_amd.dft import DFTGroundState
+    from sirius_amd.hamiltonian import HamiltonianK
+    from sirius_amd.multi_cg import linear_solver
+    from sirius_amd.core import la
+
+    ctx = make_context(natoms=2, gk_cutoff=4.0, pw_cutoff=10.0,
+                       device="cuda:0")
+    kset = KPointSet(ctx)
+    dft = DFTGroundState(kset).initial_state()
+    dft.find(num_dft_iter=8)
+    kp = kset.kpoints[0]
+    hk = HamiltonianK(dft.h0, kp)
+    nocc = 4
+    evq = kp.psi[0][:nocc].contiguous()
+    eig = torch.from_numpy(kp.eigvals[0][:nocc]).to(ctx.device)
+    torch.manual_seed(3)
+    B = torch.view_as_complex(
+        torch.randn(nocc, kp.num_gkvec, 2, dtype=torch.float64,
+                    device="cuda:0"))
+    ov = la.inner(evq, B)
+    B = B - la.transform(ov, evq)
+    X, it, hist = linear_solver(hk, eig, evq, B, maxiters=300, tol=1e-8)
+    from sirius_amd.multi_cg import LinearResponseOperator
+    A = LinearResponseOperator(hk, eig, evq, 1.0)
+    res = (B - A(X, torch.arange(nocc, device="cuda:0"))).abs().max()
+    assert float(res) < 1e-5, (float(res), it)
+
+
+@requires_gpu
 def test_native_ops_loaded():
     """The in-tree HIP extension (not a fallback) is what runs on GPU."""
     from sirius_amd import ops
