@@ -22,7 +22,6 @@ import torch
 
 from .core import ylm as ylm_mod
 from .core.gaunt import gaunt_rrr
-from .core.radial import RadialIntegrals
 
 
 def packed_index(xi1: int, xi2: int) -> int:

@@ -10,7 +10,6 @@ energy (energy.cpp:142-165) → convergence on |ΔE| and RMS (:349-356).
 
 from __future__ import annotations
 
-import json
 import math
 import time
 

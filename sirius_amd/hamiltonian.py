@@ -27,7 +27,6 @@ import torch
 
 from .core import la
 from .core import ylm as ylm_mod
-from .core.radial import RadialIntegrals
 
 
 class BetaProjectors:

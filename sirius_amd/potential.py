@@ -16,7 +16,6 @@ import math
 import numpy as np
 import torch
 
-from .core.radial import RadialIntegrals
 from . import xc as xc_mod
 
 

@@ -19,7 +19,6 @@ mirroring lattice_relaxation.hpp:44-76.
 
 from __future__ import annotations
 
-import json
 
 import numpy as np
 
