@@ -233,3 +233,29 @@ def test_anderson_stable_mixer_fixed_point():
         mx.set_input({"x": M @ cur + b})
         hist.append(mx.mix())
     assert hist[-1] < 1e-10, hist[-5:]
+
+
+def test_counters_and_checkpoint_cli(tmp_path):
+    """Observability counters populate and the save/restart CLI round-trips."""
+    from sirius_amd.models.synthetic import make_context
+    from sirius_amd.kpoint import KPointSet
+    from sirius_amd.dft import DFTGroundState
+    from sirius_amd.checkpoint import save_state, load_state
+
+    ctx = make_context(natoms=2, gk_cutoff=4.0, pw_cutoff=10.0, device="cpu")
+    kset = KPointSet(ctx)
+    dft = DFTGroundState(kset).initial_state()
+    r = dft.find(num_dft_iter=4)
+    c = r["counters"]
+    assert c["local_operator_num_applied"] > 0
+    assert c["num_itsol_steps"] > 0
+    assert c["band_evp_work_count"] > 0
+    p = str(tmp_path / "state.npz")
+    save_state(p, dft)
+
+    ctx2 = make_context(natoms=2, gk_cutoff=4.0, pw_cutoff=10.0, device="cpu")
+    kset2 = KPointSet(ctx2)
+    dft2 = DFTGroundState(kset2).initial_state()
+    load_state(p, dft2)
+    import torch
+    assert torch.allclose(dft2.density.rho_r, dft.density.rho_r, atol=1e-12)
