@@ -259,3 +259,21 @@ def test_counters_and_checkpoint_cli(tmp_path):
     load_state(p, dft2)
     import torch
     assert torch.allclose(dft2.density.rho_r, dft.density.rho_r, atol=1e-12)
+
+
+def test_radial_solver_hydrogen():
+    """Radial bound states reproduce the exact Coulomb spectrum
+    (Radial_solver seam for the LAPW branch)."""
+    import numpy as np
+    from sirius_amd.core.radial_solver import bound_states, hydrogenic_levels
+
+    r = np.geomspace(1e-6, 80.0, 1300)
+    for l in (0, 1):
+        e, R = bound_states(r, -1.0 / r, l, nstates=3)
+        ref = hydrogenic_levels(1.0, l, 3)
+        assert np.abs(e - ref).max() < 1e-4, (l, e, ref)
+    # 1s radial function: R_10 = 2 e^{-r}
+    e, R = bound_states(r, -1.0 / r, 0, nstates=1)
+    ref_R = 2.0 * np.exp(-r)
+    sel = (r > 0.1) & (r < 5.0)
+    assert np.abs(R[0][sel] - ref_R[sel]).max() < 1e-3
