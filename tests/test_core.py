@@ -277,3 +277,14 @@ def test_radial_solver_hydrogen():
     ref_R = 2.0 * np.exp(-r)
     sel = (r > 0.1) & (r < 5.0)
     assert np.abs(R[0][sel] - ref_R[sel]).max() < 1e-3
+
+
+def test_atomic_lda_helium():
+    """Self-consistent spherical LDA He vs NIST LSD reference
+    (ε_1s = −0.570425, E_tot = −2.8348 with VWN; PZ differs ~1e-3)."""
+    from sirius_amd.atoms import solve_atom
+
+    r = solve_atom(2)
+    assert r["converged"]
+    assert abs(r["levels"][(1, 0)] - (-0.570425)) < 2e-3
+    assert abs(r["etot"] - (-2.8348)) < 5e-3
