@@ -69,3 +69,102 @@ def hydrogenic_levels(zn: float, l: int, n_levels: int = 4) -> np.ndarray:
     """Exact Coulomb levels −Z²/2n² for n = l+1, … (test reference)."""
     n = np.arange(l + 1, l + 1 + n_levels)
     return -zn**2 / (2.0 * n.astype(np.float64) ** 2)
+
+
+SPEED_OF_LIGHT = 137.035999084   # Hartree atomic units
+
+
+def bound_state_sr(r: np.ndarray, v: np.ndarray, n: int, l: int,
+                   enu: float | None = None, rel: bool = True,
+                   tol: float = 1e-10):
+    """One bound state (n, l) of the scalar-relativistic (Koelling-Harmon)
+    radial equation by outward RK4 shooting with node-count bisection
+    (reference: Radial_solver::integrate_forward_rk4 + the bound-state
+    search of src/radial/radial_solver.hpp:786-990).
+
+        u'' = [l(l+1)/r² + 2M(V−E)] u + (M'/M)(u' − u/r),
+        M   = 1 + (E − V)/2c²      (M ≡ 1 for rel=False)
+
+    Returns (E, R on r) with ∫R²r²dr = 1. The target state has
+    n − l − 1 radial nodes.
+    """
+    from scipy.interpolate import CubicSpline
+
+    r = np.asarray(r, dtype=np.float64)
+    v = np.asarray(v, dtype=np.float64)
+    vs = CubicSpline(r, v)
+    c2 = SPEED_OF_LIGHT ** 2 if rel else 1e30
+    nodes_target = n - l - 1
+
+    # integration grid: log-spaced, dense
+    t = np.linspace(np.log(max(r[0], 1e-8)), np.log(r[-1]), 6000)
+    rg = np.exp(t)
+    vg = vs(rg)
+    dvg = vs(rg, 1)
+
+    def shoot(E):
+        """integrate u outward; return (nodes, u, u(rmax) sign function)."""
+        M = 1.0 + (E - vg) / (2 * c2)
+        dM = -dvg / (2 * c2)
+        A = l * (l + 1) / rg**2 + 2.0 * M * (vg - E)
+        B = dM / M
+        u = np.zeros_like(rg)
+        up = np.zeros_like(rg)
+        u[0] = rg[0] ** (l + 1)
+        up[0] = (l + 1) * rg[0] ** l
+
+        def rhs(i, frac, ui, upi):
+            # linear interpolation of coefficients inside the step
+            a = A[i] * (1 - frac) + A[min(i + 1, len(rg) - 1)] * frac
+            b = B[i] * (1 - frac) + B[min(i + 1, len(rg) - 1)] * frac
+            rr = rg[i] * (1 - frac) + rg[min(i + 1, len(rg) - 1)] * frac
+            return upi, a * ui + b * (upi - ui / rr)
+
+        nodes = 0
+        scale = 0.0
+        for i in range(len(rg) - 1):
+            h = rg[i + 1] - rg[i]
+            k1u, k1p = rhs(i, 0.0, u[i], up[i])
+            k2u, k2p = rhs(i, 0.5, u[i] + 0.5 * h * k1u, up[i] + 0.5 * h * k1p)
+            k3u, k3p = rhs(i, 0.5, u[i] + 0.5 * h * k2u, up[i] + 0.5 * h * k2p)
+            k4u, k4p = rhs(i, 1.0, u[i] + h * k3u, up[i] + h * k3p)
+            u[i + 1] = u[i] + h / 6 * (k1u + 2 * k2u + 2 * k3u + k4u)
+            up[i + 1] = up[i] + h / 6 * (k1p + 2 * k2p + 2 * k3p + k4p)
+            if u[i + 1] * u[i] < 0:
+                nodes += 1
+            m = abs(u[i + 1])
+            if m > 1e20:             # renormalize both to avoid overflow
+                u[: i + 2] /= m
+                up[: i + 2] /= m
+                # (node count unaffected: uniform positive scaling)
+        return nodes, u
+
+    # bracket E by node count
+    elo = (max(vg.min(), -50.0 * (abs(v[np.searchsorted(r, 0.5)]) + 1.0))
+           if enu is None else enu - 5.0)
+    ehi = 10.0
+    for _ in range(200):
+        em = 0.5 * (elo + ehi)
+        nodes, u = shoot(em)
+        if nodes > nodes_target:
+            ehi = em
+        elif nodes < nodes_target:
+            elo = em
+        else:
+            # right node count: tail sign decides (decaying solution has
+            # u(rmax) -> 0; too-high E adds a node / flips the tail)
+            if u[-1] * u[np.argmax(np.abs(u))] > 0:
+                elo = em
+            else:
+                ehi = em
+        if ehi - elo < tol:
+            break
+    E = 0.5 * (elo + ehi)
+    _, u = shoot(E)
+    R = u / rg
+    nrm = np.trapezoid(u * u, rg)
+    R = R / np.sqrt(abs(nrm))
+    if R[np.argmax(np.abs(R[:2000]))] < 0:
+        R = -R
+    Rr = CubicSpline(rg, R)(np.clip(r, rg[0], rg[-1]))
+    return E, Rr

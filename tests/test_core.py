@@ -288,3 +288,24 @@ def test_atomic_lda_helium():
     assert r["converged"]
     assert abs(r["levels"][(1, 0)] - (-0.570425)) < 2e-3
     assert abs(r["etot"] - (-2.8348)) < 5e-3
+
+
+def test_radial_solver_scalar_relativistic():
+    """Koelling-Harmon shooting solver: the H 1s relativistic shift
+    matches perturbation theory (−5α²/8 = −6.66e-6 Ha) and the Z=30 1s
+    level lands on the Dirac value (−455.4 Ha) to <0.1%."""
+    import numpy as np
+    from sirius_amd.core.radial_solver import bound_state_sr
+
+    r = np.geomspace(1e-7, 50.0, 1500)
+    v = -1.0 / r
+    e_nr, _ = bound_state_sr(r, v, 1, 0, rel=False)
+    e_sr, _ = bound_state_sr(r, v, 1, 0, rel=True)
+    assert abs(e_nr - (-0.5)) < 5e-6
+    assert abs((e_sr - e_nr) - (-6.66e-6)) < 1.5e-6
+
+    r2 = np.geomspace(1e-7, 5.0, 1500)
+    e30, _ = bound_state_sr(r2, -30.0 / r2, 1, 0, rel=True)
+    c = 137.035999084
+    e_dirac = c * c * (np.sqrt(1 - (30.0 / c) ** 2) - 1.0)
+    assert abs(e30 - e_dirac) / abs(e_dirac) < 1e-3, (e30, e_dirac)
