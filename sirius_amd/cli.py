@@ -107,9 +107,20 @@ def main(argv=None):
     if args.forces:
         f = dft.forces()
         out["ground_state"]["forces"] = f["total"].tolist()
+        if args.verbosity >= 1:
+            print("total forces [Ha/bohr]:")
+            for ia, row in enumerate(f["total"]):
+                print(f"  atom {ia:3d}: "
+                      + "  ".join(f"{x:+.8f}" for x in row))
     if args.stress:
         st = dft.stress()
         out["ground_state"]["stress"] = st["total"].T.tolist()
+        if args.verbosity >= 1:
+            print("stress tensor [Ha/bohr^3]:")
+            for row in st["total"].T:
+                print("  " + "  ".join(f"{x:+.8f}" for x in row))
+            p3 = -sum(st["total"][i][i] for i in range(3)) / 3.0
+            print(f"  pressure: {p3 * 29421.02648438959:+.4f} kbar")
     path = args.output or f"output_{int(time.time())}.json"
     with open(path, "w") as f:
         json.dump(out, f, indent=2)
