@@ -120,7 +120,8 @@ def main(argv=None):
             for row in st["total"].T:
                 print("  " + "  ".join(f"{x:+.8f}" for x in row))
             p3 = -sum(st["total"][i][i] for i in range(3)) / 3.0
-            print(f"  pressure: {p3 * 29421.02648438959:+.4f} kbar")
+            # au2kbar = 2.94210119e5 (reference src/geometry/stress.cpp:557)
+            print(f"  pressure: {p3 * 2.94210119e5:+.4f} kbar")
     path = args.output or f"output_{int(time.time())}.json"
     with open(path, "w") as f:
         json.dump(out, f, indent=2)

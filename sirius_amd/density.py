@@ -411,6 +411,13 @@ class Density:
 
         import os as _os
         if getattr(cfg_mixer, "use_hartree", False) \
+                and not _os.environ.get("SIRIUS_AMD_USE_HARTREE_METRIC"):
+            import sys
+            print("[sirius_amd] warning: mixer.use_hartree requested by the "
+                  "deck but gated off (set SIRIUS_AMD_USE_HARTREE_METRIC=1 "
+                  "to enable the Coulomb mixing metric); mixing with the "
+                  "plain L2 metric instead", file=sys.stderr)
+        if getattr(cfg_mixer, "use_hartree", False) \
                 and _os.environ.get("SIRIUS_AMD_USE_HARTREE_METRIC"):
             # Coulomb metric over the COARSE G set only, no 4π
             # (periodic_function_property_modified(use_coarse_gvec=true),

@@ -26,9 +26,13 @@ def occupancy(kind: str, x: np.ndarray, w: float) -> np.ndarray:
         return 0.5 * (1.0 + erf(z)) + np.where(z * z > 200, 0.0,
                                                np.exp(-np.clip(z * z, None, 200)) / math.sqrt(2 * math.pi))
     if kind == "methfessel_paxton":  # order 1
-        # f = f_gauss + A1 H1(t) exp(-t^2), A1 = -1/(4 sqrt(pi)), H1 = 2t
+        # Reference: smearing.cpp methfessel_paxton::occupancy (z = -x/w):
+        # f = 0.5(1-erf(z)) + A1 H1(z) exp(-z^2), A1 = -1/(4 sqrt(pi)),
+        # H1(z) = 2z.  With t = x/w = -z this is
+        # f = 0.5(1+erf(t)) + t exp(-t^2)/(2 sqrt(pi)).
         g = 0.5 * (1.0 + erf(t))
-        return g - t * np.exp(-np.clip(t * t, None, 200)) / (2.0 * math.sqrt(math.pi))
+        f = g + t * np.exp(-np.clip(t * t, None, 200)) / (2.0 * math.sqrt(math.pi))
+        return np.where(f < 1e-30, 0.0, f)
     raise ValueError(f"unknown smearing: {kind}")
 
 
@@ -48,8 +52,25 @@ def entropy(kind: str, x: np.ndarray, w: float) -> np.ndarray:
         z2 = np.clip(z * z, None, 200)
         return -np.exp(-z2) * (w - math.sqrt(2.0) * w * t) / (2.0 * math.sqrt(math.pi))
     if kind == "methfessel_paxton":
-        h2 = 4.0 * t * t - 2.0
-        return -np.exp(-np.clip(t * t, None, 200)) * (h2 / 4.0 + 0.5) * w / (2.0 * math.sqrt(math.pi)) * 2.0
+        # Exact port of the reference series (smearing.cpp
+        # methfessel_paxton::entropy, n=1 — QE w1gauss form).  NOTE: the
+        # reference does NOT multiply by the width here (unlike its
+        # gaussian/FD entropies); we reproduce that behavior for parity.
+        arg = np.clip(t * t, None, 200.0)
+        S = -0.5 * np.exp(-arg) / math.sqrt(math.pi)
+        hd = np.zeros_like(t)
+        hp = np.exp(-arg)
+        ni = 0
+        a = 1.0 / math.sqrt(math.pi)
+        for i in range(1, 2):  # n = 1
+            hd = 2.0 * t * hp - 2.0 * ni * hd
+            ni += 1
+            hpm1 = hp
+            hp = 2.0 * t * hd - 2.0 * ni * hp
+            ni += 1
+            a = -a / (i + 4.0)
+            S = S - a * (0.5 * hp + ni * hpm1)
+        return S
     raise ValueError(f"unknown smearing: {kind}")
 
 
