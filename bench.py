@@ -1,11 +1,20 @@
 #!/usr/bin/env python3
 """Flagship benchmark: sec/SCF-iteration of the plane-wave DFT engine.
 
-Config (BASELINE.json): Si N-atom supercell, norm-conserving PP-PW, LDA,
-k-mesh distributed over GPUs (k-point parallelism over RCCL/xGMI).
-Synthetic Si-like NC pseudopotential and random-init wavefunctions (no
-network for real UPF files); fp64 (complex128) throughout — the
-reference's working precision.
+Configs map to BASELINE.json's named configs (synthetic species — no
+network for real UPF files — with the same shapes/cutoffs/SCF work):
+
+  --model sto-uspp  (default)  BASELINE config 2: SrTiO3-shaped 5-atom
+                    ultrasoft PP-PW cell, 4x4x4 k-mesh (augmentation +
+                    Q-operator work in every timed step); 1 GPU headline,
+                    k-point parallel at N>1.
+  --model fe-paw    BASELINE config 4: Fe-bcc-shaped PAW, collinear spin,
+                    12x12x12 k-mesh, k-point parallel over RCCL/xGMI.
+  --model si512     BASELINE config 3: Si 512-atom supercell, NC, Γ-only.
+  --model si2       BASELINE config 1: Si 2-atom diamond, NC, Γ-only.
+  --model si64      round-1 trajectory config (Si 64-atom, 2x2x2 k).
+
+fp64 (complex128) throughout — the reference's working precision.
 
 Contract: one timed step = one full SCF iteration (Davidson diagonalize
 + occupancies + density + mix + potential). Rank 0 prints ONE JSON line.
@@ -22,13 +31,21 @@ import json
 import os
 import time
 
-import numpy as np
 import torch
 
-from sirius_amd.models.synthetic import make_context
+from sirius_amd.models.synthetic import make_named_context
 from sirius_amd.kpoint import KPointSet
 from sirius_amd.dft import DFTGroundState
-from sirius_amd.parallel import init_distributed, get_comm
+from sirius_amd.parallel import init_distributed
+
+
+MODEL_LABEL = {
+    "si2": "Si2-NC-LDA-Gamma",
+    "si64": "Si64-NC-LDA",
+    "si512": "Si512-NC-LDA-Gamma",
+    "sto-uspp": "SrTiO3-USPP-4x4x4k",
+    "fe-paw": "Fe-bcc-PAW-collinear-12x12x12k",
+}
 
 
 def main():
@@ -36,10 +53,11 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=4)
     ap.add_argument("--warmup", type=int, default=4)
-    ap.add_argument("--natoms", type=int, default=64)
-    ap.add_argument("--gk-cutoff", type=float, default=5.0)
-    ap.add_argument("--pw-cutoff", type=float, default=14.0)
-    ap.add_argument("--ngridk", type=int, nargs=3, default=[2, 2, 2])
+    ap.add_argument("--model", type=str, default="sto-uspp",
+                    choices=sorted(MODEL_LABEL))
+    ap.add_argument("--gk-cutoff", type=float, default=None)
+    ap.add_argument("--pw-cutoff", type=float, default=None)
+    ap.add_argument("--ngridk", type=int, nargs=3, default=None)
     ap.add_argument("--device", type=str, default=None)
     args = ap.parse_args()
 
@@ -52,14 +70,21 @@ def main():
     if use_gpu and device.startswith("cuda"):
         torch.cuda.set_device(device)
 
-    ctx = make_context(natoms=args.natoms, device=device,
-                       gk_cutoff=args.gk_cutoff, pw_cutoff=args.pw_cutoff,
-                       ngridk=tuple(args.ngridk))
+    overrides = {}
+    if args.gk_cutoff is not None:
+        overrides["gk_cutoff"] = args.gk_cutoff
+    if args.pw_cutoff is not None:
+        overrides["pw_cutoff"] = args.pw_cutoff
+    if args.ngridk is not None:
+        overrides["ngridk"] = tuple(args.ngridk)
+
+    ctx = make_named_context(args.model, device=device, **overrides)
     kset = KPointSet(ctx)
     dft = DFTGroundState(kset)
 
     if comm.rank == 0:
-        print(f"# natoms={ctx.unit_cell.num_atoms} nbands={ctx.num_bands} "
+        print(f"# model={args.model} natoms={ctx.unit_cell.num_atoms} "
+              f"nbands={ctx.num_bands} nspins={ctx.num_spins} "
               f"nk={kset.num_kpoints} (local {len(kset.kpoints)}) "
               f"nGk~{kset.kpoints[0].num_gkvec if kset.kpoints else 0} "
               f"fine_dims={ctx.fft_fine.dims} device={device}", flush=True)
@@ -113,13 +138,17 @@ def main():
             "dtype": "fp64",
             "data": "synthetic",
             "config": {
-                "model": f"Si{ctx.unit_cell.num_atoms}-NC-LDA",
+                "model": MODEL_LABEL[args.model],
+                "baseline_config": {
+                    "si2": 1, "sto-uspp": 2, "si512": 3, "fe-paw": 4,
+                    "si64": None}[args.model],
                 "natoms": ctx.unit_cell.num_atoms,
                 "num_bands": ctx.num_bands,
-                "ngridk": list(args.ngridk),
+                "num_spins": ctx.num_spins,
+                "ngridk": list(ctx.cfg.parameters.ngridk),
                 "num_kpoints": kset.num_kpoints,
-                "gk_cutoff": args.gk_cutoff,
-                "pw_cutoff": args.pw_cutoff,
+                "gk_cutoff": ctx.gk_cutoff,
+                "pw_cutoff": ctx.pw_cutoff,
                 "parallelism": f"kp{comm.size}",
             },
         }), flush=True)

@@ -16,7 +16,7 @@ import math
 import numpy as np
 from scipy.special import erf
 
-from ..cell import AtomType, BetaProjector, AtomicWf, UnitCell
+from ..cell import AtomType, BetaProjector, AtomicWf, QRadialFunction, UnitCell
 from ..config import Config
 
 
@@ -78,6 +78,172 @@ def make_synthetic_cell(natoms: int) -> UnitCell:
                 for p in _DIAMOND8:
                     pos.append(("Si", (p + [i, j, k]) / n))
     return UnitCell(lat, {"Si": at}, pos)
+
+
+def _normalize_rb(r: np.ndarray, f: np.ndarray) -> np.ndarray:
+    return f / math.sqrt(np.trapezoid(f * f, r))
+
+
+def uspp_atom_type(label: str, zn: float, beta_spec, rc: float = 1.6,
+                   rloc: float = 1.2, q_scale: float = 0.15,
+                   rho_alpha: float = 0.35) -> AtomType:
+    """Synthetic ultrasoft species: Gaussian betas per (l, sigma, d_ion)
+    spec plus Gaussian augmentation Q_ij^l(r) with the UPF storage
+    convention (file stores r^2*Q; small-r behavior r^{l+2}).  Shapes and
+    work match a real USPP (reference: atom_type.cpp:498 read_pseudo_uspp).
+    """
+    at = AtomType(label)
+    at.symbol = label
+    at.zn = zn
+    r = np.geomspace(1e-7, 12.0, 1400)
+    at.r = r
+    at.vloc_r = -(zn / r) * erf(r / rloc)
+
+    dvals = []
+    for (l, sig, dval) in beta_spec:
+        b = r ** (l + 1) * np.exp(-(r ** 2) / (2 * sig ** 2))
+        at.beta.append(BetaProjector(l=l, j=None, f_r=_normalize_rb(r, b)))
+        dvals.append(dval)
+    at.d_ion = np.diag(dvals)
+
+    # augmentation Q_ij^l: all pairs, allowed l of matching parity
+    sq = 0.9 * rc
+    nb = len(at.beta)
+    for j in range(nb):
+        lj = at.beta[j].l
+        for i in range(j + 1):
+            li = at.beta[i].l
+            for l3 in range(abs(li - lj), li + lj + 1):
+                if (li + lj + l3) % 2 != 0:
+                    continue
+                amp = q_scale if i == j else 0.4 * q_scale
+                f = amp * r ** (l3 + 2) * np.exp(-(r ** 2) / (2 * sq ** 2))
+                # normalize the l=0 diagonal moment to ~amp
+                norm = np.trapezoid(r ** (l3 + 2) * np.exp(-(r ** 2) / (2 * sq ** 2)), r)
+                f = f / norm * amp
+                at.q_radial.append(QRadialFunction(i=i, j=j, l=l3, f_r=f))
+
+    rho = zn * (rho_alpha / math.pi) ** 1.5 * np.exp(-rho_alpha * r ** 2)
+    at.rho_total_4pir2 = 4 * math.pi * r ** 2 * rho
+
+    # LCAO wfs: one per beta channel's l, filled to zn
+    occ_left = zn
+    seen_l = []
+    for (l, sig, _d) in beta_spec:
+        if l in seen_l:
+            continue
+        seen_l.append(l)
+        cap = 2.0 * (2 * l + 1)
+        chi = r ** (l + 1) * np.exp(-(r ** 2) / (2 * (sig + 0.6) ** 2))
+        at.atomic_wfs.append(AtomicWf(n=l + 1, l=l, occ=min(cap, max(occ_left, 0.0)),
+                                      f_r=_normalize_rb(r, chi)))
+        occ_left -= cap
+    at.is_norm_conserving = False
+    at.is_ultrasoft = True
+    return at
+
+
+def paw_like_atom_type(label: str, zn: float, beta_spec, rc: float = 1.8,
+                       **kw) -> AtomType:
+    """Synthetic PAW species: USPP base + AE/PS partial waves differing
+    inside r_cut and a small AE core density (reference:
+    atom_type.cpp:644 read_pseudo_paw)."""
+    at = uspp_atom_type(label, zn, beta_spec, rc=rc, **kw)
+    at.is_paw = True
+    at.is_ultrasoft = False
+    r = at.r
+    at.paw_cutoff_index = int(np.searchsorted(r, 2.5 * rc))
+    at.paw_core_energy = -2.0
+    # AE core: localized Gaussian, 2 electrons' worth, stored as rho_c(r)
+    ac = 6.0
+    at.paw_ae_core = 2.0 * (ac / math.pi) ** 1.5 * np.exp(-ac * r ** 2)
+    at.rho_core_r = 0.05 * np.exp(-r ** 2)  # smooth ps core (NLCC-like)
+    at.core_correction = True
+    occ_map = {0: 2.0, 1: 0.0, 2: 6.0}
+    for b in at.beta:
+        ps = np.asarray(b.f_r)
+        # AE partial wave: ps + oscillatory bump inside r_cut (vanishes at rc)
+        bump = 0.3 * r ** (b.l + 1) * np.exp(-(r / (0.45 * rc)) ** 2) \
+            * np.cos(2.5 * r / rc)
+        ae = ps + np.where(r < rc, bump, 0.0)
+        at.paw_ps_wfs.append(ps)
+        at.paw_ae_wfs.append(ae)
+        at.paw_wf_occ.append(occ_map.get(b.l, 0.0))
+    return at
+
+
+A0_STO = 7.38  # bohr (3.905 A cubic perovskite)
+A0_FE = 5.42   # bohr bcc
+
+
+def make_sto_cell() -> UnitCell:
+    """SrTiO3-shaped 5-atom cubic perovskite with synthetic USPPs
+    (BASELINE config 2 analog; augmentation + Q ops in every SCF step)."""
+    sr = uspp_atom_type("Sr", 10.0, [(0, 1.2, 1.0), (1, 1.3, 0.8)], rc=2.0)
+    ti = uspp_atom_type("Ti", 12.0, [(0, 1.0, 1.1), (1, 1.1, 0.9), (2, 0.9, 1.4)], rc=1.8)
+    o = uspp_atom_type("O", 6.0, [(0, 0.8, 1.2), (1, 0.9, 1.0)], rc=1.3)
+    lat = np.eye(3) * A0_STO
+    pos = [("Sr", np.array([0.0, 0.0, 0.0])),
+           ("Ti", np.array([0.5, 0.5, 0.5])),
+           ("O", np.array([0.5, 0.5, 0.0])),
+           ("O", np.array([0.0, 0.5, 0.5])),
+           ("O", np.array([0.5, 0.0, 0.5]))]
+    return UnitCell(lat, {"Sr": sr, "Ti": ti, "O": o}, pos)
+
+
+def make_fe_cell() -> UnitCell:
+    """Fe-bcc-shaped 1-atom cell, synthetic PAW, collinear spin
+    (BASELINE config 4 analog)."""
+    fe = paw_like_atom_type("Fe", 8.0, [(0, 1.0, 1.0), (2, 0.85, 1.5)], rc=1.7)
+    lat = 0.5 * A0_FE * np.array([[1, 1, -1], [-1, 1, 1], [1, -1, 1]], dtype=float)
+    uc = UnitCell(lat, {"Fe": fe}, [("Fe", np.array([0.0, 0.0, 0.0]))])
+    uc.vector_fields[0] = [0.0, 0.0, 2.0]
+    return uc
+
+
+def make_named_context(model: str, device: str | None = None, **overrides):
+    """BASELINE-named synthetic configs for bench.py:
+
+      si2      — config 1: Si 2-atom diamond, NC, Gamma-only
+      sto-uspp — config 2: SrTiO3-shaped 5-atom USPP, 4x4x4 k
+      si512    — config 3: Si 512-atom supercell, NC, Gamma-only
+      fe-paw   — config 4: Fe-bcc-shaped PAW, collinear, 12x12x12 k
+      si64     — round-1 trajectory config (Si 64-atom, 2x2x2 k)
+    """
+    from ..context import SimulationContext
+
+    if model in ("si2", "si64", "si512"):
+        natoms = {"si2": 2, "si64": 64, "si512": 512}[model]
+        ngridk = {"si2": (1, 1, 1), "si64": (2, 2, 2), "si512": (1, 1, 1)}[model]
+        kw = dict(gk_cutoff=5.0, pw_cutoff=14.0, ngridk=ngridk)
+        kw.update(overrides)
+        cfg, _ = make_synthetic_config(natoms=natoms, **kw)
+        uc = make_synthetic_cell(natoms)
+        return SimulationContext(cfg, unit_cell=uc, device=device)
+    if model == "sto-uspp":
+        kw = dict(gk_cutoff=6.0, pw_cutoff=20.0, ngridk=(4, 4, 4))
+        kw.update(overrides)
+        cfg, _ = make_synthetic_config(natoms=5, **kw)
+        return SimulationContext(cfg, unit_cell=make_sto_cell(), device=device)
+    if model == "fe-paw":
+        kw = dict(gk_cutoff=6.0, pw_cutoff=20.0, ngridk=(12, 12, 12),
+                  smearing_width=0.02)
+        kw.update(overrides)
+        cfg = Config({
+            "parameters": {
+                "xc_functionals": ["XC_LDA_X", "XC_LDA_C_PZ"],
+                "gk_cutoff": kw["gk_cutoff"],
+                "pw_cutoff": kw["pw_cutoff"],
+                "ngridk": list(kw["ngridk"]),
+                "smearing": "gaussian",
+                "smearing_width": kw["smearing_width"],
+                "num_mag_dims": 1,
+                "use_symmetry": False,
+            },
+            "mixer": {"type": "anderson", "beta": 0.7},
+        })
+        return SimulationContext(cfg, unit_cell=make_fe_cell(), device=device)
+    raise ValueError(f"unknown bench model: {model}")
 
 
 def make_synthetic_config(natoms: int = 8, gk_cutoff: float = 5.0,

@@ -1,0 +1,44 @@
+"""BASELINE-named synthetic bench models run end-to-end on CPU.
+
+These are the configs bench.py exposes via --model (BASELINE.json
+configs 1-4 analogs, synthetic species).  Quick sanity: SCF iterations
+produce finite energies and the USPP/PAW machinery (augmentation,
+Q-operator, PAW on-site terms) is exercised.
+"""
+
+import math
+
+import pytest
+
+from sirius_amd.models.synthetic import make_named_context
+from sirius_amd.kpoint import KPointSet
+from sirius_amd.dft import DFTGroundState
+
+
+def _run(model, niter=3, **kw):
+    ctx = make_named_context(model, **kw)
+    kset = KPointSet(ctx)
+    dft = DFTGroundState(kset).initial_state()
+    res = dft.find(num_dft_iter=niter)
+    etot = res["energy"]["total"]
+    assert math.isfinite(etot) and etot < 0.0
+    return ctx, res
+
+
+def test_sto_uspp_model():
+    ctx, res = _run("sto-uspp", ngridk=(1, 1, 1))
+    assert ctx.unit_cell.num_atoms == 5
+    # augmentation is active (ultrasoft species)
+    assert any(at.is_ultrasoft for at in ctx.unit_cell.atom_types.values())
+    assert ctx.unit_cell.atom_types["Ti"].num_beta == 3
+
+
+def test_fe_paw_model():
+    ctx, res = _run("fe-paw", ngridk=(1, 1, 1))
+    assert ctx.num_spins == 2
+    assert ctx.unit_cell.atom_types["Fe"].is_paw
+
+
+def test_si2_model():
+    ctx, res = _run("si2")
+    assert ctx.unit_cell.num_atoms == 2
