@@ -14,7 +14,9 @@ import torch
 
 _ext = None
 _ext_zgemm = None
+_ext_radial = None
 _tried = False
+_tried_radial = False
 
 _SRC_DIR = os.path.join(os.path.dirname(os.path.abspath(__file__)), "src")
 _BUILD_DIR = os.path.join(os.path.dirname(os.path.abspath(__file__)), "_build")
@@ -47,7 +49,37 @@ def build_extensions(verbose: bool = False):
         build_directory=zdir,
         verbose=verbose,
     )
+    build_radial(verbose=verbose)
     return _ext
+
+
+def build_radial(verbose: bool = False):
+    """Compile the CPU radial-ODE extension (LAPW radial solver)."""
+    global _ext_radial, _tried_radial
+    _tried_radial = True
+    from torch.utils.cpp_extension import load
+
+    rdir = os.path.join(_BUILD_DIR, "radial")
+    os.makedirs(rdir, exist_ok=True)
+    _ext_radial = load(
+        name="sirius_amd_radial",
+        sources=[os.path.join(_SRC_DIR, "radial_ode.cpp")],
+        extra_cflags=["-O3"],
+        build_directory=rdir,
+        verbose=verbose,
+    )
+    return _ext_radial
+
+
+def get_radial():
+    """The radial-ODE solver extension (CPU; required for the FP-LAPW
+    branch everywhere — raises if it cannot build)."""
+    global _ext_radial
+    if _ext_radial is None and not _tried_radial:
+        build_radial()
+    if _ext_radial is None:
+        raise RuntimeError("sirius_amd radial-ODE extension failed to build")
+    return _ext_radial
 
 
 def get_ext(required: bool | None = None):
