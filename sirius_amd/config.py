@@ -56,6 +56,8 @@ _DEFAULTS: dict[str, dict[str, Any]] = {
         "vk": [],
         "gamma_point": False,
         "nn_radius": -1,
+        "molecule": False,
+        "auto_rmt": 1,
         "hubbard_correction": False,
         "so_correction": False,
     },

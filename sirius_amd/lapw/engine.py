@@ -1,0 +1,971 @@
+"""FP-LAPW ground state engine: density, potential, Hamiltonian, SCF.
+
+Reference behavior (all citations into /root/reference):
+- step function: src/lapw/step_function.hpp
+- initial density: src/density/density.cpp:291 (initial_density_full_pot)
+- MT density: density.cpp:796 (add_k_point_contribution_dm_fplapw),
+  :1561 (reduce_density_matrix), :1600 (generate_valence_mt)
+- Poisson: src/potential/poisson.cpp (pseudo-charge method, Weinert)
+- MT XC: src/potential/xc_mt.cpp
+- Hamiltonian/overlap setup: src/hamiltonian/hamiltonian_k.cpp:299
+  (set_fv_h_o + apw_lo/lo_lo/it blocks), hamiltonian.cpp:91
+  (apply_hmt_to_apw); radial integrals: src/unit_cell/atom.hpp:132
+- dense FV solve: src/hamiltonian/diagonalize_fp.hpp:29
+- fv states: src/k_point/generate_fv_states.cpp:21
+- energies: src/dft/energy.cpp (FP branch: E = ekin + exc + vha/2 + enuc)
+
+The muffin-tin basis is one symmetry class per atom; radial integrals
+are evaluated with precomputed cubic-spline quadrature weights so every
+per-iteration integral is a tensor contraction.
+"""
+
+from __future__ import annotations
+
+import math
+import sys
+import time
+
+import numpy as np
+import torch
+from scipy.interpolate import CubicSpline
+
+from ..config import Config
+from ..cell import UnitCell
+from ..context import SimulationContext
+from ..core.ylm import ylm as _ylm, lmmax as _lmmax
+from ..core.radial import sbessel
+from ..kpoint import KPointSet
+from .. import xc as xc_mod
+from .species import FPAtomType
+from .basis import AtomSymmetryClass, Y00
+from .sht import SHT, l_by_lm, rlm_to_ylm, ylm_to_rlm, gaunt_hybrid
+from .matching import MatchingCoefficients
+
+FOURPI = 4.0 * math.pi
+PSEUDO_DENSITY_ORDER = 9   # reference potential.hpp pseudo_density_order_
+
+
+def spline_weights(r: np.ndarray) -> np.ndarray:
+    """w such that w . f = integral of the natural-cubic-spline
+    interpolant of f over [r0, rN] (exact, linear in f)."""
+    n = len(r)
+    eye = np.eye(n)
+    cs = CubicSpline(r, eye, axis=0)
+    return cs.integrate(r[0], r[-1])
+
+
+def make_lapw_context(cfg: Config, base_dir: str = ".", device=None):
+    """Build a SimulationContext with FP-LAPW species."""
+    import os
+
+    ucfg = cfg.unit_cell
+    lat = np.asarray(ucfg.lattice_vectors, dtype=np.float64) \
+        * float(ucfg.lattice_vectors_scale)
+    p = cfg.parameters
+    types = {}
+    for lab in ucfg.atom_types:
+        path = os.path.join(base_dir, ucfg.atom_files[lab])
+        types[lab] = FPAtomType.from_file(
+            lab, path, lmax_apw=int(p.lmax_apw),
+            radial_grid=cfg.settings.radial_grid)
+        # duck-type the PP AtomType flags the shared context checks
+        types[lab].is_paw = False
+        types[lab].is_ultrasoft = False
+        types[lab].is_norm_conserving = False
+        types[lab].spin_orbit = False
+        types[lab].num_beta = 0
+    pos = []
+    atoms_cfg = ucfg.atoms if isinstance(ucfg.atoms, dict) else dict(ucfg.atoms)
+    for lab in ucfg.atom_types:
+        for v in atoms_cfg.get(lab, []):
+            pos.append((lab, np.asarray(v[:3], dtype=np.float64)))
+    uc = UnitCell(lat, types, pos)
+
+    # gk cutoff from rgkmax: aw_cutoff / min R_mt (simulation_context.cpp:292)
+    if float(p.aw_cutoff) > 0:
+        min_rmt = min(at.rmt for at in types.values())
+        cfg._data["parameters"]["gk_cutoff"] = float(p.aw_cutoff) / min_rmt
+        cfg.parameters.gk_cutoff = float(p.aw_cutoff) / min_rmt
+
+    ctx = SimulationContext(cfg, unit_cell=uc, device=device)
+    ctx.full_potential = True
+    ctx.lmax_apw = int(p.lmax_apw)
+    ctx.lmax_rho = int(p.lmax_rho)
+    ctx.lmax_pot = int(p.lmax_pot)
+    ctx.lmmax_rho = _lmmax(ctx.lmax_rho)
+    ctx.lmmax_pot = _lmmax(ctx.lmax_pot)
+    ctx.molecule = bool(getattr(p, "molecule", False))
+    ctx.valence_relativity = str(p.valence_relativity)
+    ctx.core_relativity = str(p.core_relativity)
+    return ctx
+
+
+# ---------------------------------------------------------------- utilities
+class MTGrids:
+    """Per-atom-type cached grids/weights for MT integrals."""
+
+    def __init__(self, ctx):
+        self.w = {}          # spline quadrature weights per type
+        self.r2w = {}        # r^2 * w
+        for lab, at in ctx.unit_cell.atom_types.items():
+            w = spline_weights(at.r)
+            self.w[lab] = w
+            self.r2w[lab] = w * at.r ** 2
+
+
+def mt_inner(ctx, grids: MTGrids, f_mt, g_mt) -> float:
+    """Sum_ia Sum_lm int f_lm g_lm r^2 dr."""
+    tot = 0.0
+    for ia, (lab, _) in enumerate(ctx.unit_cell.atoms):
+        tot += float(np.einsum("lr,lr,r->", f_mt[ia], g_mt[ia],
+                               grids.r2w[lab]))
+    return tot
+
+
+def it_inner(ctx, f_rg, g_rg) -> float:
+    """Interstitial inner product with the step function weight."""
+    n = ctx.fft_fine.size
+    return float((f_rg * g_rg * ctx.theta_rg).sum().real) \
+        * ctx.unit_cell.omega / n
+
+
+# ------------------------------------------------------------ step function
+def init_step_function(ctx):
+    """theta(G) on the fine sphere + theta(r) on the fine grid
+    (reference step_function.hpp:78-129)."""
+    uc = ctx.unit_cell
+    gl = ctx.gvec_fine
+    glen = gl.gk_len
+    out = np.zeros(gl.num_gvec, dtype=np.complex128)
+    tau = uc.atom_positions_frac()
+    m = gl.miller.astype(np.float64)
+    for lab in uc.type_labels:
+        ia = uc.atoms_of_type(lab)
+        if len(ia) == 0:
+            continue
+        R = uc.atom_types[lab].rmt
+        ff = np.where(glen < 1e-12, R ** 3 / 3.0,
+                      (np.sin(glen * R) - glen * R * np.cos(glen * R))
+                      / np.maximum(glen, 1e-12) ** 3)
+        ph = np.exp(-2j * math.pi * (tau[ia] @ m.T)).sum(axis=0)
+        out -= FOURPI / uc.omega * ff * ph
+    iz = gl.index_of_zero()
+    out[iz] += 1.0
+    theta_pw = torch.from_numpy(out).to(ctx.device)
+    theta_rg = ctx.fft_fine.to_real(theta_pw).real
+    vit = float(theta_rg.sum()) * uc.omega / ctx.fft_fine.size
+    vmt = sum(FOURPI / 3 * uc.atom_types[lab].rmt ** 3 for lab, _ in uc.atoms)
+    if abs(vit - (uc.omega - vmt)) > 1e-8 * uc.omega:
+        print(f"[sirius_amd] warning: step function IT volume error "
+              f"{vit - (uc.omega - vmt):.3e}", file=sys.stderr)
+    ctx.theta_pw = theta_pw
+    ctx.theta_rg = theta_rg
+    ctx.volume_it = uc.omega - vmt
+
+
+# --------------------------------------------------- surface sums (fg*fl*yg)
+def sum_fg_fl_yg(ctx, lmax: int, fpw: np.ndarray, fl: dict) -> np.ndarray:
+    """flm[lm, ia] = Sum_G 4pi i^l f_l(|G|, type) f(G) Y*_lm(G^) e^{iG tau}
+    (reference src/lapw/sum_fg_fl_yg.hpp).  fl: label -> [lmax+1, nG]."""
+    uc = ctx.unit_cell
+    gl = ctx.gvec_fine
+    nlm = _lmmax(lmax)
+    flm = np.zeros((nlm, uc.num_atoms), dtype=np.complex128)
+    ylm_c = ctx.gvec_ylm_fine[:, :nlm].conj()          # [nG, lm]
+    zil = np.array([1j ** l for l in range(lmax + 1)])
+    lbl = l_by_lm(lmax)
+    tau = uc.atom_positions_frac()
+    m = gl.miller.astype(np.float64)
+    for lab in uc.type_labels:
+        ias = uc.atoms_of_type(lab)
+        if len(ias) == 0:
+            continue
+        zl = FOURPI * fl[lab] * zil[:, None] * fpw[None, :]   # [l, nG]
+        zm = zl[lbl, :] * ylm_c.T                              # [lm, nG]
+        ph = np.exp(2j * math.pi * (m @ tau[ias].T))           # [nG, na]
+        flm[:, ias] += zm @ ph
+    return flm
+
+
+# ------------------------------------------------------------------ density
+class FPDensity:
+    def __init__(self, ctx, grids: MTGrids):
+        self.ctx = ctx
+        self.grids = grids
+        uc = ctx.unit_cell
+        ng = ctx.gvec_fine.num_gvec
+        self.rho_pw = torch.zeros(ng, dtype=ctx.dtype, device=ctx.device)
+        self.rho_rg = torch.zeros(ctx.fft_fine.dims, dtype=ctx.rdtype,
+                                  device=ctx.device)
+        self.rho_mt = [np.zeros((ctx.lmmax_rho, uc.atom_types[lab].nmtp))
+                       for lab, _ in uc.atoms]
+        self.mixer = None
+
+    # -- charge bookkeeping -------------------------------------------------
+    def total_charge(self) -> float:
+        ctx = self.ctx
+        it = float((self.rho_rg * ctx.theta_rg).sum()) \
+            * ctx.unit_cell.omega / ctx.fft_fine.size
+        mt = 0.0
+        for ia, (lab, _) in enumerate(ctx.unit_cell.atoms):
+            mt += FOURPI * Y00 * float(self.grids.r2w[lab] @ self.rho_mt[ia][0])
+        return it + mt
+
+    def normalize(self):
+        ne = self.ctx.unit_cell.num_electrons
+        tot = self.total_charge()
+        s = ne / tot
+        self.rho_pw *= s
+        self.rho_rg *= s
+        for ia in range(len(self.rho_mt)):
+            self.rho_mt[ia] *= s
+
+    # -- initial density ----------------------------------------------------
+    def initial(self):
+        """Superposition of free-atom densities (density.cpp:291)."""
+        ctx = self.ctx
+        uc = ctx.unit_cell
+        gl = ctx.gvec_fine
+        # smoothed free-atom density -> PW via shell form factors
+        ff = {}
+        for lab, at in uc.atom_types.items():
+            r = at.free_atom_r
+            rho = at.free_atom_rho.copy()
+            R = at.rmt
+            irmt = np.searchsorted(r, R)
+            irmt = min(irmt, len(r) - 1)
+            Rg = r[irmt]
+            sm = rho * 0.5 * (1 + np.array(
+                [math.erf((x / Rg - 0.5) * 10) for x in r]))
+            mask = r <= Rg
+            smooth = np.where(mask, sm, rho)
+            q = gl.shell_len
+            x = np.outer(r, q)
+            j0 = np.where(x > 1e-12, np.sin(x) / np.maximum(x, 1e-300), 1.0)
+            integ = smooth[:, None] * j0 * (r ** 2)[:, None]
+            ff[lab] = CubicSpline(r, integ, axis=0).integrate(r[0], r[-1])
+        v = ctx.make_periodic_function(ff)          # (4pi/omega) sum ff e^{-iG tau}
+        self.rho_pw = v.clone()
+        self.rho_rg = ctx.fft_fine.to_real(self.rho_pw).real.clamp(min=0.0)
+        self.rho_pw = ctx.fft_fine.to_pw(self.rho_rg.to(ctx.dtype))
+
+        # MT: Y00 channel from the (unsmoothed) free atom density
+        for ia, (lab, _) in enumerate(uc.atoms):
+            at = uc.atom_types[lab]
+            self.rho_mt[ia][:] = 0.0
+            self.rho_mt[ia][0] = at.free_atom_density(at.r) / Y00
+
+        # boundary match: flm at R from the PW side minus free-atom value
+        lmax = ctx.lmax_rho
+        fl = {lab: ctx.sbessel_mt[lab][:lmax + 1] for lab in uc.type_labels}
+        flm = sum_fg_fl_yg(ctx, lmax, self.rho_pw.cpu().numpy(), fl)
+        lbl = l_by_lm(lmax)
+        for ia, (lab, _) in enumerate(uc.atoms):
+            at = uc.atom_types[lab]
+            f = flm[:, ia].copy()
+            f[0] -= at.free_atom_density(at.rmt) / Y00
+            glm = ylm_to_rlm(f)
+            rr2 = (at.r / at.rmt) ** 2
+            self.rho_mt[ia] += glm[:, None] * rr2[None, :]
+        self.normalize()
+
+    # -- valence + core -----------------------------------------------------
+    def generate(self, kset, engine, symmetrize: bool = True):
+        ctx = self.ctx
+        uc = ctx.unit_cell
+        # interstitial: accumulate |psi(r)|^2 on the coarse grid
+        coarse = ctx.fft_coarse
+        rho_c = torch.zeros(coarse.dims, dtype=ctx.rdtype, device=ctx.device)
+        dm = [np.zeros((uc.atom_types[lab].mt_basis_size,
+                        uc.atom_types[lab].mt_basis_size, ctx.num_spins),
+                       dtype=np.complex128) for lab, _ in uc.atoms]
+        for kp in kset.kpoints:
+            # map G+k sphere coefficients into the coarse FFT grid
+            for ispn in range(ctx.num_spins):
+                occ = kp.occ[ispn]
+                nocc = int(np.sum(occ > 1e-12))
+                if nocc == 0:
+                    continue
+                w = torch.from_numpy(occ[:nocc] * kp.weight).to(ctx.device)
+                psi = kp.psi[ispn, :nocc, :kp.num_gkvec]
+                psir = kp.fft.to_real(psi)
+                rho_c += torch.einsum(
+                    "b,bxyz->xyz", w.to(ctx.rdtype),
+                    (psir.real ** 2 + psir.imag ** 2))
+                # MT density matrix
+                mt = kp.mt_coeffs  # list per atom: [nfv, mt_basis]
+                for ia, (lab, _) in enumerate(uc.atoms):
+                    c = mt[ia][:nocc]
+                    wocc = (occ[:nocc] * kp.weight)[:, None]
+                    dm[ia][:, :, ispn] += (c.conj() * wocc).T @ c
+        rho_c *= 1.0 / uc.omega
+        if kset.comm.active:
+            kset.comm.allreduce_(rho_c)
+            for ia in range(uc.num_atoms):
+                t = torch.from_numpy(dm[ia])
+                kset.comm.allreduce_(t)
+                dm[ia] = t.numpy()
+        self.dm = dm
+
+        # coarse real grid -> fine PW
+        rho_pw_c = ctx.fft_coarse.to_pw(rho_c.to(ctx.dtype))
+        rho_pw = torch.zeros(ctx.gvec_fine.num_gvec, dtype=ctx.dtype,
+                             device=ctx.device)
+        rho_pw[ctx.coarse_to_fine] = rho_pw_c
+        if symmetrize and ctx.symmetry is not None:
+            from ..symmetry import symmetrize_rho_g
+            rho_pw = symmetrize_rho_g(rho_pw, ctx.gvec_fine, ctx.symmetry.ops)
+        self.rho_pw = rho_pw
+        self.rho_rg = ctx.fft_fine.to_real(rho_pw).real
+
+        # MT density from the density matrix
+        if symmetrize and ctx.symmetry is not None:
+            dm = engine.symmetrize_mt_dm(dm)
+            self.dm = dm
+        for ia, (lab, _) in enumerate(uc.atoms):
+            at = uc.atom_types[lab]
+            asc = engine.classes[ia]
+            self.rho_mt[ia][:] = self._mt_density_one(at, asc, dm[ia][:, :, 0])
+            if ctx.num_spins == 2:
+                raise NotImplementedError("collinear LAPW density TODO")
+
+        # core
+        for ia, (lab, _) in enumerate(uc.atoms):
+            asc = engine.classes[ia]
+            self.rho_mt[ia][0] += asc.ae_core_density / Y00
+
+    def _mt_density_one(self, at, asc, zdens):
+        """reduce_density_matrix + expand: rho_lm(r)
+        (density.cpp:1561-1734)."""
+        ctx = self.ctx
+        G = engine_gaunt(ctx, at)            # [L1, L3, L2] <Y|R|Y>
+        nrf = at.num_rf
+        lmmax = ctx.lmmax_rho
+        # mt_dm[lm3, pair]
+        mt_dm = np.zeros((lmmax, nrf * (nrf + 1) // 2))
+        for i2 in range(nrf):
+            l2 = at.indexr[i2][0]
+            xi2_0 = _first_xi(at, i2)
+            for i1 in range(i2 + 1):
+                l1 = at.indexr[i1][0]
+                xi1_0 = _first_xi(at, i1)
+                offs = i2 * (i2 + 1) // 2 + i1
+                # sum over m1, m2 with gaunt
+                blk = zdens[xi1_0:xi1_0 + 2 * l1 + 1, xi2_0:xi2_0 + 2 * l2 + 1]
+                lm1 = at.indexb[xi1_0][2]
+                lm2 = at.indexb[xi2_0][2]
+                g = G[lm1:lm1 + 2 * l1 + 1, :lmmax, lm2:lm2 + 2 * l2 + 1]
+                mt_dm[:, offs] += np.real(np.einsum("ab,acb->c", blk, g))
+        # rho_lm(r) = sum_pairs mt_dm * u_i1 u_i2 * (2 - delta)
+        pair_f = np.empty((nrf * (nrf + 1) // 2, at.nmtp))
+        for i2 in range(nrf):
+            for i1 in range(i2 + 1):
+                n = 1.0 if i1 == i2 else 2.0
+                pair_f[i2 * (i2 + 1) // 2 + i1] = n * asc.u[i1] * asc.u[i2]
+        return mt_dm @ pair_f
+
+    # -- mixing -------------------------------------------------------------
+    def mixer_init(self, cfg_mixer):
+        from ..mixer import Component, Linear, Anderson, Broyden2, AndersonStable
+
+        ctx = self.ctx
+        omega = ctx.unit_cell.omega
+
+        def inner_pw(x, y):
+            return omega * float(torch.vdot(x, y).real)
+
+        comps = [Component("rho_pw", inner=inner_pw)]
+
+        def make_inner_mt(lab):
+            r2w = torch.from_numpy(self.grids.r2w[lab])
+
+            def inner_mt(x, y):
+                return float(torch.einsum("lr,lr,r->", x, y, r2w))
+            return inner_mt
+
+        for ia, (lab, _) in enumerate(ctx.unit_cell.atoms):
+            comps.append(Component(f"rho_mt_{ia}", inner=make_inner_mt(lab)))
+        kind = cfg_mixer.type
+        cls = {"linear": Linear, "anderson": Anderson, "broyden2": Broyden2,
+               "anderson_stable": AndersonStable}[kind]
+        self.mixer = cls(comps, max_history=int(cfg_mixer.max_history),
+                         beta=float(cfg_mixer.beta))
+        self.mixer.initialize(self._mix_value())
+
+    def _mix_value(self):
+        v = {"rho_pw": self.rho_pw.clone()}
+        for ia in range(len(self.rho_mt)):
+            v[f"rho_mt_{ia}"] = torch.from_numpy(self.rho_mt[ia].copy())
+        return v
+
+    def _set_from_mix(self, v):
+        self.rho_pw = v["rho_pw"].clone()
+        for ia in range(len(self.rho_mt)):
+            self.rho_mt[ia] = v[f"rho_mt_{ia}"].numpy().copy()
+        self.rho_rg = self.ctx.fft_fine.to_real(self.rho_pw).real
+
+    def mix(self) -> float:
+        self.mixer.set_input(self._mix_value())
+        rms = self.mixer.mix()
+        self._set_from_mix(self.mixer.get_output())
+        return rms
+
+
+def _first_xi(at, idxrf):
+    """First basis-function index of radial function idxrf."""
+    xi = 0
+    for i in range(idxrf):
+        l = at.indexr[i][0]
+        xi += 2 * l + 1
+    return xi
+
+
+_gaunt_cache = {}
+
+
+def engine_gaunt(ctx, at):
+    key = (at.lmax_apw, ctx.lmax_rho)
+    if key not in _gaunt_cache:
+        lmax_b = max(at.lmax_apw, max((lo.l for lo in at.lo_descriptors),
+                                      default=0))
+        _gaunt_cache[key] = gaunt_hybrid(lmax_b, max(ctx.lmax_rho, ctx.lmax_pot),
+                                         lmax_b)
+    return _gaunt_cache[key]
+
+
+# ---------------------------------------------------------------- potential
+class FPPotential:
+    def __init__(self, ctx, grids: MTGrids):
+        self.ctx = ctx
+        self.grids = grids
+        uc = ctx.unit_cell
+        ng = ctx.gvec_fine.num_gvec
+        self.vha_pw = torch.zeros(ng, dtype=ctx.dtype, device=ctx.device)
+        self.vha_rg = torch.zeros(ctx.fft_fine.dims, dtype=ctx.rdtype,
+                                  device=ctx.device)
+        self.vha_mt = [np.zeros((ctx.lmmax_pot, uc.atom_types[lab].nmtp))
+                       for lab, _ in uc.atoms]
+        self.vxc_rg = torch.zeros_like(self.vha_rg)
+        self.exc_rg = torch.zeros_like(self.vha_rg)
+        self.vxc_mt = [np.zeros_like(v) for v in self.vha_mt]
+        self.exc_mt = [np.zeros_like(v) for v in self.vha_mt]
+        self.veff_rg = torch.zeros_like(self.vha_rg)
+        self.veff_mt = [np.zeros_like(v) for v in self.vha_mt]
+        self.veff_pw = torch.zeros(ng, dtype=ctx.dtype, device=ctx.device)
+        self.vh_el = np.zeros(uc.num_atoms)
+        self.energy_vha = 0.0
+        self.sht = SHT(max(ctx.lmax_rho, ctx.lmax_pot))
+
+        # tables: j_l(GR), moments, gamma factors (potential.cpp:146-213)
+        gl = ctx.gvec_fine
+        glen = gl.gk_len
+        lmax = max(ctx.lmax_rho, ctx.lmax_pot) + PSEUDO_DENSITY_ORDER + 1
+        ctx.sbessel_mt = {}
+        self.sbessel_mom = {}
+        self.gamma_factors_R = {}
+        for lab, at in uc.atom_types.items():
+            R = at.rmt
+            tbl = np.empty((lmax + 1, gl.num_gvec))
+            for l in range(lmax + 1):
+                tbl[l] = sbessel(l, glen * R)
+            ctx.sbessel_mt[lab] = tbl
+            mom = np.zeros((ctx.lmax_rho + 1, gl.num_gvec))
+            nz = glen > 1e-12
+            for l in range(ctx.lmax_rho + 1):
+                mom[l, nz] = R ** (l + 2) * tbl[l + 1, nz] / glen[nz]
+            iz = gl.index_of_zero()
+            if iz >= 0:
+                mom[0, iz] = R ** 3 / 3.0
+            self.sbessel_mom[lab] = mom
+            gf = np.empty(ctx.lmax_rho + 1)
+            for l in range(ctx.lmax_rho + 1):
+                Rl = R ** l
+                n_min = 2 * l + 3
+                n_max = (2 * l + 1) + (2 * PSEUDO_DENSITY_ORDER + 2)
+                f1 = 1.0
+                f2 = 1.0
+                for n in range(n_min, n_max + 1, 2):
+                    if f1 < Rl:
+                        f1 *= n / 2.0
+                    else:
+                        f2 *= n / 2.0
+                gf[l] = (f1 / Rl) * f2
+            self.gamma_factors_R[lab] = gf
+
+        # Ylm of fine G vectors (shared)
+        if not hasattr(ctx, "gvec_ylm_fine"):
+            gk = gl.gkvec_cart
+            rr = np.linalg.norm(gk, axis=1)
+            with np.errstate(invalid="ignore"):
+                th = np.where(rr > 1e-12,
+                              np.arccos(np.clip(gk[:, 2] / np.maximum(rr, 1e-300), -1, 1)), 0.0)
+                ph = np.where(rr > 1e-12, np.arctan2(gk[:, 1], gk[:, 0]), 0.0)
+            ctx.gvec_ylm_fine = _ylm(max(ctx.lmax_rho, ctx.lmax_pot), th, ph)
+
+    # -- radial MT Poisson (potential.hpp:298-386 poisson_vmt) --------------
+    def _poisson_vmt_one(self, at, rho_mt):
+        """Returns (vha_mt [lmmax_pot, nr], qmt_rlm [lmmax_rho])."""
+        ctx = self.ctx
+        r = at.r
+        lbl = l_by_lm(ctx.lmax_rho)
+        qmt = np.zeros(ctx.lmmax_rho)
+        v = np.zeros((ctx.lmmax_pot, at.nmtp))
+        R = at.rmt
+        for lm in range(ctx.lmmax_rho):
+            l = int(lbl[lm])
+            rl = rho_mt[lm]
+            f1 = rl * r ** (l + 2)
+            g1 = CubicSpline(r, f1).antiderivative()(r)
+            qmt[lm] = g1[-1]
+            if lm < ctx.lmmax_pot:
+                f2 = rl * r ** (1 - l)
+                g2 = CubicSpline(r, f2).antiderivative()(r)
+                d1 = 1.0 / R ** (2 * l + 1)
+                vlm = (1.0 - (r / R) ** (2 * l + 1)) * g1 / r ** (l + 1) \
+                    + (g2[-1] - g2) * r ** l \
+                    - (g1[-1] - g1) * r ** l * d1
+                v[lm] = vlm * FOURPI / (2 * l + 1)
+        # nuclear: +zn/R (|_VHA_AUX off branch); -zn/r is applied later
+        v[0] += at.zn / R / Y00
+        qmt[0] -= at.zn * Y00
+        return v, qmt
+
+    # -- generate -----------------------------------------------------------
+    def generate(self, density: FPDensity):
+        ctx = self.ctx
+        uc = ctx.unit_cell
+        gl = ctx.gvec_fine
+        glen = gl.gk_len
+        iz = gl.index_of_zero()
+
+        # 1) MT Poisson -> true multipole moments (complex Ylm)
+        qmt_c = np.zeros((ctx.lmmax_rho, uc.num_atoms), dtype=np.complex128)
+        for ia, (lab, _) in enumerate(uc.atoms):
+            at = uc.atom_types[lab]
+            v, qmt_r = self._poisson_vmt_one(at, density.rho_mt[ia])
+            self.vha_mt[ia][:] = v
+            qmt_c[:, ia] = rlm_to_ylm(qmt_r)
+
+        # 2) interstitial moments inside MT spheres
+        rho_pw = density.rho_pw.cpu().numpy()
+        fl = {lab: self.sbessel_mom[lab] for lab in uc.type_labels}
+        qit = sum_fg_fl_yg(ctx, ctx.lmax_rho, rho_pw, fl)
+
+        # 3) pseudo-charge PW correction (poisson.cpp:39-152)
+        rho_mod = rho_pw.copy()
+        lbl = l_by_lm(ctx.lmax_rho)
+        tau = uc.atom_positions_frac()
+        m = gl.miller.astype(np.float64)
+        for lab in uc.type_labels:
+            ias = uc.atoms_of_type(lab)
+            at = uc.atom_types[lab]
+            R = at.rmt
+            qa = qmt_c[:, ias] - qit[:, ias]                     # [lm, na]
+            pf = np.exp(2j * math.pi * (m @ tau[ias].T))         # [nG, na]
+            qapf = qa @ pf.conj().T                              # [lm, nG]
+            nz = glen > 1e-12
+            gR = glen[nz] * R
+            gRn = (2.0 / gR) ** (PSEUDO_DENSITY_ORDER + 1)
+            ylm = ctx.gvec_ylm_fine[nz][:, :ctx.lmmax_rho]       # [nG, lm]
+            zil_conj = np.array([np.conj(1j ** l)
+                                 for l in range(ctx.lmax_rho + 1)])
+            contrib = np.zeros(nz.sum(), dtype=np.complex128)
+            for l in range(ctx.lmax_rho + 1):
+                sel = lbl == l
+                zt1 = np.einsum("gm,mg->g", ylm[:, sel], qapf[sel][:, nz])
+                jl = ctx.sbessel_mt[lab][l + PSEUDO_DENSITY_ORDER + 1][nz]
+                contrib += (FOURPI / uc.omega) * zil_conj[l] * zt1 \
+                    * self.gamma_factors_R[lab][l] * jl * gRn
+            rho_mod[nz] += contrib
+            rho_mod[iz] += FOURPI / uc.omega * Y00 \
+                * np.real(qmt_c[0, ias] - qit[0, ias]).sum()
+
+        # 4) Hartree PW
+        vh = np.zeros_like(rho_mod)
+        nz = glen > 1e-12
+        if ctx.molecule:
+            R_cut = 0.5 * uc.omega ** (1.0 / 3)
+            vh[nz] = FOURPI * rho_mod[nz] / glen[nz] ** 2 \
+                * (1.0 - np.cos(glen[nz] * R_cut))
+        else:
+            vh[nz] = FOURPI * rho_mod[nz] / glen[nz] ** 2
+        vh[iz] = 0.0
+        self.vha_pw = torch.from_numpy(vh).to(ctx.device)
+
+        # 5) MT boundary values -> homogeneous solution
+        fl_mt = {lab: ctx.sbessel_mt[lab][:ctx.lmax_pot + 1]
+                 for lab in uc.type_labels}
+        vmtlm = sum_fg_fl_yg(ctx, ctx.lmax_pot, vh, fl_mt)
+        lbl_p = l_by_lm(ctx.lmax_pot)
+        for ia, (lab, _) in enumerate(uc.atoms):
+            at = uc.atom_types[lab]
+            vlm = ylm_to_rlm(vmtlm[:, ia])
+            rRl = (at.r[None, :] / at.rmt) ** lbl_p[:, None]
+            self.vha_mt[ia] += vlm[:, None] * rRl
+            self.vh_el[ia] = Y00 * self.vha_mt[ia][0, 0]
+            # add nucleus -z/r
+            self.vha_mt[ia][0] -= at.zn / at.r / Y00
+
+        self.vha_rg = ctx.fft_fine.to_real(self.vha_pw).real
+
+        # 6) energy_vha = <rho|V_H> with the complete potential
+        grids = self.grids
+        self.energy_vha = it_inner(ctx, density.rho_rg, self.vha_rg) \
+            + mt_inner(ctx, grids, density.rho_mt, self.vha_mt)
+
+        # 7) XC
+        self._xc(density)
+
+        # 8) effective potential
+        self.veff_rg = self.vha_rg + self.vxc_rg
+        for ia in range(uc.num_atoms):
+            self.veff_mt[ia] = self.vha_mt[ia] + self.vxc_mt[ia]
+        vtheta = (self.veff_rg * ctx.theta_rg).to(ctx.dtype)
+        self.veff_pw = ctx.fft_fine.to_pw(vtheta)
+
+        self.energy_veff = it_inner(ctx, density.rho_rg, self.veff_rg) \
+            + mt_inner(ctx, grids, density.rho_mt, self.veff_mt)
+        self.energy_exc = it_inner(ctx, density.rho_rg, self.exc_rg) \
+            + mt_inner(ctx, grids, density.rho_mt, self.exc_mt)
+        self.energy_enuc = -0.5 * sum(
+            uc.atom_types[lab].zn * self.vh_el[ia]
+            for ia, (lab, _) in enumerate(uc.atoms))
+
+    def _xc(self, density: FPDensity):
+        ctx = self.ctx
+        if ctx.is_gga:
+            raise NotImplementedError("LAPW GGA TODO")
+        eps, vx, _ = xc_mod.evaluate(ctx.xc_names,
+                                     density.rho_rg.clamp(min=0.0))
+        self.exc_rg = eps
+        self.vxc_rg = vx
+        sht = self.sht
+        for ia, (lab, _) in enumerate(ctx.unit_cell.atoms):
+            rho_tp = sht.rlm_backward[:, :ctx.lmmax_rho] @ density.rho_mt[ia]
+            e_tp, v_tp, _ = xc_mod.evaluate(
+                ctx.xc_names, torch.from_numpy(rho_tp.clip(min=0.0)))
+            self.exc_mt[ia] = (sht.rlm_forward[:ctx.lmmax_pot]
+                               @ e_tp.numpy())
+            self.vxc_mt[ia] = (sht.rlm_forward[:ctx.lmmax_pot]
+                               @ v_tp.numpy())
+
+    def update_atomic_potential(self, classes):
+        for ia, asc in enumerate(classes):
+            vs = Y00 * self.veff_mt[ia][0]
+            asc.set_spherical_potential(vs)
+
+
+# ------------------------------------------------------------- ground state
+class FPGroundState:
+    def __init__(self, kset: KPointSet):
+        self.kset = kset
+        self.ctx = kset.ctx
+        ctx = self.ctx
+        uc = ctx.unit_cell
+        if not hasattr(ctx, "theta_pw"):
+            init_step_function(ctx)
+        self.grids = MTGrids(ctx)
+        self.potential = FPPotential(ctx, self.grids)
+        self.density = FPDensity(ctx, self.grids)
+        self.classes = [AtomSymmetryClass(uc.atom_types[lab],
+                                          valence_relativity=ctx.valence_relativity,
+                                          core_relativity=ctx.core_relativity)
+                        for lab, _ in uc.atoms]
+        self.matching = {}
+        for ik, kp in enumerate(kset.kpoints):
+            self.matching[ik] = MatchingCoefficients(ctx, kp.gkvec)
+        # G1-G2 lookup on the fine sphere
+        self._g12 = {}
+        n1, n2, n3 = ctx.fft_fine.dims
+        lut = np.full(n1 * n2 * n3, -1, dtype=np.int64)
+        mm = ctx.gvec_fine.miller
+        key = (np.mod(mm[:, 0], n1) * n2 + np.mod(mm[:, 1], n2)) * n3 \
+            + np.mod(mm[:, 2], n3)
+        lut[key] = np.arange(len(mm))
+        self._fine_lut = lut
+        self.scf_energies = []
+
+    # ---------------------------------------------------------------- bands
+    def _g12_index(self, kp):
+        ik = id(kp)
+        if ik not in self._g12:
+            ctx = self.ctx
+            n1, n2, n3 = ctx.fft_fine.dims
+            m = kp.gkvec.miller
+            d = m[:, None, :] - m[None, :, :]
+            key = (np.mod(d[..., 0], n1) * n2 + np.mod(d[..., 1], n2)) * n3 \
+                + np.mod(d[..., 2], n3)
+            idx = self._fine_lut[key]
+            assert (idx >= 0).all(), "G1-G2 outside fine sphere"
+            self._g12[ik] = torch.from_numpy(idx)
+        return self._g12[ik]
+
+    def _hmt_full(self, ia):
+        """Full MT Hamiltonian matrix between basis functions
+        (radial_integrals_sum_L3 over the Gaunt-packed potential)."""
+        ctx = self.ctx
+        uc = ctx.unit_cell
+        lab, _ = uc.atoms[ia]
+        at = uc.atom_types[lab]
+        asc = self.classes[ia]
+        G = engine_gaunt(ctx, at)            # [L1, L3, L2]
+        # h radial integrals [lm, nrf, nrf]
+        nrf = at.num_rf
+        lmmax = ctx.lmmax_pot
+        vmt = self.potential.veff_mt[ia]     # [lmmax_pot, nr]
+        r2w = self.grids.r2w[lab]
+        # parity-allowed integrals
+        hri = np.zeros((lmmax, nrf, nrf))
+        u = asc.u
+        lbl = l_by_lm(ctx.lmax_pot)
+        vmt_w = vmt * r2w[None, :]
+        # lm > 0 channels
+        hri[1:] = np.einsum("ar,br,lr->lab", u, u, vmt_w[1:], optimize=True)
+        # parity zero-out
+        lrf = np.array([at.indexr[i][0] for i in range(nrf)])
+        par = (lbl[:, None, None] + lrf[None, :, None] + lrf[None, None, :]) % 2
+        hri[par == 1] = 0.0
+        hri[0] = asc.h_spherical
+        # hmt[xi1, xi2] = sum_lm3 G[lm1, lm3, lm2] hri[lm3, rf1, rf2]
+        nb = at.mt_basis_size
+        lm_of = np.array([b[2] for b in at.indexb])
+        rf_of = np.array([b[4] for b in at.indexb])
+        hmt = np.einsum("acb,cab->ab",
+                        G[np.ix_(lm_of, np.arange(lmmax), lm_of)],
+                        hri[:, rf_of][:, :, rf_of], optimize=True)
+        return hmt
+
+    def _omt_full(self, ia):
+        uc = self.ctx.unit_cell
+        lab, _ = uc.atoms[ia]
+        at = uc.atom_types[lab]
+        asc = self.classes[ia]
+        nb = at.mt_basis_size
+        omt = np.zeros((nb, nb))
+        for xi1 in range(nb):
+            l1, m1, lm1, o1, rf1 = at.indexb[xi1]
+            for xi2 in range(nb):
+                l2, m2, lm2, o2, rf2 = at.indexb[xi2]
+                if lm1 == lm2:
+                    omt[xi1, xi2] = asc.o_radial[(l1, o1, o2)]
+        return omt
+
+    def _basis_c(self, ik, kp, ia):
+        """C[mt_basis, N]: MT expansion of the N = ngk+nlo basis fns."""
+        ctx = self.ctx
+        uc = ctx.unit_cell
+        lab, _ = uc.atoms[ia]
+        at = uc.atom_types[lab]
+        alm = self.matching[ik].generate(ia, self.classes[ia])  # [ngk, naw]
+        ngk = kp.num_gkvec
+        nlo_tot = self._num_lo_total()
+        C = torch.zeros(at.mt_basis_size, ngk + nlo_tot,
+                        dtype=torch.complex128)
+        C[:at.mt_aw_basis_size, :ngk] = alm.T
+        # lo columns of THIS atom
+        off = ngk + self._lo_offset(ia)
+        for j in range(at.mt_lo_basis_size):
+            C[at.mt_aw_basis_size + j, off + j] = 1.0
+        return C
+
+    def _num_lo_total(self):
+        uc = self.ctx.unit_cell
+        return sum(uc.atom_types[lab].mt_lo_basis_size for lab, _ in uc.atoms)
+
+    def _lo_offset(self, ia):
+        uc = self.ctx.unit_cell
+        return sum(uc.atom_types[lab].mt_lo_basis_size
+                   for lab, _ in uc.atoms[:ia])
+
+    def set_fv_h_o(self, ik, kp):
+        ctx = self.ctx
+        uc = ctx.unit_cell
+        ngk = kp.num_gkvec
+        nlo = self._num_lo_total()
+        N = ngk + nlo
+        H = torch.zeros(N, N, dtype=torch.complex128)
+        O = torch.zeros(N, N, dtype=torch.complex128)
+
+        # interstitial
+        g12 = self._g12_index(kp)
+        veff_d = self.potential.veff_pw.cpu()
+        theta_d = ctx.theta_pw.cpu()
+        vblock = veff_d[g12]
+        tblock = theta_d[g12]
+        gk = kp.gkvec.gkvec_t.cpu()          # [ngk, 3] cartesian G+k
+        tdot = 0.5 * (gk @ gk.T).to(torch.complex128)
+        if ctx.valence_relativity == "none":
+            H[:ngk, :ngk] += vblock + tdot * tblock
+        else:
+            raise NotImplementedError(
+                f"valence relativity {ctx.valence_relativity} TODO")
+        O[:ngk, :ngk] += tblock
+
+        # MT
+        for ia in range(uc.num_atoms):
+            C = self._basis_c(ik, kp, ia)
+            hmt = torch.from_numpy(self._hmt_full(ia)).to(torch.complex128)
+            omt = torch.from_numpy(self._omt_full(ia)).to(torch.complex128)
+            Ch = C.conj().T
+            H += Ch @ (hmt @ C)
+            O += Ch @ (omt @ C)
+        return H, O
+
+    def diagonalize_fv(self, ik, kp):
+        from scipy.linalg import eigh
+        ctx = self.ctx
+        H, O = self.set_fv_h_o(ik, kp)
+        nfv = ctx.num_bands
+        herm = max(float((H - H.conj().T).abs().max()),
+                   float((O - O.conj().T).abs().max()))
+        if herm > 1e-9:
+            print(f"[sirius_amd] warning: H/O hermiticity error {herm:.2e}",
+                  file=sys.stderr)
+        w, v = eigh(H.numpy(), O.numpy(), subset_by_index=[0, nfv - 1])
+        kp.fv_eval = w
+        kp.fv_evec = v                      # [N, nfv]
+        kp.eigvals[0, :] = w
+        self.generate_fv_states(ik, kp)
+
+    def generate_fv_states(self, ik, kp):
+        ctx = self.ctx
+        uc = ctx.unit_cell
+        ngk = kp.num_gkvec
+        Z = torch.from_numpy(kp.fv_evec).to(torch.complex128)
+        kp.psi[0, :, :ngk] = Z[:ngk, :].T
+        kp.mt_coeffs = []
+        for ia in range(uc.num_atoms):
+            C = self._basis_c(ik, kp, ia)
+            kp.mt_coeffs.append((C @ Z).T.numpy())   # [nfv, mt_basis]
+
+    # ------------------------------------------------------- symmetrization
+    def symmetrize_mt_dm(self, dm):
+        """Average the MT density matrices over the space group
+        (LAPW analogue of symmetrize_density_matrix.hpp; the MT basis is
+        complex Ylm x radial, so per-l Ylm rotation matrices
+        Uy = C D^l(S) C^H act blockwise)."""
+        ctx = self.ctx
+        if ctx.symmetry is None:
+            return dm
+        from ..symmetry import rlm_rotation_matrices
+        from .sht import _conv_matrices
+        uc = ctx.unit_cell
+        ops = ctx.symmetry.ops
+        out = [np.zeros_like(d) for d in dm]
+        lmax_b = max(max(at.lmax_apw, max((lo.l for lo in at.lo_descriptors),
+                                          default=0))
+                     for at in uc.atom_types.values())
+        Cc = _conv_matrices(lmax_b)
+        Ts = {}
+        for iop, op in enumerate(ops):
+            Dl = rlm_rotation_matrices(lmax_b, op.S)
+            Ufull = np.zeros(((lmax_b + 1) ** 2, (lmax_b + 1) ** 2))
+            i0 = 0
+            for l in range(lmax_b + 1):
+                n = 2 * l + 1
+                Ufull[i0:i0 + n, i0:i0 + n] = Dl[l]
+                i0 += n
+            Uy = Cc @ Ufull @ Cc.conj().T
+            for lab in uc.type_labels:
+                at = uc.atom_types[lab]
+                Ts[(iop, lab)] = self._basis_rot(at, Uy)
+        for iop, op in enumerate(ops):
+            for ia, (lab, _) in enumerate(uc.atoms):
+                src = int(op.perm[ia])
+                T = Ts[(iop, lab)]
+                for ispn in range(dm[ia].shape[2]):
+                    out[ia][:, :, ispn] += T.conj().T @ dm[src][:, :, ispn] @ T
+        for ia in range(len(out)):
+            out[ia] /= len(ops)
+        return out
+
+    def _basis_rot(self, at, Uy):
+        nb = at.mt_basis_size
+        T = np.zeros((nb, nb), dtype=np.complex128)
+        for xi1 in range(nb):
+            l1, m1, lm1, o1, rf1 = at.indexb[xi1]
+            for xi2 in range(nb):
+                l2, m2, lm2, o2, rf2 = at.indexb[xi2]
+                if rf1 == rf2:
+                    T[xi1, xi2] = Uy[lm1, lm2]
+        return T
+
+    # -------------------------------------------------------------- SCF
+    def initial_state(self):
+        self.density.initial()
+        self.potential.generate(self.density)
+        return self
+
+    def scf_iteration(self, itsol_tol=None):
+        ctx = self.ctx
+        # 1) refresh radial basis from the current spherical potential
+        self.potential.update_atomic_potential(self.classes)
+        for asc in self.classes:
+            asc.generate_radial_functions()
+        # 2) diagonalize all k
+        for ik, kp in enumerate(self.kset.kpoints):
+            self.diagonalize_fv(ik, kp)
+        # 3) occupancies
+        self.kset.find_band_occupancies()
+        # 4) new density (+ core with current potential)
+        for asc in self.classes:
+            asc.generate_core_charge_density()
+        self.density.generate(self.kset, self)
+
+    def total_energy(self) -> dict:
+        ctx = self.ctx
+        pot = self.potential
+        kset = self.kset
+        core_sum = sum(asc.core_eval_sum for asc in self.classes)
+        val_sum = kset.valence_eval_sum()
+        ekin = core_sum + val_sum - pot.energy_veff
+        etot = ekin + pot.energy_exc + 0.5 * pot.energy_vha + pot.energy_enuc
+        return {
+            "total": etot, "ekin": ekin, "exc": pot.energy_exc,
+            "vha": pot.energy_vha, "enuc": pot.energy_enuc,
+            "veff": pot.energy_veff, "core_eval_sum": core_sum,
+            "valence_eval_sum": val_sum, "entropy_sum": kset.entropy_sum(),
+            "free": etot + kset.entropy_sum(),
+        }
+
+    def find(self, density_tol=None, energy_tol=None, num_dft_iter=None,
+             callback=None) -> dict:
+        ctx = self.ctx
+        p = ctx.cfg.parameters
+        density_tol = density_tol if density_tol is not None else p.density_tol
+        energy_tol = energy_tol if energy_tol is not None else p.energy_tol
+        num_dft_iter = num_dft_iter if num_dft_iter is not None \
+            else p.num_dft_iter
+
+        self.density.mixer_init(ctx.cfg.mixer)
+        eold = 0.0
+        etot_hist, rms_hist = [], []
+        num_iter = -1
+        t0 = time.time()
+        for it in range(num_dft_iter):
+            self.scf_iteration()
+            rms = self.density.mix()
+            self.potential.generate(self.density)
+            en = self.total_energy()
+            etot = en["total"]
+            etot_hist.append(etot)
+            rms_hist.append(rms)
+            if callback:
+                callback(it, etot, rms)
+            if ctx.cfg.control.verbosity >= 1:
+                print(f"iter {it:3d}  etot {etot: .10f}  "
+                      f"dE {etot - eold: .3e}  rms {rms:.3e}", flush=True)
+            if abs(etot - eold) < energy_tol and rms < density_tol:
+                num_iter = it
+                break
+            eold = etot
+        en = self.total_energy()
+        return {
+            "converged": num_iter >= 0,
+            "num_scf_iterations": num_iter,
+            "energy": en,
+            "etot_history": etot_hist,
+            "rms_history": rms_hist,
+            "scf_time": time.time() - t0,
+            "efermi": self.kset.energy_fermi,
+        }

@@ -86,6 +86,63 @@ def lda_c_pw(rho: torch.Tensor):
     return eps, vrho
 
 
+def _vwn_eps(x, A, b, c, x0):
+    """VWN interpolation e(x), de/dx with x = sqrt(rs)."""
+    Q = math.sqrt(4 * c - b * b)
+    X = x * x + b * x + c
+    X0 = x0 * x0 + b * x0 + c
+    t = 2 * x + b
+    atn = torch.atan(Q / t)
+    e = A * (torch.log(x * x / X) + 2 * b / Q * atn
+             - b * x0 / X0 * (torch.log((x - x0) ** 2 / X)
+                              + 2 * (b + 2 * x0) / Q * atn))
+    de = A * (2.0 / x - t / X - 4 * b / (Q * Q + t * t)
+              - b * x0 / X0 * (2.0 / (x - x0) - t / X
+                               - 4 * (b + 2 * x0) / (Q * Q + t * t)))
+    return e, de
+
+
+def lda_c_vwn(rho: torch.Tensor):
+    """XC_LDA_C_VWN (VWN5), unpolarized (libxc parameters:
+    A=0.0310907, b=3.72744, c=12.9352, x0=-0.10498)."""
+    rho = _safe_rho(rho)
+    rs = (3.0 / (4.0 * math.pi * rho)) ** _THIRD
+    x = torch.sqrt(rs)
+    e, de = _vwn_eps(x, 0.0310907, 3.72744, 12.9352, -0.10498)
+    # v = e + rho de/drho = e - (rs/3) de/drs = e - (x/6) de/dx
+    vrho = e - x / 6.0 * de
+    return e, vrho
+
+
+def lda_c_vwn_spin(ru: torch.Tensor, rd: torch.Tensor):
+    """XC_LDA_C_VWN spin-polarized (VWN5 interpolation with the spin
+    stiffness alpha_c; standard libxc scheme)."""
+    rho = _safe_rho(ru + rd)
+    ru = _safe_rho(ru)
+    rd = _safe_rho(rd)
+    z = (ru - rd) / rho
+    rs = (3.0 / (4.0 * math.pi * rho)) ** _THIRD
+    x = torch.sqrt(rs)
+    eP, deP = _vwn_eps(x, 0.0310907, 3.72744, 12.9352, -0.10498)
+    eF, deF = _vwn_eps(x, 0.01554535, 7.06042, 18.0578, -0.32500)
+    eA, deA = _vwn_eps(x, -1.0 / (6.0 * math.pi ** 2), 1.13107, 13.0045,
+                       -0.00475840)
+    fz = _fzeta(z)
+    dfz = _dfzeta(z)
+    d2f0 = 4.0 / (9.0 * (2.0 ** (1.0 / 3) - 1.0))
+    z4 = z ** 4
+    g = fz / d2f0 * (1 - z4)
+    e = eP + eA * g + (eF - eP) * fz * z4
+    de_dx = deP + deA * g + (deF - deP) * fz * z4
+    de_dz = eA / d2f0 * (dfz * (1 - z4) - fz * 4 * z ** 3) \
+        + (eF - eP) * (dfz * z4 + fz * 4 * z ** 3)
+    # v_sigma = e + rho de/drho_sigma; de/drho_s = (de/drs)(drs/drho) + (de/dz)(dz/drho_s)
+    common = e - x / 6.0 * de_dx
+    vu = common + de_dz * (1.0 - z)
+    vd = common - de_dz * (1.0 + z)
+    return e, vu, vd
+
+
 # -- PBE ---------------------------------------------------------------------
 
 _PBE_KAPPA = 0.8040
@@ -350,11 +407,12 @@ def gga_c_pbesol(rho, sigma):
     return gga_c_pbe(rho, sigma, beta=_PBESOL_BETA)
 
 
-_LDA = {"XC_LDA_X": lda_x, "XC_LDA_C_PZ": lda_c_pz, "XC_LDA_C_PW": lda_c_pw}
+_LDA = {"XC_LDA_X": lda_x, "XC_LDA_C_PZ": lda_c_pz, "XC_LDA_C_PW": lda_c_pw,
+        "XC_LDA_C_VWN": lda_c_vwn}
 _GGA = {"XC_GGA_X_PBE": gga_x_pbe, "XC_GGA_C_PBE": gga_c_pbe,
         "XC_GGA_X_PBE_SOL": gga_x_pbesol, "XC_GGA_C_PBE_SOL": gga_c_pbesol}
 _LDA_SPIN = {"XC_LDA_X": lda_x_spin, "XC_LDA_C_PZ": lda_c_pz_spin,
-             "XC_LDA_C_PW": lda_c_pw_spin}
+             "XC_LDA_C_PW": lda_c_pw_spin, "XC_LDA_C_VWN": lda_c_vwn_spin}
 
 
 def evaluate_spin(names: list[str], ru: torch.Tensor, rd: torch.Tensor,
