@@ -208,10 +208,14 @@ class SimulationContext:
         self.device = torch.device(device)
 
         p = cfg.parameters
+        self.full_potential = (p.electronic_structure_method
+                               == "full_potential_lapwlo")
         self.pw_cutoff = float(p.pw_cutoff)
         self.gk_cutoff = float(p.gk_cutoff)
-        if self.pw_cutoff < 2 * self.gk_cutoff:
+        if self.pw_cutoff < 2 * self.gk_cutoff and not self.full_potential:
             # reference insists fine grid covers products of wavefunctions
+            # (PP case; LAPW honors the deck's pw_cutoff — the step
+            # function and V*theta are deliberately truncated there)
             self.pw_cutoff = 2 * self.gk_cutoff
         self.num_mag_dims = int(p.num_mag_dims)
         self.num_spins = 2 if self.num_mag_dims > 0 else 1
@@ -232,8 +236,25 @@ class SimulationContext:
         self.gvec_coarse = Gvec(uc.recip, 2 * self.gk_cutoff, dims=coarse_dims, device=device)
         self.fft_fine = SphericalFFT(self.gvec_fine)
         self.fft_coarse = SphericalFFT(self.gvec_coarse)
-        # coarse -> fine index map (gvec_base_mapping analogue)
-        self.coarse_to_fine = self.gvec_coarse.gvec_map_to(self.gvec_fine)
+        # coarse -> fine index map (gvec_base_mapping analogue); for LAPW
+        # with pw_cutoff < 2*gk the spheres only intersect — keep pair maps
+        if self.full_potential and self.pw_cutoff < 2 * self.gk_cutoff:
+            key = {tuple(mm): i for i, mm in enumerate(self.gvec_fine.miller)}
+            ic, if_ = [], []
+            for j, mm in enumerate(self.gvec_coarse.miller):
+                i = key.get(tuple(mm))
+                if i is not None:
+                    ic.append(j)
+                    if_.append(i)
+            self.coarse_fine_pairs = (
+                torch.tensor(ic, dtype=torch.long, device=self.device),
+                torch.tensor(if_, dtype=torch.long, device=self.device))
+            self.coarse_to_fine = None
+        else:
+            self.coarse_to_fine = self.gvec_coarse.gvec_map_to(self.gvec_fine)
+            n = self.gvec_coarse.num_gvec
+            self.coarse_fine_pairs = (
+                torch.arange(n, device=self.device), self.coarse_to_fine)
 
         # number of bands (simulation_context.cpp:333-352)
         nel = uc.num_electrons

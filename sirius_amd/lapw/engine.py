@@ -81,6 +81,65 @@ def make_lapw_context(cfg: Config, base_dir: str = ".", device=None):
             pos.append((lab, np.asarray(v[:3], dtype=np.float64)))
     uc = UnitCell(lat, types, pos)
 
+    # auto MT radii (reference Unit_cell::find_mt_radii, unit_cell.cpp:30:
+    # auto_rmt=1 -> R = min(rmt_max, 0.95*d_nn/2) per type, then inflate)
+    auto_rmt = int(getattr(p, "auto_rmt", 1))
+    if auto_rmt:
+        rmt_max = float(cfg.control.rmt_max)
+        labels = list(types)
+        tid = {lab: i for i, lab in enumerate(labels)}
+        nn_cut = 20.0
+        pairs = uc.nearest_neighbours_full(nn_cut)
+        # nearest neighbour distance per atom
+        nn = [None] * uc.num_atoms
+        for ia, ja, d, _, _ in pairs:
+            if nn[ia] is None or d < nn[ia][1]:
+                nn[ia] = (ja, d)
+        Rmt = {lab: 1e10 for lab in labels}
+        for ia, (lab1, _) in enumerate(uc.atoms):
+            if nn[ia] is not None:
+                ja, d = nn[ia]
+                lab2 = uc.atoms[ja][0]
+                if auto_rmt == 1:
+                    R = min(rmt_max, 0.95 * d / 2)
+                    Rmt[lab1] = min(R, Rmt[lab1])
+                    Rmt[lab2] = min(R, Rmt[lab2])
+                else:  # auto_rmt == 2
+                    s = 0.95 * d / (types[lab1].rmt + types[lab2].rmt)
+                    Rmt[lab1] = min(s, Rmt[lab1])
+                    Rmt[lab2] = min(s, Rmt[lab2])
+            else:
+                Rmt[lab1] = rmt_max if auto_rmt == 1 \
+                    else rmt_max / types[lab1].rmt
+        if auto_rmt == 2:
+            for lab in labels:
+                Rmt[lab] = min(rmt_max, types[lab].rmt * Rmt[lab])
+        # inflate (unit_cell.cpp:81-116)
+        scale_ok = {lab: True for lab in labels}
+        for ia, (lab1, _) in enumerate(uc.atoms):
+            if nn[ia] is not None:
+                ja, d = nn[ia]
+                lab2 = uc.atoms[ja][0]
+                if Rmt[lab1] + Rmt[lab2] > d * 0.94:
+                    scale_ok[lab1] = False
+                    scale_ok[lab2] = False
+        Rmt_infl = {lab: 1e10 for lab in labels}
+        for ia, (lab1, _) in enumerate(uc.atoms):
+            if nn[ia] is not None:
+                ja, d = nn[ia]
+                lab2 = uc.atoms[ja][0]
+                if scale_ok[lab1] and not scale_ok[lab2]:
+                    Rmt_infl[lab1] = min(Rmt_infl[lab1],
+                                         min(rmt_max, 0.95 * (d - Rmt[lab2])))
+                else:
+                    Rmt_infl[lab1] = Rmt[lab1]
+            else:
+                Rmt_infl[lab1] = Rmt[lab1]
+        for lab in labels:
+            if Rmt_infl[lab] < 0.3:
+                raise RuntimeError(f"auto MT radius too small for {lab}")
+            types[lab].set_rmt(Rmt_infl[lab])
+
     # gk cutoff from rgkmax: aw_cutoff / min R_mt (simulation_context.cpp:292)
     if float(p.aw_cutoff) > 0:
         min_rmt = min(at.rmt for at in types.values())
@@ -307,11 +366,13 @@ class FPDensity:
                 dm[ia] = t.numpy()
         self.dm = dm
 
-        # coarse real grid -> fine PW
+        # coarse real grid -> fine PW (truncated to the fine sphere when
+        # pw_cutoff < 2*gk, as the reference does for LAPW)
         rho_pw_c = ctx.fft_coarse.to_pw(rho_c.to(ctx.dtype))
         rho_pw = torch.zeros(ctx.gvec_fine.num_gvec, dtype=ctx.dtype,
                              device=ctx.device)
-        rho_pw[ctx.coarse_to_fine] = rho_pw_c
+        ic, if_ = ctx.coarse_fine_pairs
+        rho_pw[if_] = rho_pw_c[ic]
         if symmetrize and ctx.symmetry is not None:
             from ..symmetry import symmetrize_rho_g
             rho_pw = symmetrize_rho_g(rho_pw, ctx.gvec_fine, ctx.symmetry.ops)
@@ -511,20 +572,21 @@ class FPPotential:
         qmt = np.zeros(ctx.lmmax_rho)
         v = np.zeros((ctx.lmmax_pot, at.nmtp))
         R = at.rmt
-        for lm in range(ctx.lmmax_rho):
-            l = int(lbl[lm])
-            rl = rho_mt[lm]
-            f1 = rl * r ** (l + 2)
-            g1 = CubicSpline(r, f1).antiderivative()(r)
-            qmt[lm] = g1[-1]
-            if lm < ctx.lmmax_pot:
-                f2 = rl * r ** (1 - l)
-                g2 = CubicSpline(r, f2).antiderivative()(r)
-                d1 = 1.0 / R ** (2 * l + 1)
-                vlm = (1.0 - (r / R) ** (2 * l + 1)) * g1 / r ** (l + 1) \
-                    + (g2[-1] - g2) * r ** l \
-                    - (g1[-1] - g1) * r ** l * d1
-                v[lm] = vlm * FOURPI / (2 * l + 1)
+        lv = lbl[:ctx.lmmax_rho].astype(np.int64)
+        f1 = rho_mt[:ctx.lmmax_rho] * r[None, :] ** (lv[:, None] + 2)
+        g1 = CubicSpline(r, f1, axis=1).antiderivative()(r)   # [lm, nr]
+        qmt[:] = g1[:, -1]
+        npot = ctx.lmmax_pot
+        lp = lv[:npot]
+        f2 = rho_mt[:npot] * r[None, :] ** (1 - lp[:, None])
+        g2 = CubicSpline(r, f2, axis=1).antiderivative()(r)
+        d1 = 1.0 / R ** (2 * lp + 1)
+        rl = r[None, :] ** lp[:, None]
+        v[:] = ((1.0 - (r[None, :] / R) ** (2 * lp[:, None] + 1))
+                * g1[:npot] / r[None, :] ** (lp[:, None] + 1)
+                + (g2[:, -1:] - g2) * rl
+                - (g1[:npot, -1:] - g1[:npot]) * rl * d1[:, None]) \
+            * (FOURPI / (2 * lp[:, None] + 1))
         # nuclear: +zn/R (|_VHA_AUX off branch); -zn/r is applied later
         v[0] += at.zn / R / Y00
         qmt[0] -= at.zn * Y00
@@ -674,9 +736,14 @@ class FPGroundState:
         self.matching = {}
         for ik, kp in enumerate(kset.kpoints):
             self.matching[ik] = MatchingCoefficients(ctx, kp.gkvec)
-        # G1-G2 lookup on the fine sphere
+        # G1-G2 lookup on the fine sphere.  The key grid must cover the
+        # full difference range |G1-G2| <= 2*gk without wrap-around, so
+        # use the larger of the fine/coarse FFT dims per axis.
         self._g12 = {}
-        n1, n2, n3 = ctx.fft_fine.dims
+        dims = tuple(max(a, b) for a, b in zip(ctx.fft_fine.dims,
+                                               ctx.fft_coarse.dims))
+        self._lut_dims = dims
+        n1, n2, n3 = dims
         lut = np.full(n1 * n2 * n3, -1, dtype=np.int64)
         mm = ctx.gvec_fine.miller
         key = (np.mod(mm[:, 0], n1) * n2 + np.mod(mm[:, 1], n2)) * n3 \
@@ -690,14 +757,18 @@ class FPGroundState:
         ik = id(kp)
         if ik not in self._g12:
             ctx = self.ctx
-            n1, n2, n3 = ctx.fft_fine.dims
+            n1, n2, n3 = self._lut_dims
             m = kp.gkvec.miller
             d = m[:, None, :] - m[None, :, :]
             key = (np.mod(d[..., 0], n1) * n2 + np.mod(d[..., 1], n2)) * n3 \
                 + np.mod(d[..., 2], n3)
             idx = self._fine_lut[key]
-            assert (idx >= 0).all(), "G1-G2 outside fine sphere"
-            self._g12[ik] = torch.from_numpy(idx)
+            # G1-G2 outside the fine sphere: theta/V truncated to zero
+            # there (reference LAPW behavior when pw_cutoff < 2*gk)
+            mask = idx >= 0
+            idx = np.where(mask, idx, 0)
+            self._g12[ik] = (torch.from_numpy(idx),
+                             torch.from_numpy(mask.astype(np.float64)))
         return self._g12[ik]
 
     def _hmt_full(self, ia):
@@ -777,6 +848,22 @@ class FPGroundState:
         return sum(uc.atom_types[lab].mt_lo_basis_size
                    for lab, _ in uc.atoms[:ia])
 
+    def _hmt_cached(self, ia):
+        c = getattr(self, "_hmt_cache", None)
+        if c is None:
+            self._hmt_cache = c = {}
+        if ia not in c:
+            c[ia] = torch.from_numpy(self._hmt_full(ia)).to(torch.complex128)
+        return c[ia]
+
+    def _omt_cached(self, ia):
+        c = getattr(self, "_omt_cache", None)
+        if c is None:
+            self._omt_cache = c = {}
+        if ia not in c:
+            c[ia] = torch.from_numpy(self._omt_full(ia)).to(torch.complex128)
+        return c[ia]
+
     def set_fv_h_o(self, ik, kp):
         ctx = self.ctx
         uc = ctx.unit_cell
@@ -787,44 +874,68 @@ class FPGroundState:
         O = torch.zeros(N, N, dtype=torch.complex128)
 
         # interstitial
-        g12 = self._g12_index(kp)
+        g12, g12_mask = self._g12_index(kp)
         veff_d = self.potential.veff_pw.cpu()
         theta_d = ctx.theta_pw.cpu()
-        vblock = veff_d[g12]
-        tblock = theta_d[g12]
+        vblock = veff_d[g12] * g12_mask
+        tblock = theta_d[g12] * g12_mask
         gk = kp.gkvec.gkvec_t.cpu()          # [ngk, 3] cartesian G+k
         tdot = 0.5 * (gk @ gk.T).to(torch.complex128)
-        if ctx.valence_relativity == "none":
+        if ctx.valence_relativity in ("none", "koelling_harmon"):
+            # KH relativity lives entirely in the radial functions; the
+            # interstitial kinetic term is the default branch
+            # (hamiltonian_k.cpp:683-687)
             H[:ngk, :ngk] += vblock + tdot * tblock
+        elif ctx.valence_relativity in ("zora", "iora"):
+            rm_inv = self.potential.rm_inv_pw.cpu()[g12] * g12_mask
+            H[:ngk, :ngk] += vblock + tdot * rm_inv
+            if ctx.valence_relativity == "iora":
+                sq_alpha_half = 0.5 / 137.035999139 ** 2
+                rm2_inv = self.potential.rm2_inv_pw.cpu()[g12] * g12_mask
+                O[:ngk, :ngk] += tdot * sq_alpha_half * rm2_inv
         else:
             raise NotImplementedError(
                 f"valence relativity {ctx.valence_relativity} TODO")
         O[:ngk, :ngk] += tblock
 
         # MT
+        self._C_cache = getattr(self, "_C_cache", {})
         for ia in range(uc.num_atoms):
             C = self._basis_c(ik, kp, ia)
-            hmt = torch.from_numpy(self._hmt_full(ia)).to(torch.complex128)
-            omt = torch.from_numpy(self._omt_full(ia)).to(torch.complex128)
+            self._C_cache[(ik, ia)] = C
+            hmt = self._hmt_cached(ia)
+            omt = self._omt_cached(ia)
             Ch = C.conj().T
             H += Ch @ (hmt @ C)
             O += Ch @ (omt @ C)
         return H, O
 
-    def diagonalize_fv(self, ik, kp):
+    def diagonalize_fv(self, ik, kp, dense_threshold: int = 2500):
+        """Dense generalized solve for small bases (rocSOLVER/LAPACK wins
+        there), block Davidson with lo extra basis beyond
+        (reference default: diagonalize_fp_fv_davidson)."""
         from scipy.linalg import eigh
         ctx = self.ctx
-        H, O = self.set_fv_h_o(ik, kp)
         nfv = ctx.num_bands
-        herm = max(float((H - H.conj().T).abs().max()),
-                   float((O - O.conj().T).abs().max()))
-        if herm > 1e-9:
-            print(f"[sirius_amd] warning: H/O hermiticity error {herm:.2e}",
-                  file=sys.stderr)
-        w, v = eigh(H.numpy(), O.numpy(), subset_by_index=[0, nfv - 1])
-        kp.fv_eval = w
-        kp.fv_evec = v                      # [N, nfv]
-        kp.eigvals[0, :] = w
+        N = kp.num_gkvec + self._num_lo_total()
+        if N <= dense_threshold:
+            H, O = self.set_fv_h_o(ik, kp)
+            herm = max(float((H - H.conj().T).abs().max()),
+                       float((O - O.conj().T).abs().max()))
+            if herm > 1e-9:
+                print(f"[sirius_amd] warning: H/O hermiticity error {herm:.2e}",
+                      file=sys.stderr)
+            w, v = eigh(H.numpy(), O.numpy(), subset_by_index=[0, nfv - 1])
+            kp.fv_eval = w
+            kp.fv_evec = v                      # [N, nfv]
+            kp.eigvals[0, :] = w
+        else:
+            from .davidson_fp import davidson_fv
+            for ia in range(ctx.unit_cell.num_atoms):
+                self._C_cache = getattr(self, "_C_cache", {})
+                if (ik, ia) not in self._C_cache:
+                    self._C_cache[(ik, ia)] = self._basis_c(ik, kp, ia)
+            davidson_fv(self, ik, kp, nfv, tol=self._itsol_tol)
         self.generate_fv_states(ik, kp)
 
     def generate_fv_states(self, ik, kp):
@@ -835,7 +946,9 @@ class FPGroundState:
         kp.psi[0, :, :ngk] = Z[:ngk, :].T
         kp.mt_coeffs = []
         for ia in range(uc.num_atoms):
-            C = self._basis_c(ik, kp, ia)
+            C = self._C_cache.get((ik, ia))
+            if C is None:
+                C = self._basis_c(ik, kp, ia)
             kp.mt_coeffs.append((C @ Z).T.numpy())   # [nfv, mt_basis]
 
     # ------------------------------------------------------- symmetrization
@@ -855,20 +968,23 @@ class FPGroundState:
         lmax_b = max(max(at.lmax_apw, max((lo.l for lo in at.lo_descriptors),
                                           default=0))
                      for at in uc.atom_types.values())
-        Cc = _conv_matrices(lmax_b)
-        Ts = {}
-        for iop, op in enumerate(ops):
-            Dl = rlm_rotation_matrices(lmax_b, op.S)
-            Ufull = np.zeros(((lmax_b + 1) ** 2, (lmax_b + 1) ** 2))
-            i0 = 0
-            for l in range(lmax_b + 1):
-                n = 2 * l + 1
-                Ufull[i0:i0 + n, i0:i0 + n] = Dl[l]
-                i0 += n
-            Uy = Cc @ Ufull @ Cc.conj().T
-            for lab in uc.type_labels:
-                at = uc.atom_types[lab]
-                Ts[(iop, lab)] = self._basis_rot(at, Uy)
+        Ts = getattr(self, "_sym_rot_cache", None)
+        if Ts is None:
+            Cc = _conv_matrices(lmax_b)
+            Ts = {}
+            for iop, op in enumerate(ops):
+                Dl = rlm_rotation_matrices(lmax_b, op.S)
+                Ufull = np.zeros(((lmax_b + 1) ** 2, (lmax_b + 1) ** 2))
+                i0 = 0
+                for l in range(lmax_b + 1):
+                    n = 2 * l + 1
+                    Ufull[i0:i0 + n, i0:i0 + n] = Dl[l]
+                    i0 += n
+                Uy = Cc @ Ufull @ Cc.conj().T
+                for lab in uc.type_labels:
+                    at = uc.atom_types[lab]
+                    Ts[(iop, lab)] = self._basis_rot(at, Uy)
+            self._sym_rot_cache = Ts
         for iop, op in enumerate(ops):
             for ia, (lab, _) in enumerate(uc.atoms):
                 src = int(op.perm[ia])
@@ -898,10 +1014,16 @@ class FPGroundState:
 
     def scf_iteration(self, itsol_tol=None):
         ctx = self.ctx
+        self._itsol_tol = itsol_tol if itsol_tol is not None else 1e-10
         # 1) refresh radial basis from the current spherical potential
         self.potential.update_atomic_potential(self.classes)
         for asc in self.classes:
             asc.generate_radial_functions()
+        self._hmt_cache = {}
+        self._omt_cache = {}
+        self._C_cache = {}
+        for attr in ("_vtheta_coarse", "_theta_coarse"):
+            self.__dict__.pop(attr, None)
         # 2) diagonalize all k
         for ik, kp in enumerate(self.kset.kpoints):
             self.diagonalize_fv(ik, kp)

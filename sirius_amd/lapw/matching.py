@@ -86,15 +86,28 @@ class MatchingCoefficients:
                     A[dm, order] = asc.sd[dm, idxrf]
             Ainv.append(np.linalg.inv(A))
 
-        alm = np.empty((ng, at.mt_aw_basis_size), dtype=np.complex128)
-        for xi in range(at.mt_aw_basis_size):
-            l, m, lm, nu, idxrf = at.indexb[xi]
+        # vectorized: zt[(l,nu)] = sum_dm alm_b[dm,:,l] Ainv_l[nu,dm]
+        nxa = at.mt_aw_basis_size
+        key = "_xi_maps"
+        maps = getattr(at, key, None)
+        if maps is None:
+            lm_of = np.array([b[2] for b in at.indexb[:nxa]])
+            l_of = np.array([b[0] for b in at.indexb[:nxa]])
+            nu_of = np.array([b[3] for b in at.indexb[:nxa]])
+            maps = (lm_of, l_of, nu_of)
+            setattr(at, key, maps)
+        lm_of, l_of, nu_of = maps
+        max_nu = max(at.aw_order(l) for l in range(at.lmax_apw + 1))
+        zt_ln = np.zeros((ng, at.lmax_apw + 1, max_nu), dtype=np.complex128)
+        for l in range(at.lmax_apw + 1):
             naw = at.aw_order(l)
-            zt = np.zeros(ng, dtype=np.complex128)
-            for dm in range(naw):
-                zt += alm_b[dm, :, l] * Ainv[l][nu, dm]
-            if conjugate:
-                alm[:, xi] = np.conj(phase * zt) * self.gkvec_ylm[:, lm]
-            else:
-                alm[:, xi] = phase * zt * np.conj(self.gkvec_ylm[:, lm])
+            # [ng, dm] @ [dm, nu] -> [ng, nu]
+            zt_ln[:, l, :naw] = np.ascontiguousarray(
+                alm_b[:naw, :, l].T) @ Ainv[l].T
+        zt = zt_ln[:, l_of, nu_of]                    # [ng, nxa]
+        yl = self.gkvec_ylm[:, lm_of]
+        if conjugate:
+            alm = np.conj(phase[:, None] * zt) * yl
+        else:
+            alm = phase[:, None] * zt * np.conj(yl)
         return torch.from_numpy(alm)

@@ -54,6 +54,7 @@ class FPAtomType:
         # MT radial grid (reference Radial_grid_exp: x_i = r0 (R/r0)^{t^p})
         kind, p = [s.strip() for s in radial_grid.split(",")]
         p = float(p)
+        self._grid_p = p
         t = (np.arange(self.nmtp) / (self.nmtp - 1)) ** p
         if kind != "exponential":
             raise ValueError(f"unsupported radial grid: {kind}")
@@ -155,6 +156,15 @@ class FPAtomType:
                 self.indexb.append((l, m, lm, o, idxrf))
         self.mt_basis_size = len(self.indexb)
         self.mt_lo_basis_size = self.mt_basis_size - self.mt_aw_basis_size
+
+    def set_rmt(self, rmt: float):
+        """Rebuild the MT radial grid with a new sphere radius (used by
+        auto_rmt; reference unit_cell.cpp:850-855 keeps nmtp and rmin)."""
+        self.rmt = float(rmt)
+        t = (np.arange(self.nmtp) / (self.nmtp - 1)) ** self._grid_p
+        self.r = self.rmin * (self.rmt / self.rmin) ** t
+        self.r[0] = self.rmin
+        self.r[-1] = self.rmt
 
     def rf_index(self, l: int, order: int) -> int:
         return self._rf_by_lo[(l, order)]
