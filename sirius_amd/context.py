@@ -257,7 +257,11 @@ class SimulationContext:
                 torch.arange(n, device=self.device), self.coarse_to_fine)
 
         # number of bands (simulation_context.cpp:333-352)
-        nel = uc.num_electrons
+        # FP-LAPW: valence electrons only (cores are solved separately);
+        # PP: all pseudo electrons are valence (simulation_context.cpp:333)
+        nel = sum(getattr(uc.atom_types[lab], "num_valence_electrons", None)
+                  or uc.atom_types[lab].zn for lab, _ in uc.atoms)
+        self.num_valence_electrons = float(nel)
         nbnd = int(nel / 2.0) + max(10, int(0.1 * nel))
         if self.num_mag_dims == 3:
             nbnd *= 2

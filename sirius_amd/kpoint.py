@@ -148,7 +148,7 @@ class KPointSet:
         ctx = self.ctx
         p = ctx.cfg.parameters
         self.sync_band()
-        ne = ctx.unit_cell.num_electrons
+        ne = getattr(ctx, 'num_valence_electrons', None) or ctx.unit_cell.num_electrons
         eig = self._all_eig  # [nk, num_spin_steps, nb]
         nk, nspin, nb = eig.shape
         eigf = eig.reshape(nk * nspin, nb)

@@ -43,15 +43,28 @@ class FVOperator:
         # coarse-grid V*theta and theta (real space)
         if not hasattr(engine, "_vtheta_coarse"):
             ic, if_ = ctx.coarse_fine_pairs
-            vpw_c = torch.zeros(ctx.gvec_coarse.num_gvec, dtype=ctx.dtype,
+
+            def _to_coarse(f_pw_fine):
+                c = torch.zeros(ctx.gvec_coarse.num_gvec, dtype=ctx.dtype,
                                 device=ctx.device)
-            tpw_c = torch.zeros_like(vpw_c)
-            vpw_c[ic] = engine.potential.veff_pw[if_]
-            tpw_c[ic] = ctx.theta_pw[if_]
-            engine._vtheta_coarse = ctx.fft_coarse.to_real(vpw_c).real
-            engine._theta_coarse = ctx.fft_coarse.to_real(tpw_c).real
+                c[ic] = f_pw_fine[if_]
+                return ctx.fft_coarse.to_real(c).real
+
+            engine._vtheta_coarse = _to_coarse(engine.potential.veff_pw)
+            engine._theta_coarse = _to_coarse(ctx.theta_pw)
+            rel = ctx.valence_relativity
+            if rel in ("zora", "iora"):
+                engine._kin_coarse = _to_coarse(engine.potential.rm_inv_pw)
+            else:
+                engine._kin_coarse = engine._theta_coarse
+            if rel == "iora":
+                engine._okin_coarse = _to_coarse(engine.potential.rm2_inv_pw)
+            else:
+                engine._okin_coarse = None
         self.vtheta_rg = engine._vtheta_coarse
         self.theta_rg = engine._theta_coarse
+        self.kin_rg = engine._kin_coarse
+        self.okin_rg = engine._okin_coarse
         self.gk = kp.gkvec.gkvec_t.to(torch.float64)   # [ngk, 3]
         # per-atom C and hmt/omt
         self.C = []
@@ -70,14 +83,19 @@ class FVOperator:
         hx = torch.zeros_like(x)
         ox = torch.zeros_like(x)
 
-        # interstitial (V*theta) psi + theta-weighted kinetic + theta overlap
+        # interstitial (V*theta) psi + kinetic (theta- or ZORA-mass-
+        # weighted) + theta overlap (+ IORA overlap correction)
         psir = kp.fft.to_real(c)
         hx[:, :ngk] += kp.fft.to_pw(self.vtheta_rg * psir)
         ox[:, :ngk] += kp.fft.to_pw(self.theta_rg * psir)
+        sq_alpha_half = 0.5 / 137.035999139 ** 2
         for ax in range(3):
             gax = self.gk[:, ax]
             gpsir = kp.fft.to_real(c * gax)
-            hx[:, :ngk] += 0.5 * gax * kp.fft.to_pw(self.theta_rg * gpsir)
+            hx[:, :ngk] += 0.5 * gax * kp.fft.to_pw(self.kin_rg * gpsir)
+            if self.okin_rg is not None:
+                ox[:, :ngk] += 0.5 * sq_alpha_half * gax * kp.fft.to_pw(
+                    self.okin_rg * gpsir)
 
         # MT: S = C x ; hx += C^H hmt S ; ox += C^H omt S
         for ia in range(len(self.C)):
@@ -97,8 +115,19 @@ class FVOperator:
         v0 = float(self.engine.potential.veff_pw[
             self.engine.ctx.gvec_fine.index_of_zero()].real)
         g2 = (self.gk ** 2).sum(-1)
-        hd[:ngk] = 0.5 * g2 * theta0 + v0
+        rel = self.engine.ctx.valence_relativity
+        if rel in ("zora", "iora"):
+            kin0 = float(self.engine.potential.rm_inv_pw[
+                self.engine.ctx.gvec_fine.index_of_zero()].real)
+        else:
+            kin0 = theta0
+        hd[:ngk] = 0.5 * g2 * kin0 + v0
         od[:ngk] = theta0
+        if rel == "iora":
+            sq_alpha_half = 0.5 / 137.035999139 ** 2
+            rm2_0 = float(self.engine.potential.rm2_inv_pw[
+                self.engine.ctx.gvec_fine.index_of_zero()].real)
+            od[:ngk] += 0.5 * sq_alpha_half * g2 * rm2_0
         for ia in range(len(self.C)):
             C = self.C[ia]
             hd += torch.einsum("an,ab,bn->n", C.conj(), self.hmt[ia], C).real

@@ -75,11 +75,28 @@ def make_lapw_context(cfg: Config, base_dir: str = ".", device=None):
         types[lab].spin_orbit = False
         types[lab].num_beta = 0
     pos = []
+    vfields = []
+    units = getattr(ucfg, "atom_coordinate_units", "lattice")
+    inv_lat = np.linalg.inv(lat)
+    from ..constants import bohr_to_ang
     atoms_cfg = ucfg.atoms if isinstance(ucfg.atoms, dict) else dict(ucfg.atoms)
     for lab in ucfg.atom_types:
         for v in atoms_cfg.get(lab, []):
-            pos.append((lab, np.asarray(v[:3], dtype=np.float64)))
+            v = np.asarray(v, dtype=np.float64)
+            x = v[:3]
+            if units in ("au", "a.u."):
+                x = x @ inv_lat
+            elif units in ("A", "angstrom"):
+                x = (x / bohr_to_ang) @ inv_lat
+            pos.append((lab, x))
+            vf = np.zeros(3)
+            if len(v) >= 6:
+                vf = v[3:6]
+            elif len(v) == 4:
+                vf[2] = v[3]
+            vfields.append(vf)
     uc = UnitCell(lat, types, pos)
+    uc.vector_fields = np.array(vfields)
 
     # auto MT radii (reference Unit_cell::find_mt_radii, unit_cell.cpp:30:
     # auto_rmt=1 -> R = min(rmt_max, 0.95*d_nn/2) per type, then inflate)
@@ -684,6 +701,15 @@ class FPPotential:
             self.veff_mt[ia] = self.vha_mt[ia] + self.vxc_mt[ia]
         vtheta = (self.veff_rg * ctx.theta_rg).to(ctx.dtype)
         self.veff_pw = ctx.fft_fine.to_pw(vtheta)
+        # relativistic inverse-mass tables (generate_pw_coeffs.cpp:32-56)
+        if ctx.valence_relativity in ("zora", "iora"):
+            sq_alpha_half = 0.5 / 137.035999139 ** 2
+            M = 1.0 - sq_alpha_half * self.veff_rg
+            self.rm_inv_pw = ctx.fft_fine.to_pw(
+                (ctx.theta_rg / M).to(ctx.dtype))
+            if ctx.valence_relativity == "iora":
+                self.rm2_inv_pw = ctx.fft_fine.to_pw(
+                    (ctx.theta_rg / M ** 2).to(ctx.dtype))
 
         self.energy_veff = it_inner(ctx, density.rho_rg, self.veff_rg) \
             + mt_inner(ctx, grids, density.rho_mt, self.veff_mt)
