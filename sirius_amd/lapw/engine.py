@@ -839,12 +839,17 @@ class FPGroundState:
         asc = self.classes[ia]
         nb = at.mt_basis_size
         omt = np.zeros((nb, nb))
+        iora = self.ctx.valence_relativity == "iora"
         for xi1 in range(nb):
             l1, m1, lm1, o1, rf1 = at.indexb[xi1]
             for xi2 in range(nb):
                 l2, m2, lm2, o2, rf2 = at.indexb[xi2]
                 if lm1 == lm2:
                     omt[xi1, xi2] = asc.o_radial[(l1, o1, o2)]
+                    if iora:
+                        # IORA overlap correction (hamiltonian_k.cpp:546,
+                        # :592, :636 + add_o1mt_to_apw)
+                        omt[xi1, xi2] += asc.o1_radial[rf1, rf2]
         return omt
 
     def _basis_c(self, ik, kp, ia):
