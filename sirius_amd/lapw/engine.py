@@ -275,6 +275,12 @@ class FPDensity:
                                   device=ctx.device)
         self.rho_mt = [np.zeros((ctx.lmmax_rho, uc.atom_types[lab].nmtp))
                        for lab, _ in uc.atoms]
+        # collinear magnetization (num_mag_dims == 1)
+        self.nmag = 1 if ctx.num_mag_dims == 1 else 0
+        if self.nmag:
+            self.mag_pw = torch.zeros_like(self.rho_pw)
+            self.mag_rg = torch.zeros_like(self.rho_rg)
+            self.mag_mt = [np.zeros_like(m) for m in self.rho_mt]
         self.mixer = None
 
     # -- charge bookkeeping -------------------------------------------------
@@ -345,19 +351,38 @@ class FPDensity:
             self.rho_mt[ia] += glm[:, None] * rr2[None, :]
         self.normalize()
 
+        # initial MT magnetization from the deck moments
+        # (density.cpp:413-452: rho * smooth falloff, scaled to the moment)
+        if self.nmag:
+            for ia, (lab, _) in enumerate(uc.atoms):
+                at = uc.atom_types[lab]
+                v = uc.vector_fields[ia]
+                length = abs(v[2])
+                x = at.r / at.rmt
+                rho_s = self.rho_mt[ia][0] * Y00 * (1 - 3 * x ** 2 + 2 * x ** 3)
+                q = FOURPI * float(self.grids.r2w[lab] @ rho_s)
+                vz = v[2]
+                if q < length and length > 0:
+                    vz *= q / length
+                    length = q
+                if length > 1e-8:
+                    self.mag_mt[ia][0] = rho_s * vz / q / Y00
+
     # -- valence + core -----------------------------------------------------
     def generate(self, kset, engine, symmetrize: bool = True):
         ctx = self.ctx
         uc = ctx.unit_cell
-        # interstitial: accumulate |psi(r)|^2 on the coarse grid
+        # interstitial: accumulate |psi(r)|^2 on the coarse grid per spin
         coarse = ctx.fft_coarse
-        rho_c = torch.zeros(coarse.dims, dtype=ctx.rdtype, device=ctx.device)
+        nsp = ctx.num_spins
+        rho_cs = [torch.zeros(coarse.dims, dtype=ctx.rdtype, device=ctx.device)
+                  for _ in range(nsp)]
         dm = [np.zeros((uc.atom_types[lab].mt_basis_size,
-                        uc.atom_types[lab].mt_basis_size, ctx.num_spins),
+                        uc.atom_types[lab].mt_basis_size, nsp),
                        dtype=np.complex128) for lab, _ in uc.atoms]
         for kp in kset.kpoints:
             # map G+k sphere coefficients into the coarse FFT grid
-            for ispn in range(ctx.num_spins):
+            for ispn in range(nsp):
                 occ = kp.occ[ispn]
                 nocc = int(np.sum(occ > 1e-12))
                 if nocc == 0:
@@ -365,18 +390,20 @@ class FPDensity:
                 w = torch.from_numpy(occ[:nocc] * kp.weight).to(ctx.device)
                 psi = kp.psi[ispn, :nocc, :kp.num_gkvec]
                 psir = kp.fft.to_real(psi)
-                rho_c += torch.einsum(
+                rho_cs[ispn] += torch.einsum(
                     "b,bxyz->xyz", w.to(ctx.rdtype),
                     (psir.real ** 2 + psir.imag ** 2))
                 # MT density matrix
-                mt = kp.mt_coeffs  # list per atom: [nfv, mt_basis]
+                mt = kp.mt_coeffs[ispn]  # list per atom: [nfv, mt_basis]
                 for ia, (lab, _) in enumerate(uc.atoms):
                     c = mt[ia][:nocc]
                     wocc = (occ[:nocc] * kp.weight)[:, None]
                     dm[ia][:, :, ispn] += (c.conj() * wocc).T @ c
-        rho_c *= 1.0 / uc.omega
+        for ispn in range(nsp):
+            rho_cs[ispn] *= 1.0 / uc.omega
         if kset.comm.active:
-            kset.comm.allreduce_(rho_c)
+            for ispn in range(nsp):
+                kset.comm.allreduce_(rho_cs[ispn])
             for ia in range(uc.num_atoms):
                 t = torch.from_numpy(dm[ia])
                 kset.comm.allreduce_(t)
@@ -385,16 +412,28 @@ class FPDensity:
 
         # coarse real grid -> fine PW (truncated to the fine sphere when
         # pw_cutoff < 2*gk, as the reference does for LAPW)
-        rho_pw_c = ctx.fft_coarse.to_pw(rho_c.to(ctx.dtype))
-        rho_pw = torch.zeros(ctx.gvec_fine.num_gvec, dtype=ctx.dtype,
-                             device=ctx.device)
         ic, if_ = ctx.coarse_fine_pairs
-        rho_pw[if_] = rho_pw_c[ic]
+
+        def to_fine(rc):
+            pw_c = ctx.fft_coarse.to_pw(rc.to(ctx.dtype))
+            pw = torch.zeros(ctx.gvec_fine.num_gvec, dtype=ctx.dtype,
+                             device=ctx.device)
+            pw[if_] = pw_c[ic]
+            return pw
+
+        rho_pw = to_fine(sum(rho_cs))
+        mag_pw = to_fine(rho_cs[0] - rho_cs[1]) if self.nmag else None
         if symmetrize and ctx.symmetry is not None:
             from ..symmetry import symmetrize_rho_g
             rho_pw = symmetrize_rho_g(rho_pw, ctx.gvec_fine, ctx.symmetry.ops)
+            if self.nmag:
+                mag_pw = symmetrize_rho_g(mag_pw, ctx.gvec_fine,
+                                          ctx.symmetry.ops)
         self.rho_pw = rho_pw
         self.rho_rg = ctx.fft_fine.to_real(rho_pw).real
+        if self.nmag:
+            self.mag_pw = mag_pw
+            self.mag_rg = ctx.fft_fine.to_real(mag_pw).real
 
         # MT density from the density matrix
         if symmetrize and ctx.symmetry is not None:
@@ -403,9 +442,13 @@ class FPDensity:
         for ia, (lab, _) in enumerate(uc.atoms):
             at = uc.atom_types[lab]
             asc = engine.classes[ia]
-            self.rho_mt[ia][:] = self._mt_density_one(at, asc, dm[ia][:, :, 0])
-            if ctx.num_spins == 2:
-                raise NotImplementedError("collinear LAPW density TODO")
+            d0 = self._mt_density_one(at, asc, dm[ia][:, :, 0])
+            if nsp == 2:
+                d1 = self._mt_density_one(at, asc, dm[ia][:, :, 1])
+                self.rho_mt[ia][:] = d0 + d1
+                self.mag_mt[ia][:] = d0 - d1
+            else:
+                self.rho_mt[ia][:] = d0
 
         # core
         for ia, (lab, _) in enumerate(uc.atoms):
@@ -442,6 +485,21 @@ class FPDensity:
                 pair_f[i2 * (i2 + 1) // 2 + i1] = n * asc.u[i1] * asc.u[i2]
         return mt_dm @ pair_f
 
+    def total_magnetization(self):
+        """Total and per-atom MT z-moments (collinear)."""
+        if not self.nmag:
+            return 0.0, []
+        ctx = self.ctx
+        it = float((self.mag_rg * ctx.theta_rg).sum()) \
+            * ctx.unit_cell.omega / ctx.fft_fine.size
+        per_atom = []
+        tot = it
+        for ia, (lab, _) in enumerate(ctx.unit_cell.atoms):
+            m = FOURPI * Y00 * float(self.grids.r2w[lab] @ self.mag_mt[ia][0])
+            per_atom.append(m)
+            tot += m
+        return tot, per_atom
+
     # -- mixing -------------------------------------------------------------
     def mixer_init(self, cfg_mixer):
         from ..mixer import Component, Linear, Anderson, Broyden2, AndersonStable
@@ -463,6 +521,11 @@ class FPDensity:
 
         for ia, (lab, _) in enumerate(ctx.unit_cell.atoms):
             comps.append(Component(f"rho_mt_{ia}", inner=make_inner_mt(lab)))
+        if self.nmag:
+            comps.append(Component("mag_pw", inner=inner_pw))
+            for ia, (lab, _) in enumerate(ctx.unit_cell.atoms):
+                comps.append(Component(f"mag_mt_{ia}",
+                                       inner=make_inner_mt(lab)))
         kind = cfg_mixer.type
         cls = {"linear": Linear, "anderson": Anderson, "broyden2": Broyden2,
                "anderson_stable": AndersonStable}[kind]
@@ -474,6 +537,10 @@ class FPDensity:
         v = {"rho_pw": self.rho_pw.clone()}
         for ia in range(len(self.rho_mt)):
             v[f"rho_mt_{ia}"] = torch.from_numpy(self.rho_mt[ia].copy())
+        if self.nmag:
+            v["mag_pw"] = self.mag_pw.clone()
+            for ia in range(len(self.mag_mt)):
+                v[f"mag_mt_{ia}"] = torch.from_numpy(self.mag_mt[ia].copy())
         return v
 
     def _set_from_mix(self, v):
@@ -481,6 +548,11 @@ class FPDensity:
         for ia in range(len(self.rho_mt)):
             self.rho_mt[ia] = v[f"rho_mt_{ia}"].numpy().copy()
         self.rho_rg = self.ctx.fft_fine.to_real(self.rho_pw).real
+        if self.nmag:
+            self.mag_pw = v["mag_pw"].clone()
+            for ia in range(len(self.mag_mt)):
+                self.mag_mt[ia] = v[f"mag_mt_{ia}"].numpy().copy()
+            self.mag_rg = self.ctx.fft_fine.to_real(self.mag_pw).real
 
     def mix(self) -> float:
         self.mixer.set_input(self._mix_value())
@@ -530,6 +602,10 @@ class FPPotential:
         self.veff_rg = torch.zeros_like(self.vha_rg)
         self.veff_mt = [np.zeros_like(v) for v in self.vha_mt]
         self.veff_pw = torch.zeros(ng, dtype=ctx.dtype, device=ctx.device)
+        if ctx.num_mag_dims == 1:
+            self.beff_rg = torch.zeros_like(self.vha_rg)
+            self.beff_mt = [np.zeros_like(v) for v in self.vha_mt]
+            self.beff_pw = torch.zeros(ng, dtype=ctx.dtype, device=ctx.device)
         self.vh_el = np.zeros(uc.num_atoms)
         self.energy_vha = 0.0
         self.sht = SHT(max(ctx.lmax_rho, ctx.lmax_pot))
@@ -711,6 +787,13 @@ class FPPotential:
                 self.rm2_inv_pw = ctx.fft_fine.to_pw(
                     (ctx.theta_rg / M ** 2).to(ctx.dtype))
 
+        if ctx.num_mag_dims == 1:
+            self.beff_pw = ctx.fft_fine.to_pw(
+                (self.beff_rg * ctx.theta_rg).to(ctx.dtype))
+            self.energy_bxc = it_inner(ctx, density.mag_rg, self.beff_rg) \
+                + mt_inner(ctx, grids, density.mag_mt, self.beff_mt)
+        else:
+            self.energy_bxc = 0.0
         self.energy_veff = it_inner(ctx, density.rho_rg, self.veff_rg) \
             + mt_inner(ctx, grids, density.rho_mt, self.veff_mt)
         self.energy_exc = it_inner(ctx, density.rho_rg, self.exc_rg) \
@@ -723,11 +806,33 @@ class FPPotential:
         ctx = self.ctx
         if ctx.is_gga:
             raise NotImplementedError("LAPW GGA TODO")
+        sht = self.sht
+        if ctx.num_mag_dims == 1:
+            # collinear: evaluate per spin channel (xc_rg_magnetic /
+            # xc_mt_magnetic); vxc = (v_up+v_dn)/2, B_z = (v_up-v_dn)/2
+            ru = (0.5 * (density.rho_rg + density.mag_rg)).clamp(min=0.0)
+            rd = (0.5 * (density.rho_rg - density.mag_rg)).clamp(min=0.0)
+            eps, vu, vd = xc_mod.evaluate_spin(ctx.xc_names, ru, rd)[:3]
+            self.exc_rg = eps
+            self.vxc_rg = 0.5 * (vu + vd)
+            self.beff_rg = 0.5 * (vu - vd)
+            for ia, (lab, _) in enumerate(ctx.unit_cell.atoms):
+                B = sht.rlm_backward[:, :ctx.lmmax_rho]
+                rho_tp = B @ density.rho_mt[ia]
+                mag_tp = B @ density.mag_mt[ia]
+                ru_tp = torch.from_numpy((0.5 * (rho_tp + mag_tp)).clip(min=0))
+                rd_tp = torch.from_numpy((0.5 * (rho_tp - mag_tp)).clip(min=0))
+                e_tp, vu_tp, vd_tp = xc_mod.evaluate_spin(
+                    ctx.xc_names, ru_tp, rd_tp)[:3]
+                F = sht.rlm_forward[:ctx.lmmax_pot]
+                self.exc_mt[ia] = F @ e_tp.numpy()
+                self.vxc_mt[ia] = F @ (0.5 * (vu_tp + vd_tp)).numpy()
+                self.beff_mt[ia] = F @ (0.5 * (vu_tp - vd_tp)).numpy()
+            return
         eps, vx, _ = xc_mod.evaluate(ctx.xc_names,
                                      density.rho_rg.clamp(min=0.0))
         self.exc_rg = eps
         self.vxc_rg = vx
-        sht = self.sht
         for ia, (lab, _) in enumerate(ctx.unit_cell.atoms):
             rho_tp = sht.rlm_backward[:, :ctx.lmmax_rho] @ density.rho_mt[ia]
             e_tp, v_tp, _ = xc_mod.evaluate(
@@ -832,6 +937,34 @@ class FPGroundState:
                         hri[:, rf_of][:, :, rf_of], optimize=True)
         return hmt
 
+    def _bmt_full(self, ia):
+        """MT matrix of the z magnetic field between basis functions
+        (reference Atom::b_radial_integrals + apply_bmt,
+        hamiltonian.cpp:147-205)."""
+        ctx = self.ctx
+        uc = ctx.unit_cell
+        lab, _ = uc.atoms[ia]
+        at = uc.atom_types[lab]
+        asc = self.classes[ia]
+        G = engine_gaunt(ctx, at)
+        nrf = at.num_rf
+        lmmax = ctx.lmmax_pot
+        bmt_lm = self.potential.beff_mt[ia]       # [lmmax_pot, nr]
+        r2w = self.grids.r2w[lab]
+        u = asc.u
+        lbl = l_by_lm(ctx.lmax_pot)
+        b_w = bmt_lm * r2w[None, :]
+        bri = np.einsum("ar,br,lr->lab", u, u, b_w, optimize=True)
+        lrf = np.array([at.indexr[i][0] for i in range(nrf)])
+        par = (lbl[:, None, None] + lrf[None, :, None] + lrf[None, None, :]) % 2
+        bri[par == 1] = 0.0
+        lm_of = np.array([b[2] for b in at.indexb])
+        rf_of = np.array([b[4] for b in at.indexb])
+        bmt = np.einsum("acb,cab->ab",
+                        G[np.ix_(lm_of, np.arange(lmmax), lm_of)],
+                        bri[:, rf_of][:, :, rf_of], optimize=True)
+        return torch.from_numpy(bmt).to(torch.complex128)
+
     def _omt_full(self, ia):
         uc = self.ctx.unit_cell
         lab, _ = uc.atoms[ia]
@@ -885,6 +1018,14 @@ class FPGroundState:
             self._hmt_cache = c = {}
         if ia not in c:
             c[ia] = torch.from_numpy(self._hmt_full(ia)).to(torch.complex128)
+        return c[ia]
+
+    def _bmt_cached(self, ia):
+        c = getattr(self, "_bmt_cache", None)
+        if c is None:
+            self._bmt_cache = c = {}
+        if ia not in c:
+            c[ia] = self._bmt_full(ia)
         return c[ia]
 
     def _omt_cached(self, ia):
@@ -974,13 +1115,43 @@ class FPGroundState:
         uc = ctx.unit_cell
         ngk = kp.num_gkvec
         Z = torch.from_numpy(kp.fv_evec).to(torch.complex128)
-        kp.psi[0, :, :ngk] = Z[:ngk, :].T
-        kp.mt_coeffs = []
+        mt_fv = []
         for ia in range(uc.num_atoms):
             C = self._C_cache.get((ik, ia))
             if C is None:
                 C = self._basis_c(ik, kp, ia)
-            kp.mt_coeffs.append((C @ Z).T.numpy())   # [nfv, mt_basis]
+            mt_fv.append(C @ Z)                     # [mt_basis, nfv]
+        if ctx.num_mag_dims == 0:
+            kp.psi[0, :, :ngk] = Z[:ngk, :].T
+            kp.mt_coeffs = [[m.T.numpy() for m in mt_fv]]
+            return
+        # second variation (diagonalize_fp_sv, diagonalize_fp.hpp:343):
+        # B matrix between fv states, then H_s = diag(e_fv) +/- B
+        nfv = Z.shape[1]
+        pw = Z[:ngk, :].T.contiguous()              # [nfv, ngk]
+        if not hasattr(self, "_bztheta_coarse"):
+            ic, if_ = ctx.coarse_fine_pairs
+            bpw_c = torch.zeros(ctx.gvec_coarse.num_gvec, dtype=ctx.dtype)
+            bpw_c[ic] = self.potential.beff_pw[if_]
+            self._bztheta_coarse = ctx.fft_coarse.to_real(bpw_c).real
+        psir = kp.fft.to_real(pw)
+        bpsi_pw = kp.fft.to_pw(self._bztheta_coarse * psir)
+        B = pw.conj() @ bpsi_pw.T                    # [nfv, nfv]
+        for ia in range(uc.num_atoms):
+            bmt = self._bmt_cached(ia)
+            S = mt_fv[ia]
+            B += S.conj().T @ (bmt @ S)
+        B = 0.5 * (B + B.conj().T)
+        ev = torch.from_numpy(kp.fv_eval)
+        kp.mt_coeffs = [None, None]
+        kp.sv_evec = [None, None]
+        for ispn, sgn in ((0, 1.0), (1, -1.0)):
+            Hs = sgn * B + torch.diag(ev).to(torch.complex128)
+            w, U = torch.linalg.eigh(Hs)
+            kp.eigvals[ispn, :nfv] = w.numpy()
+            kp.sv_evec[ispn] = U
+            kp.psi[ispn, :, :ngk] = (pw.T @ U).T
+            kp.mt_coeffs[ispn] = [(m @ U).T.numpy() for m in mt_fv]
 
     # ------------------------------------------------------- symmetrization
     def symmetrize_mt_dm(self, dm):
@@ -1052,8 +1223,10 @@ class FPGroundState:
             asc.generate_radial_functions()
         self._hmt_cache = {}
         self._omt_cache = {}
+        self._bmt_cache = {}
         self._C_cache = {}
-        for attr in ("_vtheta_coarse", "_theta_coarse"):
+        for attr in ("_vtheta_coarse", "_theta_coarse", "_kin_coarse",
+                     "_okin_coarse", "_bztheta_coarse"):
             self.__dict__.pop(attr, None)
         # 2) diagonalize all k
         for ik, kp in enumerate(self.kset.kpoints):
@@ -1071,11 +1244,12 @@ class FPGroundState:
         kset = self.kset
         core_sum = sum(asc.core_eval_sum for asc in self.classes)
         val_sum = kset.valence_eval_sum()
-        ekin = core_sum + val_sum - pot.energy_veff
+        bxc = getattr(pot, "energy_bxc", 0.0)
+        ekin = core_sum + val_sum - pot.energy_veff - bxc
         etot = ekin + pot.energy_exc + 0.5 * pot.energy_vha + pot.energy_enuc
         return {
             "total": etot, "ekin": ekin, "exc": pot.energy_exc,
-            "vha": pot.energy_vha, "enuc": pot.energy_enuc,
+            "vha": pot.energy_vha, "enuc": pot.energy_enuc, "bxc": bxc,
             "veff": pot.energy_veff, "core_eval_sum": core_sum,
             "valence_eval_sum": val_sum, "entropy_sum": kset.entropy_sum(),
             "free": etot + kset.entropy_sum(),

@@ -34,8 +34,11 @@ constexpr double SQ_ALPHA_HALF = 0.5 / REST_ENERGY;
 enum class Rel { none = 0, kh = 1, zora = 2, iora = 3, dirac = 4 };
 
 // ---------------------------------------------------------------- spline
-// Natural cubic spline on a non-uniform grid; evaluation + exact
-// piecewise integration (mirrors the role of the reference Spline).
+// Cubic spline with NOT-A-KNOT boundary conditions on a non-uniform
+// grid (the reference Spline uses not-a-knot, spline.hpp:21; scipy's
+// CubicSpline default, used on the Python side, is also not-a-knot —
+// keeping all interpolations consistent matters at the 1e-5 Ha level
+// for deep-core integrals).
 struct Spline {
     std::vector<double> x, a, b, c, d;  // f(t) = a + b*dt + c*dt^2 + d*dt^3
 
@@ -50,23 +53,46 @@ struct Spline {
         b.assign(n, 0.0);
         c.assign(n, 0.0);
         d.assign(n, 0.0);
-        if (n < 3) return;
-        std::vector<double> h(n - 1), al(n, 0.0), l(n), mu(n), z(n);
+        if (n < 4) return;
+        std::vector<double> h(n - 1);
         for (int i = 0; i < n - 1; i++) h[i] = xs[i + 1] - xs[i];
-        for (int i = 1; i < n - 1; i++)
-            al[i] = 3.0 * ((ys[i + 1] - ys[i]) / h[i] - (ys[i] - ys[i - 1]) / h[i - 1]);
-        l[0] = 1.0; mu[0] = 0.0; z[0] = 0.0;
-        for (int i = 1; i < n - 1; i++) {
-            l[i] = 2.0 * (xs[i + 1] - xs[i - 1]) - h[i - 1] * mu[i - 1];
-            mu[i] = h[i] / l[i];
-            z[i] = (al[i] - h[i - 1] * z[i - 1]) / l[i];
+        // solve for interior c_1..c_{n-2}; c_0, c_{n-1} eliminated via
+        // not-a-knot: d3 continuity at x_1 and x_{n-2}:
+        //   c0 = c1 + h0*(c1 - c2)/h1
+        //   c_{n-1} = c_{n-2} + h_{n-2}*(c_{n-2} - c_{n-3})/h_{n-3}
+        int m = n - 2;
+        std::vector<double> dl(m, 0.0), dm(m, 0.0), du(m, 0.0), rhs(m, 0.0);
+        for (int i = 1; i <= n - 2; i++) {
+            int j = i - 1;
+            dl[j] = h[i - 1];
+            dm[j] = 2.0 * (h[i - 1] + h[i]);
+            du[j] = h[i];
+            rhs[j] = 3.0 * ((ys[i + 1] - ys[i]) / h[i] - (ys[i] - ys[i - 1]) / h[i - 1]);
         }
-        l[n - 1] = 1.0; z[n - 1] = 0.0; c[n - 1] = 0.0;
-        for (int j = n - 2; j >= 0; j--) {
-            c[j] = z[j] - mu[j] * c[j + 1];
+        // fold the eliminated boundary c's into the first/last rows
+        // row j=0 couples c0: dm += h0*(1 + h0/h1), du += -h0*h0/h1
+        dm[0] += h[0] * (1.0 + h[0] / h[1]);
+        du[0] += -h[0] * h[0] / h[1];
+        dm[m - 1] += h[n - 2] * (1.0 + h[n - 2] / h[n - 3]);
+        dl[m - 1] += -h[n - 2] * h[n - 2] / h[n - 3];
+        // Thomas solve
+        for (int i = 1; i < m; i++) {
+            double w = dl[i] / dm[i - 1];
+            dm[i] -= w * du[i - 1];
+            rhs[i] -= w * rhs[i - 1];
+        }
+        c[n - 2] = rhs[m - 1] / dm[m - 1];
+        for (int i = m - 2; i >= 0; i--) {
+            c[i + 1] = (rhs[i] - du[i] * c[i + 2]) / dm[i];
+        }
+        c[0] = c[1] + h[0] * (c[1] - c[2]) / h[1];
+        c[n - 1] = c[n - 2] + h[n - 2] * (c[n - 2] - c[n - 3]) / h[n - 3];
+        for (int j = 0; j < n - 1; j++) {
             b[j] = (ys[j + 1] - ys[j]) / h[j] - h[j] * (c[j + 1] + 2.0 * c[j]) / 3.0;
             d[j] = (c[j + 1] - c[j]) / (3.0 * h[j]);
         }
+        // derivative coefficients at the last point (for deriv())
+        b[n - 1] = b[n - 2] + 2.0 * c[n - 2] * h[n - 2] + 3.0 * d[n - 2] * h[n - 2] * h[n - 2];
     }
 
     // value inside interval i at offset dt

@@ -83,6 +83,20 @@ def find_space_group(cell, tol: float = 1e-6) -> list[SymOp]:
     A = cell.lattice
     pos = cell.atom_positions_frac() % 1.0
     labels = [lab for lab, _ in cell.atoms]
+    # collinear magnetic structure: distinguish sublattices by the sign
+    # of the initial moment so the detected group is the magnetic
+    # subgroup (reference uses spin rotations of the full group; the
+    # colored subgroup is physically equivalent for symmetrization/IBZ)
+    vf = getattr(cell, "vector_fields", None)
+    if vf is not None and np.abs(vf).max() > 1e-8:
+        def _color(m):
+            if m > 1e-8:
+                return "+"
+            if m < -1e-8:
+                return "-"
+            return "0"
+        labels = [f"{lab}{_color(vf[i][2])}" for i, (lab, _)
+                  in enumerate(cell.atoms)]
     na = len(labels)
     Ws = _lattice_point_group(A, tol)
 
