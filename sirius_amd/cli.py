@@ -64,6 +64,65 @@ def main(argv=None):
             json.dump(rres, f, indent=2, default=float)
         return 0 if rres["converged"] else 1
 
+    if cfg.parameters.electronic_structure_method == "full_potential_lapwlo":
+        # FP-LAPW branch (reference: the same sirius.scf entry point
+        # dispatches on electronic_structure_method)
+        from .lapw.engine import make_lapw_context, FPGroundState
+
+        ctx = make_lapw_context(cfg, base_dir=base, device=args.device)
+        ctx.cfg._data["control"]["verbosity"] = 0
+        ctx.cfg.control.verbosity = 0
+        kset = KPointSet(ctx)
+        gs = FPGroundState(kset).initial_state()
+        cb = None
+        if args.verbosity >= 1:
+            def cb(it, etot, rms):
+                print(f"iter {it:3d}  Etot {etot:+.10f} Ha  rms {rms:.3e}",
+                      flush=True)
+        res = gs.find(num_dft_iter=args.num_iter, callback=cb)
+        mtot, per = gs.density.total_magnetization() \
+            if getattr(gs.density, "nmag", 0) else (0.0, [])
+        out = {
+            "ground_state": {
+                "energy": res["energy"],
+                "converged": res["converged"],
+                "num_scf_iterations": res["num_scf_iterations"],
+                "etot_history": res["etot_history"],
+                "rms_history": [float(x) for x in res["rms_history"]],
+                "scf_time": res["scf_time"],
+                "efermi": res["efermi"],
+                "magnetisation": {
+                    "total": [0.0, 0.0, mtot],
+                    "atoms": [[0.0, 0.0, m] for m in per]},
+                "core_leakage": sum(a.core_leakage for a in gs.classes),
+            },
+            "context": {
+                "num_bands": ctx.num_bands,
+                "num_kpoints": kset.num_kpoints,
+                "fft_grid": list(ctx.fft_fine.dims),
+                "num_gvec": ctx.gvec_fine.num_gvec,
+                "device": str(ctx.device),
+            },
+        }
+        etot = res["energy"]["total"]
+        print(f"total energy: {etot:.10f} Ha  "
+              f"({res['num_scf_iterations']} SCF iterations, "
+              f"{res['scf_time']:.1f} s)")
+        rc = 0
+        if args.test_against:
+            with open(args.test_against) as f:
+                refd = json.load(f)
+            ref = refd["ground_state"]["energy"]["total"]
+            de = abs(etot - ref)
+            ok = de < 1e-5
+            print(f"test_against: ref {ref:.10f}  |dE| = {de:.2e}  "
+                  f"{'OK' if ok else 'FAIL'}")
+            rc = 0 if ok else 1
+        if args.output:
+            with open(args.output, "w") as f:
+                json.dump(out, f, indent=2, default=float)
+        return rc
+
     ctx = SimulationContext(cfg, base_dir=base, device=args.device)
     kset = KPointSet(ctx)
     dft = DFTGroundState(kset).initial_state()
