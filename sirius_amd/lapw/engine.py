@@ -802,10 +802,80 @@ class FPPotential:
             uc.atom_types[lab].zn * self.vh_el[ia]
             for ia, (lab, _) in enumerate(uc.atoms))
 
+    def _grad_rg(self, f_rg):
+        """Cartesian gradient of a fine-grid field via iG."""
+        ctx = self.ctx
+        fg = ctx.fft_fine.to_pw(f_rg.to(ctx.dtype))
+        gv = ctx.gvec_fine.gkvec_t
+        return [ctx.fft_fine.to_real(1j * gv[:, d] * fg).real
+                for d in range(3)]
+
+    def _div_rg(self, F):
+        ctx = self.ctx
+        gv = ctx.gvec_fine.gkvec_t
+        out = None
+        for d in range(3):
+            fg = ctx.fft_fine.to_pw(F[d].to(ctx.dtype))
+            t = ctx.fft_fine.to_real(1j * gv[:, d] * fg).real
+            out = t if out is None else out + t
+        return out
+
+    def _xc_mt_gga_one(self, at, rho_mt):
+        """GGA in one MT sphere (reference xc_mt_nonmagnetic GGA branch,
+        xc_mt.cpp:46-110, divergence form)."""
+        ctx = self.ctx
+        sht = self.sht
+        r = at.r
+        if not hasattr(sht, "_sgrad"):
+            from .sht import rlm_surface_grad
+            sht._sgrad = rlm_surface_grad(sht.lmax, sht.theta, sht.phi)
+            st, ct = np.sin(sht.theta), np.cos(sht.theta)
+            sht._rhat = np.stack([st * np.cos(sht.phi),
+                                  st * np.sin(sht.phi), ct], axis=1)
+        B = sht.rlm_backward[:, :ctx.lmmax_rho]
+        SG = sht._sgrad[:, :ctx.lmmax_rho]        # [ntp, lm, 3]
+        rhat = sht._rhat                           # [ntp, 3]
+        drho = CubicSpline(r, rho_mt, axis=1).derivative()(r)   # [lm, nr]
+        rho_tp = B @ rho_mt                        # [ntp, nr]
+        dr_tp = B @ drho
+        ang_tp = np.einsum("tlx,lr->txr", SG, rho_mt / r[None, :])
+        grad = dr_tp[:, None, :] * rhat[:, :, None] + ang_tp    # [ntp,3,nr]
+        sigma = np.einsum("txr,txr->tr", grad, grad)
+        eps, vrho, vsigma = xc_mod.evaluate(
+            ctx.xc_names, torch.from_numpy(rho_tp.clip(min=0.0)),
+            torch.from_numpy(sigma))
+        vs = vsigma.numpy()
+        F = sht.rlm_forward[:ctx.lmmax_rho]
+        div = np.zeros_like(rho_tp)
+        for x in range(3):
+            gx_tp = vs * grad[:, x, :]
+            gx_lm = F @ gx_tp
+            dgx = CubicSpline(r, gx_lm, axis=1).derivative()(r)
+            div += (B @ dgx) * rhat[:, x:x + 1] \
+                + np.einsum("tl,lr->tr", SG[:, :, x], gx_lm / r[None, :])
+        vxc_tp = vrho.numpy() - 2.0 * div
+        Fp = sht.rlm_forward[:ctx.lmmax_pot]
+        return Fp @ eps.numpy(), Fp @ vxc_tp
+
     def _xc(self, density: FPDensity):
         ctx = self.ctx
         if ctx.is_gga:
-            raise NotImplementedError("LAPW GGA TODO")
+            if ctx.num_mag_dims:
+                raise NotImplementedError("magnetic LAPW GGA TODO")
+            # interstitial (like xc_rg_nonmagnetic GGA, xc.cpp:26)
+            rho = density.rho_rg.clamp(min=0.0)
+            grads = self._grad_rg(density.rho_rg)
+            sigma = grads[0] ** 2 + grads[1] ** 2 + grads[2] ** 2
+            eps, vrho, vsigma = xc_mod.evaluate(ctx.xc_names, rho, sigma)
+            div = self._div_rg([vsigma * g for g in grads])
+            self.exc_rg = eps
+            self.vxc_rg = vrho - 2.0 * div
+            for ia, (lab, _) in enumerate(ctx.unit_cell.atoms):
+                at = ctx.unit_cell.atom_types[lab]
+                e_lm, v_lm = self._xc_mt_gga_one(at, density.rho_mt[ia])
+                self.exc_mt[ia] = e_lm
+                self.vxc_mt[ia] = v_lm
+            return
         sht = self.sht
         if ctx.num_mag_dims == 1:
             # collinear: evaluate per spin channel (xc_rg_magnetic /

@@ -116,6 +116,34 @@ def ylm_to_rlm(fylm: np.ndarray) -> np.ndarray:
     return np.ascontiguousarray(out.real)
 
 
+def rlm_surface_grad(lmax: int, theta: np.ndarray, phi: np.ndarray,
+                     h: float = 1e-6) -> np.ndarray:
+    """Surface gradient of R_lm on the unit sphere: [ntp, lmmax, 3]
+    cartesian components of grad_S R_lm (the gradient of the degree-0
+    homogeneous extension), via central differences — accurate to ~1e-9,
+    well below the XC quadrature level."""
+    st, ct = np.sin(theta), np.cos(theta)
+    v = np.stack([st * np.cos(phi), st * np.sin(phi), ct], axis=1)
+    out = np.empty((len(theta), _lmmax(lmax), 3))
+    for ax in range(3):
+        e = np.zeros(3)
+        e[ax] = h
+        vp = v + e
+        vm = v - e
+        vp /= np.linalg.norm(vp, axis=1, keepdims=True)
+        vm /= np.linalg.norm(vm, axis=1, keepdims=True)
+
+        def ang(w):
+            r = np.linalg.norm(w, axis=1)
+            th = np.arccos(np.clip(w[:, 2] / r, -1, 1))
+            ph = np.arctan2(w[:, 1], w[:, 0])
+            return th, ph
+        Rp = _rlm(lmax, *ang(vp))
+        Rm = _rlm(lmax, *ang(vm))
+        out[:, :, ax] = (Rp - Rm) / (2 * h)
+    return out
+
+
 @lru_cache(maxsize=8)
 def gaunt_hybrid(lmax1: int, lmax3: int, lmax2: int) -> np.ndarray:
     """<Y_{l1 m1} | R_{l3 m3} | Y_{l2 m2}> dense table

@@ -73,6 +73,11 @@ def test_nio_afm_test16():
     assert abs(mtot) < 1e-6
 
 
+def test_nio_pbe_test17():
+    gs, res = run_deck("test17")
+    assert abs(res["energy"]["total"] + 1595.92851879) < 1e-5
+
+
 def test_yn_iora_test18():
     """YN with IORA valence relativity. Documented residual: 2.2e-3 Ha
     (core-eigenvalue systematics on Y, see NEXT.md)."""
