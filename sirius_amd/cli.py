@@ -37,9 +37,9 @@ def main(argv=None):
                     help="relax atomic positions (fixed cell)")
     ap.add_argument("--vc-relax", action="store_true",
                     help="relax positions and cell")
-    ap.add_argument("--save-state", default=None, metavar="FILE.npz",
+    ap.add_argument("--save-state", default=None, metavar="FILE.h5|FILE.npz",
                     help="write the converged density/potential state")
-    ap.add_argument("--restart", default=None, metavar="FILE.npz",
+    ap.add_argument("--restart", default=None, metavar="FILE.h5|FILE.npz",
                     help="start the SCF from a saved state")
     ap.add_argument("--verbosity", "-v", type=int, default=1)
     args = ap.parse_args(argv)
@@ -127,9 +127,11 @@ def main(argv=None):
     kset = KPointSet(ctx)
     dft = DFTGroundState(kset).initial_state()
     if args.restart:
-        from .checkpoint import load_state
-
-        load_state(args.restart, dft)
+        if args.restart.endswith(".h5"):
+            from .checkpoint import load_state_h5 as _load
+        else:
+            from .checkpoint import load_state as _load
+        _load(args.restart, dft)
     cb = None
     if args.verbosity >= 1:
         def cb(it, etot, rms):
@@ -138,9 +140,11 @@ def main(argv=None):
     res = dft.find(num_dft_iter=args.num_iter, callback=cb)
     res["setup_and_scf_time"] = time.time() - t0
     if args.save_state:
-        from .checkpoint import save_state
-
-        save_state(args.save_state, dft)
+        if args.save_state.endswith(".h5"):
+            from .checkpoint import save_state_h5 as _save
+        else:
+            from .checkpoint import save_state as _save
+        _save(args.save_state, dft)
 
     out = {
         "ground_state": {
