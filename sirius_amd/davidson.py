@@ -91,10 +91,18 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
              tol_occ: float, tol_empty: float,
              num_steps: int = 20, subspace_size: int = 2,
              min_occupancy: float = 1e-14,
-             extra_ortho: bool = False) -> DavidsonResult:
+             extra_ortho: bool = False, locking: bool = True,
+             early_restart: float = 0.5) -> DavidsonResult:
     """Solve for the `nb` lowest eigenpairs of H ψ = ε S ψ (S=I or USPP S).
 
     apply_h_s(phi [n, nG]) -> (hphi, sphi|None).
+
+    Locking (reference davidson.hpp:704-723): at every subspace restart
+    the leading consecutively-converged Ritz vectors are frozen; the
+    subsequent subspace eigenproblems shrink by the locked count
+    (evp_work counts ((N - locked)/nb)^3, davidson.hpp:828-834).
+    `early_restart` triggers a restart before the subspace is full when
+    few bands remain unconverged and >5 are lockable (davidson.hpp:622).
     """
     nb, nG = psi0.shape
     num_phi_max = min(max(subspace_size * nb, nb + 1), max(nG // 2, nb))
@@ -130,19 +138,24 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
     eval_old = np.full(nb, 1e10)
     niter = 0
     converged = False
+    nlock = 0                      # number of locked (frozen) leading bands
+    eval_locked = np.zeros(0)
 
     for it in range(num_steps):
         niter = it + 1
-        ev = evals[:nb].real.cpu().numpy()
+        nb_act = nb - nlock        # bands still solved in the active block
+        ev = np.concatenate([eval_locked,
+                             evals[:nb_act].real.cpu().numpy()])
         unconv = np.nonzero(np.abs(ev - eval_old[: len(ev)]) > tol)[0]
+        unconv = unconv[unconv >= nlock]
         eval_old = ev.copy()
         if len(unconv) == 0:
             converged = True
             break
 
-        N = phi.shape[0]
+        N = phi.shape[0]           # active subspace size (excl. locked)
         # residuals of unconverged bands: r_j = (H - e_j S) phi Z_j
-        idx = torch.from_numpy(unconv).to(psi0.device)
+        idx = torch.from_numpy(unconv - nlock).to(psi0.device)
         Zs = Z[:, idx]                                     # [N, n]
         e = evals[idx].real
         hpsi = la.transform(Zs, hphi)                       # [n, nG]
@@ -172,45 +185,65 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
         res = res / torch.linalg.vector_norm(res, dim=1, keepdim=True).to(res.dtype)
 
         n_new = res.shape[0]
-        if N + n_new > num_phi_max:
-            # restart: collapse subspace to current Ritz vectors
-            Znb = Z[:, :nb]
+        # leading consecutively-converged active bands are lockable
+        lockable = int(unconv[0] - nlock) if len(unconv) else nb_act
+        should_restart = (nlock + N + n_new > num_phi_max) or \
+            (locking and lockable > 5
+             and len(unconv) < early_restart * lockable)
+        if should_restart:
+            # restart: collapse the active subspace to its Ritz vectors
+            Znb = Z[:, :nb_act]
             psi = la.transform(Znb, phi)
             hpsi_f = la.transform(Znb, hphi)
             spsi_f = la.transform(Znb, sphi) if sphi is not None else None
-            phi_buf[:nb] = psi
-            hphi_buf[:nb] = hpsi_f
-            phi = phi_buf[:nb]
-            hphi = hphi_buf[:nb]
+            phi_buf[nlock:nlock + nb_act] = psi
+            hphi_buf[nlock:nlock + nb_act] = hpsi_f
             if spsi_f is not None:
-                sphi_buf[:nb] = spsi_f
-                sphi = sphi_buf[:nb]
-            H = torch.diag(evals[:nb].to(H.dtype))
-            evals = evals[:nb].clone()
-            Z = torch.eye(nb, dtype=H.dtype, device=H.device)
-            N = nb
-            if N + n_new > num_phi_max:
-                n_new = num_phi_max - N
+                sphi_buf[nlock:nlock + nb_act] = spsi_f
+            ev_act = evals[:nb_act].clone()
+            if locking and lockable > 0:
+                # freeze the converged leading Ritz vectors
+                eval_locked = np.concatenate(
+                    [eval_locked, ev_act[:lockable].real.cpu().numpy()])
+                nlock += lockable
+                nb_act = nb - nlock
+                ev_act = ev_act[lockable:]
+            phi = phi_buf[nlock:nlock + nb_act]
+            hphi = hphi_buf[nlock:nlock + nb_act]
+            if sphi is not None:
+                sphi = sphi_buf[nlock:nlock + nb_act]
+            H = torch.diag(ev_act.to(phi_buf.dtype))
+            evals = ev_act
+            Z = torch.eye(nb_act, dtype=H.dtype, device=H.device)
+            N = nb_act
+            if nlock + N + n_new > num_phi_max:
+                n_new = num_phi_max - N - nlock
                 res = res[:n_new]
                 if n_new <= 0:
                     break
 
         hnew, snew = apply_h_s(res)
-        outs = (phi_buf[N:], hphi_buf[N:],
-                sphi_buf[N:] if sphi is not None else None)
+        lo = nlock + N
+        outs = (phi_buf[lo:], hphi_buf[lo:],
+                sphi_buf[lo:] if sphi is not None else None)
+        # project out the full basis incl. locked vectors
+        full_phi = phi_buf[:lo]
+        full_h = hphi_buf[:lo]
+        full_s = sphi_buf[:lo] if sphi is not None else None
         if extra_ortho:
-            res, hnew, snew = _ortho_joint(res, hnew, snew, phi, hphi, sphi)
-        res, hnew, snew = _ortho_joint(res, hnew, snew, phi, hphi, sphi,
-                                       outs=outs)
+            res, hnew, snew = _ortho_joint(res, hnew, snew, full_phi,
+                                           full_h, full_s)
+        res, hnew, snew = _ortho_joint(res, hnew, snew, full_phi, full_h,
+                                       full_s, outs=outs)
         if res.shape[0] == 0:
             converged = True
             break
         # subspace grown in place by _ortho_joint(outs=...)
         nn = res.shape[0]
-        phi = phi_buf[:N + nn]
-        hphi = hphi_buf[:N + nn]
+        phi = phi_buf[nlock:lo + nn]
+        hphi = hphi_buf[nlock:lo + nn]
         if sphi is not None:
-            sphi = sphi_buf[:N + nn]
+            sphi = sphi_buf[nlock:lo + nn]
         Nn = phi.shape[0]
         Hn = torch.empty(Nn, Nn, dtype=H.dtype, device=H.device)
         Hn[:N, :N] = H
@@ -221,7 +254,15 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
         evals, Z = la.eigh(H)
         evp_work += (Nn / nb) ** 3
 
-    psi = la.transform(Z[:, :nb], phi)
-    return DavidsonResult(eval=evals[:nb].real.cpu().numpy(), psi=psi,
+    nb_act = nb - nlock
+    psi_act = la.transform(Z[:, :nb_act], phi)
+    if nlock:
+        psi = torch.cat([phi_buf[:nlock], psi_act], dim=0)
+        ev_out = np.concatenate([eval_locked,
+                                 evals[:nb_act].real.cpu().numpy()])
+    else:
+        psi = psi_act
+        ev_out = evals[:nb].real.cpu().numpy()
+    return DavidsonResult(eval=ev_out, psi=psi,
                           niter=niter, converged=converged,
                           evp_work=evp_work)

@@ -114,7 +114,9 @@ def diagonalize(ctx, h0: Hamiltonian0, kset: KPointSet, itsol_tol: float) -> boo
                 kp.psi[ispn], h_diag, o_diag, occ=kp.occ[ispn],
                 tol_occ=itsol_tol, tol_empty=empy_tol,
                 num_steps=itso.num_steps, subspace_size=itso.subspace_size,
-                min_occupancy=itso.min_occupancy, extra_ortho=itso.extra_ortho)
+                min_occupancy=itso.min_occupancy, extra_ortho=itso.extra_ortho,
+                locking=bool(itso.locking),
+                early_restart=float(itso.early_restart))
             kp.psi[ispn] = res.psi
             kp.eigvals[ispn] = res.eval
             ctx.counters["num_itsol_steps"] += res.niter

@@ -59,30 +59,60 @@ class BetaProjectors:
                 cols.append(z * rl[:, ylm_mod.lm_index(l, m)] * fl[irf])
             cols_t[lab] = np.stack(cols, axis=1)
 
-        # assemble per atom with phase e^{-i(G+k)·τ}
-        blocks = []
+        # assemble per atom with phase e^{-i(G+k)·τ}.  The per-TYPE column
+        # blocks above are small ([nGk, nbf_type], computed once); the
+        # per-ATOM phase expansion is O(nGk x nbf_total) and runs on
+        # device through the beta_phase kernel (create_beta_gk.cu twin;
+        # was a host numpy loop that dominated setup at many atoms).
         self.atom_offsets = []
         self.atom_nbf = []
+        type_row_off = {}
+        off_t = 0
+        for lab in uc.type_labels:
+            type_row_off[lab] = off_t
+            off_t += uc.atom_types[lab].num_beta_lm
         off = 0
-        mk = (g.miller + g.k_frac).astype(np.float64)  # (G+k) in frac recip
         for ia, (lab, tau) in enumerate(uc.atoms):
             nb = uc.atom_types[lab].num_beta_lm
             self.atom_offsets.append(off)
             self.atom_nbf.append(nb)
             off += nb
-            if nb == 0:
-                continue
-            phase = np.exp(-2j * math.pi * (mk @ tau))  # [nGk]
-            blocks.append(cols_t[lab] * phase[:, None])
         self.num_beta_total = off
+        ngk = len(glen)
+        mk = (g.miller + g.k_frac).astype(np.float64)  # (G+k) in frac recip
         # stored TRANSPOSED [nbf_tot, nGk]: both ⟨β|ψ⟩ (Gram) and the
         # β·(D⟨β|ψ⟩) apply (transform) then run on the MFMA zgemm kernels
         # (ops/src/zgemm_gram.hip) with fully coalesced row-major access.
-        if blocks:
+        if off == 0:
+            self.beta_t = torch.zeros(0, ngk, dtype=ctx.dtype, device=dev)
+        elif str(dev).startswith("cuda"):
+            from . import ops
+
+            ext = ops.get_ext(required=True)
+            bt_types = np.concatenate(
+                [cols_t[lab].T for lab in uc.type_labels], axis=0)
+            bt_types_d = torch.from_numpy(
+                np.ascontiguousarray(bt_types)).to(dev)
+            mk_d = torch.from_numpy(np.ascontiguousarray(mk)).to(dev)
+            tau_d = torch.from_numpy(
+                uc.atom_positions_frac().copy()).to(dev)
+            t_off = torch.tensor([type_row_off[lab] for lab, _ in uc.atoms],
+                                 dtype=torch.int32, device=dev)
+            a_nbf = torch.tensor(self.atom_nbf, dtype=torch.int32, device=dev)
+            a_off = torch.tensor(self.atom_offsets, dtype=torch.int32,
+                                 device=dev)
+            self.beta_t = torch.empty(off, ngk, dtype=ctx.dtype, device=dev)
+            ext.beta_phase(bt_types_d, mk_d, tau_d, t_off, a_nbf, a_off,
+                           self.beta_t)
+        else:
+            blocks = []
+            for ia, (lab, tau) in enumerate(uc.atoms):
+                if uc.atom_types[lab].num_beta_lm == 0:
+                    continue
+                phase = np.exp(-2j * math.pi * (mk @ tau))  # [nGk]
+                blocks.append(cols_t[lab] * phase[:, None])
             bt = np.ascontiguousarray(np.concatenate(blocks, axis=1).T)
             self.beta_t = torch.from_numpy(bt).to(dev)
-        else:
-            self.beta_t = torch.zeros(0, len(glen), dtype=ctx.dtype, device=dev)
 
     def inner(self, psi: torch.Tensor) -> torch.Tensor:
         """⟨β|ψ⟩ = β^H ψ: [nbf_tot, nb] (MFMA Gram kernel; reference

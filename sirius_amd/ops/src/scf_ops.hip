@@ -252,6 +252,21 @@ void mul_veff(torch::Tensor grid, torch::Tensor veff) {
                        (double2*)grid.data_ptr(), veff.data_ptr<double>(), gs, nb);
 }
 
+void beta_phase(torch::Tensor beta_types, torch::Tensor mk, torch::Tensor tau,
+                torch::Tensor type_off, torch::Tensor atom_nbf,
+                torch::Tensor atom_off, torch::Tensor beta_out) {
+    CHECK_HIP(beta_types);
+    long ng = beta_types.size(-1);
+    int na = tau.size(0);
+    dim3 grid(grid_1d(ng, 256), na);
+    hipLaunchKernelGGL(beta_phase_kernel, grid, dim3(256), 0, cur_stream(),
+                       (const double2*)beta_types.data_ptr(),
+                       mk.data_ptr<double>(), tau.data_ptr<double>(),
+                       type_off.data_ptr<int>(), atom_nbf.data_ptr<int>(),
+                       atom_off.data_ptr<int>(),
+                       (double2*)beta_out.data_ptr(), ng, na);
+}
+
 void density_acc(torch::Tensor psi_r, torch::Tensor w, torch::Tensor rho) {
     CHECK_HIP(psi_r);
     long gs = rho.numel();
@@ -288,6 +303,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("out"), py::arg("scale") = 1.0);
     m.def("mul_veff", &mul_veff, "in-place psi(r) *= V(r)");
     m.def("density_acc", &density_acc, "rho(r) += sum_b w_b |psi_b(r)|^2");
+    m.def("beta_phase", &beta_phase,
+          "beta(G+k) per atom = type columns x e^{-i(G+k).tau} "
+          "(create_beta_gk.cu twin)");
     m.def("residual_precond", &residual_precond,
           "res=(h-e*s)psi/P fused with norm reduction; returns norms^2");
 }
