@@ -42,6 +42,7 @@ _DEFAULTS: dict[str, dict[str, Any]] = {
         "use_symmetry": True,
         "use_ibz": True,
         "num_mag_dims": 0,
+        "precision_wf": "fp64",
         "pw_cutoff": 20.0,                  # a.u.^-1, fine grid |G| cutoff
         "gk_cutoff": 6.0,                   # a.u.^-1, |G+k| cutoff
         "aw_cutoff": 0.0,

@@ -25,8 +25,11 @@ from .gvec import Gvec
 
 
 def _ext_for(t: torch.Tensor):
-    """Native extension when running on GPU (required there), else None."""
-    if t.is_cuda:
+    """Native extension when running on GPU (required there), else None.
+    The HIP kernels are fp64 (double2); complex64 tensors (fp32
+    wave-function mode) take the torch path (cgemm/cFFT, still fully on
+    device)."""
+    if t.is_cuda and t.dtype in (torch.complex128, torch.float64):
         from .. import ops
 
         return ops.get_ext(required=True)

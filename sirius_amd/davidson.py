@@ -160,7 +160,7 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
         e = evals[idx].real
         hpsi = la.transform(Zs, hphi)                       # [n, nG]
         spsi = la.transform(Zs, sphi if sphi is not None else phi)
-        if hpsi.is_cuda:
+        if hpsi.is_cuda and hpsi.dtype == torch.complex128:
             from . import ops
 
             ext = ops.get_ext(required=True)

@@ -99,25 +99,38 @@ def initialize_subspace(ctx, kp, hk):
         kp.eigvals[ispn] = evals[:nb].real.cpu().numpy()
 
 
-def diagonalize(ctx, h0: Hamiltonian0, kset: KPointSet, itsol_tol: float) -> bool:
-    """Davidson for all local k-points/spins (reference diagonalize.hpp)."""
+def diagonalize(ctx, h0: Hamiltonian0, kset: KPointSet, itsol_tol: float,
+                wf_dtype=None) -> bool:
+    """Davidson for all local k-points/spins (reference diagonalize.hpp).
+
+    wf_dtype=torch.complex64 runs the fp32 wave-function mode
+    (reference precision_wf, dft_ground_state.cpp:269-304): trial
+    vectors, H/S application and the subspace algebra all in single
+    precision; converged psi is stored back in fp64."""
     itso = ctx.cfg.iterative_solver
     empy_tol = max(itsol_tol * itso.tolerance_ratio, itso.empty_states_tolerance)
+    fp32 = wf_dtype == torch.complex64
     all_conv = True
     for kp in kset:
         hk = h0(kp)
         o_diag = hk.o_diag()
+        if fp32:
+            o_diag = o_diag.to(torch.float32)
         for ispn in range(ctx.num_spin_steps):
             h_diag = hk.h_diag(ispn)
+            psi0 = kp.psi[ispn]
+            if fp32:
+                h_diag = h_diag.to(torch.float32)
+                psi0 = psi0.to(torch.complex64)
             res = davidson(
                 lambda p, s=ispn: hk.apply_h_s(p, s),
-                kp.psi[ispn], h_diag, o_diag, occ=kp.occ[ispn],
+                psi0, h_diag, o_diag, occ=kp.occ[ispn],
                 tol_occ=itsol_tol, tol_empty=empy_tol,
                 num_steps=itso.num_steps, subspace_size=itso.subspace_size,
                 min_occupancy=itso.min_occupancy, extra_ortho=itso.extra_ortho,
                 locking=bool(itso.locking),
                 early_restart=float(itso.early_restart))
-            kp.psi[ispn] = res.psi
+            kp.psi[ispn] = res.psi.to(torch.complex128) if fp32 else res.psi
             kp.eigvals[ispn] = res.eval
             ctx.counters["num_itsol_steps"] += res.niter
             ctx.counters["band_evp_work_count"] += res.evp_work
@@ -212,13 +225,24 @@ class DFTGroundState:
         num_iter = -1
         t0 = time.time()
 
+        # fp32 wave-function mode with runtime fp64 promotion
+        # (reference precision_wf + fp32_to_fp64_rms,
+        # dft_ground_state.cpp:269-304)
+        import torch as _torch
+        wf_dtype = _torch.complex64 \
+            if str(getattr(p, "precision_wf", "fp64")) == "fp32" else None
+        fp32_rms = float(ctx.cfg.settings.fp32_to_fp64_rms) or \
+            max(10.0 * density_tol, 1e-5)
+
         for it in range(num_dft_iter):
             with profiler("scf_iteration"):
                 with profiler("Hamiltonian0"):
                     h0 = Hamiltonian0(ctx, self.potential, self.density)
                     self.h0 = h0  # kept for post-SCF forces (D at diag-time V)
                 with profiler("diagonalize"):
-                    bands_converged = diagonalize(ctx, h0, self.kset, itsol_tol)
+                    bands_converged = diagonalize(ctx, h0, self.kset,
+                                                  itsol_tol,
+                                                  wf_dtype=wf_dtype)
                 with profiler("occupancies"):
                     self.kset.find_band_occupancies()
                 with profiler("density"):
@@ -232,6 +256,13 @@ class DFTGroundState:
 
             with profiler("mix"):
                 rms = self.density.mix()
+
+            if wf_dtype is not None and rms < fp32_rms:
+                # promote to fp64 for the remaining iterations
+                wf_dtype = None
+                if ctx.cfg.control.verbosity >= 1:
+                    print(f"[scf] switching wave functions fp32 -> fp64 "
+                          f"at iteration {it} (rms {rms:.2e})", flush=True)
 
             tol = rms
             tol = min(itso.tolerance_scale[0] * tol,

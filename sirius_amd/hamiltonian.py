@@ -352,25 +352,48 @@ class HamiltonianK:
             return torch.cat([d, d])
         return d
 
+    def _fp32_tensors(self, ispn: int):
+        """complex64/float32 copies of the apply tables for the fp32
+        wave-function mode (reference precision_wf,
+        dft_ground_state.cpp:269-304); cached per Hamiltonian0."""
+        c = getattr(self, "_fp32_cache", None)
+        if c is None:
+            self._fp32_cache = c = {}
+        if ispn not in c:
+            c[ispn] = (self.h0.veff_r_coarse[ispn].to(torch.float32),
+                       self.ekin.to(torch.float32),
+                       self.bp.beta_t.to(torch.complex64)
+                       if self.bp.num_beta_total else None,
+                       self.D[ispn].to(torch.complex64),
+                       self.Q.to(torch.complex64)
+                       if self.Q is not None else None)
+        return c[ispn]
+
     def apply_h_s(self, psi: torch.Tensor, ispn: int = 0):
-        """psi [nb, nGk] (or [nb, 2·nGk] spinors) -> (hpsi, spsi)."""
+        """psi [nb, nGk] (or [nb, 2·nGk] spinors) -> (hpsi, spsi).
+
+        complex64 input runs the whole chain in fp32 (cFFT/cgemm via
+        torch; the fp64 HIP kernels are bypassed)."""
         if self.ctx.nc_magnetism:
             return self._apply_h_s_nc(psi)
         kp = self.kp
         self.ctx.counters["local_operator_num_applied"] += psi.shape[0]
+        fp32 = psi.dtype == torch.complex64
+        if fp32:
+            veff_r, ekin, beta_t, D, Q = self._fp32_tensors(ispn)
+        else:
+            veff_r, ekin = self.h0.veff_r_coarse[ispn], self.ekin
+            beta_t, D, Q = self.bp.beta_t, self.D[ispn], self.Q
         # fused local operator: FFT⁻¹ → ×V_eff(r) → FFT → +½|G+k|²ψ
-        hpsi = kp.fft.apply_veff_kinetic(psi, self.h0.veff_r_coarse[ispn],
-                                         self.ekin)
+        hpsi = kp.fft.apply_veff_kinetic(psi, veff_r, ekin)
 
         spsi = None
         if self.bp.num_beta_total:
-            bphi = self.bp.inner(psi)           # [nbf, nb]
-            la.transform(self.D[ispn] @ bphi, self.bp.beta_t,
-                         out=hpsi, accumulate=True)
-            if self.Q is not None:
+            bphi = la.inner(beta_t, psi)        # [nbf, nb]
+            la.transform(D @ bphi, beta_t, out=hpsi, accumulate=True)
+            if Q is not None:
                 spsi = psi.clone()
-                la.transform(self.Q @ bphi, self.bp.beta_t,
-                             out=spsi, accumulate=True)
+                la.transform(Q @ bphi, beta_t, out=spsi, accumulate=True)
         if self.ctx.hubbard is not None:
             self.ctx.hubbard.apply(kp, self, psi, hpsi, ispn)
         return hpsi, spsi

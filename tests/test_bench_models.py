@@ -42,3 +42,17 @@ def test_fe_paw_model():
 def test_si2_model():
     ctx, res = _run("si2")
     assert ctx.unit_cell.num_atoms == 2
+
+
+def test_fp32_wf_promotion():
+    """fp32 wave functions with runtime fp64 promotion converge to the
+    fp64 energy (reference precision_wf, dft_ground_state.cpp:269-304)."""
+    ctx64, r64 = _run("si2", niter=30)
+    from sirius_amd.models.synthetic import make_named_context
+    ctx = make_named_context("si2")
+    ctx.cfg._data["parameters"]["precision_wf"] = "fp32"
+    ctx.cfg.parameters.precision_wf = "fp32"
+    kset = KPointSet(ctx)
+    dft = DFTGroundState(kset).initial_state()
+    r32 = dft.find(num_dft_iter=30)
+    assert abs(r32["energy"]["total"] - r64["energy"]["total"]) < 1e-8
