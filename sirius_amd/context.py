@@ -277,6 +277,27 @@ class SimulationContext:
                          "band_evp_work_count": 0.0,
                          "num_itsol_steps": 0}
 
+        # band-parallel communicator grid (control.mpi_grid_dims;
+        # reference init_comm splits world into comm_k x comm_band,
+        # simulation_context.cpp:1301-1334).  MI355X design: wave
+        # functions replicated within a band group (288 GB HBM holds the
+        # largest BASELINE cells), compute split over bands; the
+        # subspace algebra stays replicated so no distributed eigensolve
+        # is needed.
+        from .parallel import get_comm as _get_comm
+        from .parallel.comm import make_band_comm
+        grid_dims = list(cfg.control.mpi_grid_dims)
+        npb = max(1, int(np.prod(grid_dims)))
+        world = _get_comm()
+        if world.active and npb > 1:
+            self.band_comm, self.kcolor, self.num_kgroups = make_band_comm(npb)
+        else:
+            from .parallel.comm import Comm as _Comm
+            self.band_comm = _Comm(None) if not world.active else _Comm()
+            self.band_comm.rank, self.band_comm.size = 0, 1
+            self.kcolor = world.rank if world.active else 0
+            self.num_kgroups = world.size if world.active else 1
+
         # crystal symmetry (space group + IBZ; reference: Crystal_symmetry)
         # (noncollinear case needs spin-rotation symmetrization — run the
         # full k-mesh instead)

@@ -178,6 +178,7 @@ class Density:
                 ng = kp.num_gkvec
                 occ = torch.from_numpy(kp.occ[0]).to(ctx.device)
                 sel = torch.nonzero(occ.abs() > min_occ).reshape(-1)
+                sel = self._band_slice(sel)
                 if len(sel) == 0:
                     continue
                 w = ((kp.weight / ctx.unit_cell.omega) * occ[sel]).to(ctx.rdtype)
@@ -196,6 +197,7 @@ class Density:
                 for ispn in range(nsp):
                     occ = torch.from_numpy(kp.occ[ispn]).to(ctx.device)
                     sel = torch.nonzero(occ.abs() > min_occ).reshape(-1)
+                    sel = self._band_slice(sel)
                     if len(sel) == 0:
                         continue
                     w = ((kp.weight / ctx.unit_cell.omega) * occ[sel]).to(ctx.rdtype)
@@ -335,6 +337,7 @@ class Density:
             for ispn in range(ctx.num_spins):
                 occ = torch.from_numpy(kp.occ[ispn]).to(ctx.device)
                 sel = torch.nonzero(occ.abs() > min_occ).reshape(-1)
+                sel = self._band_slice(sel)
                 if len(sel) == 0:
                     continue
                 bpsi = bp.inner(kp.psi[ispn][sel])     # [nbf_tot, nocc]
@@ -396,6 +399,15 @@ class Density:
         return n
 
     # -- mixing ------------------------------------------------------------
+
+    def _band_slice(self, sel):
+        """Slice an occupied-band index set across band-group ranks
+        (band-parallel density accumulation; the world allreduce then
+        sums each band exactly once)."""
+        bc = getattr(self.ctx, "band_comm", None)
+        if bc is None or not bc.active:
+            return sel
+        return sel[bc.rank::bc.size]
 
     def mixer_init(self, cfg_mixer):
         """Register mixed quantities (density.cpp:1834; inner products per

@@ -110,6 +110,8 @@ def diagonalize(ctx, h0: Hamiltonian0, kset: KPointSet, itsol_tol: float,
     itso = ctx.cfg.iterative_solver
     empy_tol = max(itsol_tol * itso.tolerance_ratio, itso.empty_states_tolerance)
     fp32 = wf_dtype == torch.complex64
+    bc = getattr(ctx, "band_comm", None)
+    band_par = bc is not None and bc.active
     all_conv = True
     for kp in kset:
         hk = h0(kp)
@@ -122,8 +124,26 @@ def diagonalize(ctx, h0: Hamiltonian0, kset: KPointSet, itsol_tol: float,
             if fp32:
                 h_diag = h_diag.to(torch.float32)
                 psi0 = psi0.to(torch.complex64)
+            if band_par:
+                # band-parallel H/S application: each band-group rank
+                # applies its slice of the trial block, results are
+                # allgathered (replicated storage, split
+                # compute — the FFT-bound apply dominates)
+                def apply_fn(p, s=ispn):
+                    n = p.shape[0]
+                    cnt = [n // bc.size + (1 if r < n % bc.size else 0)
+                           for r in range(bc.size)]
+                    o = int(np.cumsum([0] + cnt)[bc.rank])
+                    h_loc, s_loc = hk.apply_h_s(p[o:o + cnt[bc.rank]], s)
+                    h = bc.allgather_rows(h_loc, cnt)
+                    sg = bc.allgather_rows(s_loc, cnt) \
+                        if s_loc is not None else None
+                    return h, sg
+            else:
+                def apply_fn(p, s=ispn):
+                    return hk.apply_h_s(p, s)
             res = davidson(
-                lambda p, s=ispn: hk.apply_h_s(p, s),
+                apply_fn,
                 psi0, h_diag, o_diag, occ=kp.occ[ispn],
                 tol_occ=itsol_tol, tol_empty=empy_tol,
                 num_steps=itso.num_steps, subspace_size=itso.subspace_size,

@@ -111,12 +111,16 @@ class KPointSet:
         self.weights = np.asarray(weights if weights is not None
                                   else np.full(len(self.vk), 1.0 / len(self.vk)))
         self.num_kpoints = len(self.vk)
-        # chunk split over ranks (reference splindex_chunk, k_point_set.hpp:44)
-        counts = [self.num_kpoints // self.comm.size +
-                  (1 if r < self.num_kpoints % self.comm.size else 0)
-                  for r in range(self.comm.size)]
+        # chunk split over k-GROUPS (reference splindex_chunk,
+        # k_point_set.hpp:44); with band parallelism every rank of a
+        # band group holds the same k-points (ctx.kcolor/num_kgroups)
+        ngrp = getattr(ctx, "num_kgroups", self.comm.size if self.comm.active else 1)
+        color = getattr(ctx, "kcolor", self.comm.rank if self.comm.active else 0)
+        counts = [self.num_kpoints // ngrp +
+                  (1 if r < self.num_kpoints % ngrp else 0)
+                  for r in range(ngrp)]
         offs = np.cumsum([0] + counts)
-        self.local_range = (int(offs[self.comm.rank]), int(offs[self.comm.rank + 1]))
+        self.local_range = (int(offs[color]), int(offs[color + 1]))
         self.kpoints = [KPoint(ctx, self.vk[ik], self.weights[ik])
                         for ik in range(*self.local_range)]
         self.energy_fermi = 0.0
@@ -132,13 +136,17 @@ class KPointSet:
             self._all_occ = np.array([kp.occ for kp in self.kpoints])
             self._all_w = self.weights.copy()
             return
-        local = [(kp.eigvals, kp.occ) for kp in self.kpoints]
+        # gather (k_index, eig, occ); with band groups the same k is
+        # held by several ranks — dedup by global k index
+        local = [(self.local_range[0] + i, kp.eigvals, kp.occ)
+                 for i, kp in enumerate(self.kpoints)]
         gathered = self.comm.allgather_object(local)
-        eig, occ = [], []
+        by_k = {}
         for part in gathered:
-            for e, o in part:
-                eig.append(e)
-                occ.append(o)
+            for ik, e, o in part:
+                by_k[ik] = (e, o)
+        eig = [by_k[ik][0] for ik in range(self.num_kpoints)]
+        occ = [by_k[ik][1] for ik in range(self.num_kpoints)]
         self._all_eig = np.array(eig)
         self._all_occ = np.array(occ)
         self._all_w = self.weights.copy()

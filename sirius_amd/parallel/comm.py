@@ -51,6 +51,41 @@ class Comm:
         if self.active:
             dist.barrier(group=self.group)
 
+    def allgather_rows(self, t: torch.Tensor, counts: list[int]) -> torch.Tensor:
+        """Concatenate row-blocks from all ranks: local t [n_r, m] ->
+        [sum(counts), m].  Pads to max(counts) for gloo/nccl equal-shape
+        all_gather, then trims."""
+        if not self.active:
+            return t
+        nmax = max(counts)
+        m = t.shape[1] if t.dim() > 1 else 0
+        pad = torch.zeros(nmax, m, dtype=t.dtype, device=t.device)
+        pad[:t.shape[0]] = t
+        outs = [torch.empty_like(pad) for _ in range(self.size)]
+        dist.all_gather(outs, pad.contiguous(), group=self.group)
+        return torch.cat([outs[r][:counts[r]] for r in range(self.size)], dim=0)
+
+
+def make_band_comm(npb: int):
+    """Split the world into k-groups of `npb` band ranks each
+    (reference Simulation_context::init_comm, simulation_context.cpp:1301:
+    world -> comm_k x comm_band).  Rank r: band_rank = r % npb,
+    k_color = r // npb.  Returns (band_comm, k_color, num_kgroups)."""
+    world = get_comm()
+    if not world.active or npb <= 1:
+        return Comm(), 0, world.size if world.active else 1
+    if world.size % npb:
+        raise ValueError(f"world size {world.size} not divisible by "
+                         f"band-group size {npb}")
+    nk_groups = world.size // npb
+    my_group = None
+    for color in range(nk_groups):
+        ranks = list(range(color * npb, (color + 1) * npb))
+        g = dist.new_group(ranks=ranks)
+        if world.rank in ranks:
+            my_group = g
+    return Comm(my_group), world.rank // npb, nk_groups
+
 
 _comm: Optional[Comm] = None
 
