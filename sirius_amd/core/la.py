@@ -94,6 +94,40 @@ def ortho_factor(gram: torch.Tensor) -> torch.Tensor:
 _ZGRAM_MAX_MN = 1536  # above this rocBLAS catches up (36.6 vs 36.8 TF/s @ 1049)
 
 
+def inner_gamma(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    """Real Gram matrix at the Γ point: with the reality constraint
+    c(-G) = c*(G), every inner product Σ_G conj(a)b is real and equals
+    the dot product of the REAL views — a single dgemm at half the
+    complex flops (reference Γ-trick, wave_functions.hpp:1589-1696;
+    there the PW set is also halved — here storage stays full-sphere,
+    the algebra runs real)."""
+    ar = torch.view_as_real(a).reshape(a.shape[0], -1)
+    br = torch.view_as_real(b).reshape(b.shape[0], -1)
+    return ar @ br.T
+
+
+def transform_gamma(T: torch.Tensor, X: torch.Tensor,
+                    out: torch.Tensor | None = None, alpha: float = 1.0,
+                    accumulate: bool = False) -> torch.Tensor:
+    """Y (+)= α Tᵀ X with REAL T [K, M] on complex X [K, G] via the real
+    view (a single dgemm at half the complex flops) — the Γ-trick
+    counterpart of `transform`."""
+    Xr = torch.view_as_real(X).reshape(X.shape[0], -1)
+    Tt = T.transpose(0, 1)
+    if out is not None:
+        outr = torch.view_as_real(out).reshape(out.shape[0], -1)
+        if accumulate:
+            outr.addmm_(Tt, Xr, alpha=alpha)
+        else:
+            torch.mm(Tt, Xr, out=outr)
+            if alpha != 1.0:
+                outr *= alpha
+        return out
+    Yr = alpha * (Tt @ Xr)
+    return torch.view_as_complex(
+        Yr.reshape(T.shape[1], X.shape[1], 2).contiguous())
+
+
 def inner(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
     """Gram block ⟨a_i|b_j⟩ = Σ_g conj(a[i,g]) b[j,g].
 

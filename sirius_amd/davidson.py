@@ -39,7 +39,8 @@ def _inner(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
     return la.inner(a, b)
 
 
-def _ortho_joint(new, hnew, snew, phi, hphi, sphi, outs=None):
+def _ortho_joint(new, hnew, snew, phi, hphi, sphi, outs=None,
+                 gamma: bool = False):
     """Project the existing S-orthonormal subspace out of `new` and
     S-orthonormalize it, applying identical transforms to hnew/snew
     (mirrors wf::orthogonalize, wave_functions.hpp:1781-2051).
@@ -48,20 +49,26 @@ def _ortho_joint(new, hnew, snew, phi, hphi, sphi, outs=None):
     rows when rank-deficient. With `outs` = (phi_buf, h_buf, s_buf) row
     slices, the final orthonormalizing transform writes DIRECTLY into the
     caller's subspace buffers (no extra device copy).
+
+    gamma=True runs the Γ-trick real algebra (Gram matrices real by the
+    c(-G)=c*(G) constraint; dgemm transforms at half the complex cost,
+    wave_functions.hpp:1589-1696).
     """
+    inner_f = la.inner_gamma if gamma else _inner
+    trans_f = la.transform_gamma if gamma else la.transform
     s_of_new = snew if snew is not None else new
     if phi is not None and phi.shape[0]:
-        ov = _inner(sphi if sphi is not None else phi, new)   # [N, n]
-        new = la.transform(ov, phi, out=new.contiguous(), alpha=-1.0,
-                           accumulate=True)
+        ov = inner_f(sphi if sphi is not None else phi, new)   # [N, n]
+        new = trans_f(ov, phi, out=new.contiguous(), alpha=-1.0,
+                      accumulate=True)
         if hnew is not None:
-            hnew = la.transform(ov, hphi, out=hnew.contiguous(), alpha=-1.0,
-                                accumulate=True)
+            hnew = trans_f(ov, hphi, out=hnew.contiguous(), alpha=-1.0,
+                           accumulate=True)
         if snew is not None:
-            snew = la.transform(ov, sphi, out=snew.contiguous(), alpha=-1.0,
-                                accumulate=True)
+            snew = trans_f(ov, sphi, out=snew.contiguous(), alpha=-1.0,
+                           accumulate=True)
         s_of_new = snew if snew is not None else new
-    gram = _inner(new, s_of_new)
+    gram = inner_f(new, s_of_new)
     gram = 0.5 * (gram + gram.conj().T)
     n = gram.shape[0]
     try:
@@ -74,15 +81,15 @@ def _ortho_joint(new, hnew, snew, phi, hphi, sphi, outs=None):
     nkeep = t.shape[0]
     if outs is not None:
         o_phi, o_h, o_s = outs
-        new = la.transform(tT, new, out=o_phi[:nkeep])
-        hnew = la.transform(tT, hnew, out=o_h[:nkeep]) \
+        new = trans_f(tT, new, out=o_phi[:nkeep])
+        hnew = trans_f(tT, hnew, out=o_h[:nkeep]) \
             if hnew is not None else None
-        snew = la.transform(tT, snew, out=o_s[:nkeep]) \
+        snew = trans_f(tT, snew, out=o_s[:nkeep]) \
             if snew is not None else None
         return new, hnew, snew
-    new = la.transform(tT, new)
-    hnew = la.transform(tT, hnew) if hnew is not None else None
-    snew = la.transform(tT, snew) if snew is not None else None
+    new = trans_f(tT, new)
+    hnew = trans_f(tT, hnew) if hnew is not None else None
+    snew = trans_f(tT, snew) if snew is not None else None
     return new, hnew, snew
 
 
@@ -92,7 +99,8 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
              num_steps: int = 20, subspace_size: int = 2,
              min_occupancy: float = 1e-14,
              extra_ortho: bool = False, locking: bool = True,
-             early_restart: float = 0.5) -> DavidsonResult:
+             early_restart: float = 0.5, gamma: bool = False,
+             gamma_neg: torch.Tensor | None = None) -> DavidsonResult:
     """Solve for the `nb` lowest eigenpairs of H ψ = ε S ψ (S=I or USPP S).
 
     apply_h_s(phi [n, nG]) -> (hphi, sphi|None).
@@ -119,9 +127,23 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
     hphi_buf = torch.empty(num_phi_max, nG, dtype=cdt, device=dev)
     sphi_buf = None
 
+    inner_f = la.inner_gamma if gamma else _inner
+    trans_f = la.transform_gamma if gamma else la.transform
+
+    def enforce_gamma(t):
+        # re-impose c(-G) = c*(G): the real-algebra Γ path assumes it and
+        # roundoff drift otherwise grows unchecked over many steps (the
+        # reference enforces it structurally by storing half the G set,
+        # wave_functions.hpp:1589)
+        if gamma and gamma_neg is not None and t is not None:
+            t.add_(t[:, gamma_neg].conj()).mul_(0.5)
+        return t
+
     phi = psi0.clone()
+    enforce_gamma(phi)
     hphi, sphi = apply_h_s(phi)
-    phi, hphi, sphi = _ortho_joint(phi, hphi, sphi, None, None, None)
+    phi, hphi, sphi = _ortho_joint(phi, hphi, sphi, None, None, None,
+                                   gamma=gamma)
     n0 = phi.shape[0]
     phi_buf[:n0] = phi
     hphi_buf[:n0] = hphi
@@ -131,7 +153,7 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
         sphi_buf = torch.empty(num_phi_max, nG, dtype=cdt, device=dev)
         sphi_buf[:n0] = sphi
         sphi = sphi_buf[:n0]
-    H = _inner(phi, hphi)
+    H = inner_f(phi, hphi)
     H = 0.5 * (H + H.conj().T)
     evals, Z = la.eigh(H)
     evp_work = (phi.shape[0] / nb) ** 3
@@ -158,8 +180,8 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
         idx = torch.from_numpy(unconv - nlock).to(psi0.device)
         Zs = Z[:, idx]                                     # [N, n]
         e = evals[idx].real
-        hpsi = la.transform(Zs, hphi)                       # [n, nG]
-        spsi = la.transform(Zs, sphi if sphi is not None else phi)
+        hpsi = trans_f(Zs, hphi)                            # [n, nG]
+        spsi = trans_f(Zs, sphi if sphi is not None else phi)
         if hpsi.is_cuda and hpsi.dtype == torch.complex128:
             from . import ops
 
@@ -183,6 +205,8 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
             break
         res = res[keep]
         res = res / torch.linalg.vector_norm(res, dim=1, keepdim=True).to(res.dtype)
+        res = res.contiguous()
+        enforce_gamma(res)
 
         n_new = res.shape[0]
         # leading consecutively-converged active bands are lockable
@@ -193,9 +217,9 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
         if should_restart:
             # restart: collapse the active subspace to its Ritz vectors
             Znb = Z[:, :nb_act]
-            psi = la.transform(Znb, phi)
-            hpsi_f = la.transform(Znb, hphi)
-            spsi_f = la.transform(Znb, sphi) if sphi is not None else None
+            psi = trans_f(Znb, phi)
+            hpsi_f = trans_f(Znb, hphi)
+            spsi_f = trans_f(Znb, sphi) if sphi is not None else None
             phi_buf[nlock:nlock + nb_act] = psi
             hphi_buf[nlock:nlock + nb_act] = hpsi_f
             if spsi_f is not None:
@@ -212,7 +236,7 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
             hphi = hphi_buf[nlock:nlock + nb_act]
             if sphi is not None:
                 sphi = sphi_buf[nlock:nlock + nb_act]
-            H = torch.diag(ev_act.to(phi_buf.dtype))
+            H = torch.diag(ev_act.to(torch.float64 if gamma else phi_buf.dtype))
             evals = ev_act
             Z = torch.eye(nb_act, dtype=H.dtype, device=H.device)
             N = nb_act
@@ -232,9 +256,9 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
         full_s = sphi_buf[:lo] if sphi is not None else None
         if extra_ortho:
             res, hnew, snew = _ortho_joint(res, hnew, snew, full_phi,
-                                           full_h, full_s)
+                                           full_h, full_s, gamma=gamma)
         res, hnew, snew = _ortho_joint(res, hnew, snew, full_phi, full_h,
-                                       full_s, outs=outs)
+                                       full_s, outs=outs, gamma=gamma)
         if res.shape[0] == 0:
             converged = True
             break
@@ -247,7 +271,7 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
         Nn = phi.shape[0]
         Hn = torch.empty(Nn, Nn, dtype=H.dtype, device=H.device)
         Hn[:N, :N] = H
-        blk = _inner(phi, hnew)                            # [Nn, nnew]
+        blk = inner_f(phi, hnew)                           # [Nn, nnew]
         Hn[:, N:] = blk
         Hn[N:, :N] = blk[:N].conj().T
         H = 0.5 * (Hn + Hn.conj().T)
@@ -255,7 +279,7 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
         evp_work += (Nn / nb) ** 3
 
     nb_act = nb - nlock
-    psi_act = la.transform(Z[:, :nb_act], phi)
+    psi_act = trans_f(Z[:, :nb_act], phi)
     if nlock:
         psi = torch.cat([phi_buf[:nlock], psi_act], dim=0)
         ev_out = np.concatenate([eval_locked,

@@ -69,9 +69,18 @@ def initialize_subspace(ctx, kp, hk):
     n_ao = phi.shape[0] * ctx.num_spinors
     if n_ao < nb:
         gen = torch.Generator(device="cpu").manual_seed(12345 + kp.gkvec.num_gvec)
-        rnd = torch.randn((nb - n_ao + 1) // ctx.num_spinors, kp.num_gkvec, 2,
-                          generator=gen, dtype=torch.float64)
-        rnd = torch.view_as_complex(rnd.contiguous()).to(ctx.device)
+        nrnd = (nb - n_ao + 1) // ctx.num_spinors
+        if bool(ctx.cfg.parameters.gamma_point) \
+                and float(np.abs(kp.k_frac).max()) < 1e-12:
+            # Γ-trick: random REAL-space fields give exactly
+            # conjugate-symmetric c(-G) = c*(G) trial vectors
+            rr = torch.randn(nrnd, *kp.fft.dims, generator=gen,
+                             dtype=torch.float64).to(ctx.device)
+            rnd = kp.fft.to_pw(rr.to(ctx.dtype))
+        else:
+            rnd = torch.randn(nrnd, kp.num_gkvec, 2,
+                              generator=gen, dtype=torch.float64)
+            rnd = torch.view_as_complex(rnd.contiguous()).to(ctx.device)
         # damp high-G components for smoother start
         damp = 1.0 / (1.0 + kp.gkvec.gk2_t)
         rnd = rnd * damp
@@ -99,6 +108,18 @@ def initialize_subspace(ctx, kp, hk):
         kp.eigvals[ispn] = evals[:nb].real.cpu().numpy()
 
 
+def _gamma_neg_index(kp):
+    """Index of -G for every G of the k=0 sphere (cached on the kp)."""
+    idx = getattr(kp, "_gamma_neg", None)
+    if idx is None:
+        key = {tuple(m): i for i, m in enumerate(kp.gkvec.miller)}
+        neg = np.array([key[tuple(-m)] for m in kp.gkvec.miller],
+                       dtype=np.int64)
+        idx = torch.from_numpy(neg).to(kp.psi.device)
+        kp._gamma_neg = idx
+    return idx
+
+
 def diagonalize(ctx, h0: Hamiltonian0, kset: KPointSet, itsol_tol: float,
                 wf_dtype=None) -> bool:
     """Davidson for all local k-points/spins (reference diagonalize.hpp).
@@ -110,6 +131,8 @@ def diagonalize(ctx, h0: Hamiltonian0, kset: KPointSet, itsol_tol: float,
     itso = ctx.cfg.iterative_solver
     empy_tol = max(itsol_tol * itso.tolerance_ratio, itso.empty_states_tolerance)
     fp32 = wf_dtype == torch.complex64
+    use_gamma = bool(ctx.cfg.parameters.gamma_point) \
+        and not ctx.nc_magnetism and ctx.hubbard is None
     bc = getattr(ctx, "band_comm", None)
     band_par = bc is not None and bc.active
     all_conv = True
@@ -149,7 +172,11 @@ def diagonalize(ctx, h0: Hamiltonian0, kset: KPointSet, itsol_tol: float,
                 num_steps=itso.num_steps, subspace_size=itso.subspace_size,
                 min_occupancy=itso.min_occupancy, extra_ortho=itso.extra_ortho,
                 locking=bool(itso.locking),
-                early_restart=float(itso.early_restart))
+                early_restart=float(itso.early_restart),
+                gamma=use_gamma and float(np.abs(kp.k_frac).max()) < 1e-12,
+                gamma_neg=_gamma_neg_index(kp)
+                if use_gamma and float(np.abs(kp.k_frac).max()) < 1e-12
+                else None)
             kp.psi[ispn] = res.psi.to(torch.complex128) if fp32 else res.psi
             kp.eigvals[ispn] = res.eval
             ctx.counters["num_itsol_steps"] += res.niter

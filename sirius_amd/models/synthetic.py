@@ -218,6 +218,10 @@ def make_named_context(model: str, device: str | None = None, **overrides):
         kw = dict(gk_cutoff=5.0, pw_cutoff=14.0, ngridk=ngridk)
         kw.update(overrides)
         cfg, _ = make_synthetic_config(natoms=natoms, **kw)
+        if model == "si512":
+            # BASELINE config 3 is Γ-only: enable the Γ-trick real algebra
+            cfg._data["parameters"]["gamma_point"] = True
+            cfg.parameters.gamma_point = True
         uc = make_synthetic_cell(natoms)
         return SimulationContext(cfg, unit_cell=uc, device=device)
     if model == "sto-uspp":
