@@ -120,6 +120,56 @@ def _gamma_neg_index(kp):
     return idx
 
 
+def diagonalize_exact(ctx, h0: Hamiltonian0, kset: KPointSet) -> bool:
+    """Exact dense diagonalization (iterative_solver.type = "exact";
+    reference diagonalize_pp_exact, diagonalize_pp.hpp:24): build the
+    full PW Hamiltonian/overlap H(G,G') = ½|G+k|²δ + V_eff(G−G') +
+    Σ β D β†, S = I + β Q β† and solve the generalized problem.  A
+    correctness tool for small basis sets."""
+    for kp in kset:
+        hk = h0(kp)
+        ngk = kp.num_gkvec
+        if ngk > 6000:
+            raise RuntimeError(f"exact solver: basis too large ({ngk})")
+        if ctx.nc_magnetism:
+            raise NotImplementedError("exact solver for spinors TODO")
+        # V_eff(G-G') from the coarse sphere
+        veff_pw = {}
+        m_c = ctx.gvec_coarse.miller
+        lutd = {tuple(mm): i for i, mm in enumerate(m_c)}
+        mgk = kp.gkvec.miller
+        d = mgk[:, None, :] - mgk[None, :, :]
+        idx = np.empty((ngk, ngk), dtype=np.int64)
+        ok = np.ones((ngk, ngk), dtype=bool)
+        for i in range(ngk):
+            for j in range(ngk):
+                t = lutd.get(tuple(d[i, j]))
+                if t is None:
+                    ok[i, j] = False
+                    idx[i, j] = 0
+                else:
+                    idx[i, j] = t
+        for ispn in range(ctx.num_spin_steps):
+            vc = ctx.fft_coarse.to_pw(
+                hk.h0.veff_r_coarse[ispn].to(ctx.dtype)).cpu().numpy()
+            V = vc[idx] * ok
+            H = torch.from_numpy(V).to(ctx.device)
+            H += torch.diag(hk.ekin.to(H.dtype))
+            S = torch.eye(ngk, dtype=H.dtype, device=ctx.device)
+            if hk.bp.num_beta_total:
+                B = hk.bp.beta_t                    # [nbf, ngk]
+                H += B.T @ hk.D[ispn].to(H.dtype) @ B.conj()
+                if hk.Q is not None:
+                    S += B.T @ hk.Q.to(H.dtype) @ B.conj()
+            from scipy.linalg import eigh as seigh
+            w, v = seigh(H.cpu().numpy(), S.cpu().numpy(),
+                         subset_by_index=[0, ctx.num_bands - 1])
+            kp.eigvals[ispn] = w
+            kp.psi[ispn] = torch.from_numpy(
+                np.ascontiguousarray(v.T)).to(ctx.device).to(ctx.dtype)
+    return True
+
+
 def diagonalize(ctx, h0: Hamiltonian0, kset: KPointSet, itsol_tol: float,
                 wf_dtype=None) -> bool:
     """Davidson for all local k-points/spins (reference diagonalize.hpp).
@@ -129,6 +179,8 @@ def diagonalize(ctx, h0: Hamiltonian0, kset: KPointSet, itsol_tol: float,
     vectors, H/S application and the subspace algebra all in single
     precision; converged psi is stored back in fp64."""
     itso = ctx.cfg.iterative_solver
+    if str(itso.type) == "exact":
+        return diagonalize_exact(ctx, h0, kset)
     empy_tol = max(itsol_tol * itso.tolerance_ratio, itso.empty_states_tolerance)
     fp32 = wf_dtype == torch.complex64
     use_gamma = bool(ctx.cfg.parameters.gamma_point) \

@@ -44,6 +44,19 @@ def test_si2_model():
     assert ctx.unit_cell.num_atoms == 2
 
 
+def test_exact_solver_matches_davidson():
+    """iterative_solver.type=exact (reference diagonalize_pp_exact,
+    diagonalize_pp.hpp:24) agrees with the Davidson path."""
+    from sirius_amd.models.synthetic import make_named_context
+    ctx = make_named_context("si2", gk_cutoff=4.0, pw_cutoff=10.0)
+    r1 = DFTGroundState(KPointSet(ctx)).initial_state().find(num_dft_iter=15)
+    ctx2 = make_named_context("si2", gk_cutoff=4.0, pw_cutoff=10.0)
+    ctx2.cfg._data["iterative_solver"]["type"] = "exact"
+    ctx2.cfg.iterative_solver.type = "exact"
+    r2 = DFTGroundState(KPointSet(ctx2)).initial_state().find(num_dft_iter=15)
+    assert abs(r1["energy"]["total"] - r2["energy"]["total"]) < 1e-8
+
+
 def test_fp32_wf_promotion():
     """fp32 wave functions with runtime fp64 promotion converge to the
     fp64 energy (reference precision_wf, dft_ground_state.cpp:269-304)."""
