@@ -207,13 +207,15 @@ class Hamiltonian0:
                     dt1 = None
                 nbf = aug.nbf
                 paw = ctx.paw
+                # packed-triangle -> full unpack, vectorized (was an
+                # O(nbf^2) python loop per atom per iteration)
+                _x1, _x2 = np.triu_indices(nbf)
+                _p12 = _x2 * (_x2 + 1) // 2 + _x1
                 for i, ia in enumerate(ia_list):
                     def unpack(dt):
                         m = np.zeros((nbf, nbf))
-                        for xi2 in range(nbf):
-                            for xi1 in range(xi2 + 1):
-                                idx12 = xi2 * (xi2 + 1) // 2 + xi1
-                                m[xi1, xi2] = m[xi2, xi1] = dt[idx12, i]
+                        m[_x1, _x2] = dt[_p12, i]
+                        m[_x2, _x1] = dt[_p12, i]
                         return m
 
                     d0 = unpack(dt0)
