@@ -106,14 +106,29 @@ class SphericalFFT:
         g = torch.fft.fftn(fr, dim=(-3, -2, -1), norm="forward")
         return self._unpack(g.reshape(*fr.shape[:-3], self.size))
 
+    # keep the dense FFT working set bounded (large Γ-only cells: a
+    # 1000-band block on a 144³ grid would be ~50 GB of grids at once)
+    MAX_GRID_BYTES = 4 << 30
+
+    def _band_chunk(self, nb: int, itemsize: int = 16) -> int:
+        per_band = self.size * itemsize
+        return max(1, min(nb, int(self.MAX_GRID_BYTES // max(per_band, 1))))
+
     def apply_veff_kinetic(self, psi: torch.Tensor, veff_r: torch.Tensor,
                            gk2: torch.Tensor) -> torch.Tensor:
         """Fused local-operator apply: FFT⁻¹ψ → ×V(r) → FFT → +½|G+k|²ψ.
 
         psi [nb, nG], veff_r [n1,n2,n3] real, gk2 [nG] real.
-        Returns hpsi_local [nb, nG].
+        Returns hpsi_local [nb, nG].  Band-chunked to bound grid memory.
         """
         nb = psi.shape[0]
+        chunk = self._band_chunk(nb)
+        if chunk < nb:
+            out = torch.empty_like(psi)
+            for i in range(0, nb, chunk):
+                out[i:i + chunk] = self.apply_veff_kinetic(
+                    psi[i:i + chunk].contiguous(), veff_r, gk2)
+            return out
         ext = _ext_for(psi)
         grid = self._pack(psi).reshape(nb, *self.dims)
         psi_r = torch.fft.ifftn(grid, dim=(-3, -2, -1), norm="forward")
@@ -134,6 +149,13 @@ class SphericalFFT:
     def density_accumulate(self, psi: torch.Tensor, weights: torch.Tensor,
                            rho_r: torch.Tensor):
         """rho_r += Σ_b w_b |FFT⁻¹ψ_b(r)|² (in place; rho_r real [n1,n2,n3])."""
+        nb = psi.shape[0]
+        chunk = self._band_chunk(nb)
+        if chunk < nb:
+            for i in range(0, nb, chunk):
+                self.density_accumulate(psi[i:i + chunk].contiguous(),
+                                        weights[i:i + chunk], rho_r)
+            return
         psi_r = self.to_real(psi)
         ext = _ext_for(psi)
         if ext is not None:

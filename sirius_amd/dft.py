@@ -74,9 +74,14 @@ def initialize_subspace(ctx, kp, hk):
                 and float(np.abs(kp.k_frac).max()) < 1e-12:
             # Γ-trick: random REAL-space fields give exactly
             # conjugate-symmetric c(-G) = c*(G) trial vectors
-            rr = torch.randn(nrnd, *kp.fft.dims, generator=gen,
-                             dtype=torch.float64).to(ctx.device)
-            rnd = kp.fft.to_pw(rr.to(ctx.dtype))
+            chunks = []
+            step = max(1, kp.fft._band_chunk(nrnd))
+            for i in range(0, nrnd, step):
+                nn_ = min(step, nrnd - i)
+                rr = torch.randn(nn_, *kp.fft.dims, generator=gen,
+                                 dtype=torch.float64).to(ctx.device)
+                chunks.append(kp.fft.to_pw(rr.to(ctx.dtype)))
+            rnd = torch.cat(chunks, dim=0)
         else:
             rnd = torch.randn(nrnd, kp.num_gkvec, 2,
                               generator=gen, dtype=torch.float64)
