@@ -41,6 +41,8 @@ def _inner(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
 
 def _ortho_joint(new, hnew, snew, phi, hphi, sphi, outs=None,
                  gamma: bool = False):
+    if gamma and new.is_cuda:
+        gamma = False      # GPU: keep the MFMA complex path (see davidson)
     """Project the existing S-orthonormal subspace out of `new` and
     S-orthonormalize it, applying identical transforms to hnew/snew
     (mirrors wf::orthogonalize, wave_functions.hpp:1781-2051).
@@ -127,8 +129,26 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
     hphi_buf = torch.empty(num_phi_max, nG, dtype=cdt, device=dev)
     sphi_buf = None
 
-    inner_f = la.inner_gamma if gamma else _inner
-    trans_f = la.transform_gamma if gamma else la.transform
+    # Γ-trick split: the real-GEMM algebra wins on CPU (dgemm at half
+    # the complex flops); on GPU the complex Grams/transforms run on the
+    # hand-written MFMA fp64 kernels which beat rocBLAS dgemm at these
+    # shapes — there we keep complex GEMMs but still use the REAL
+    # subspace eigensolves (H is real at Γ) and the symmetry enforcement.
+    gamma_alg = gamma and not psi0.is_cuda
+    inner_c = la.inner_gamma if gamma_alg else _inner
+    trans_f = la.transform_gamma if gamma_alg else la.transform
+
+    def inner_f(a, b):
+        r = inner_c(a, b)
+        if gamma and not gamma_alg:
+            r = r.real
+        return r
+
+    def to_T(Z):
+        # transform coefficient dtype expected by trans_f
+        if gamma_alg:
+            return Z.real if Z.is_complex() else Z
+        return Z if Z.is_complex() else Z.to(psi0.dtype)
 
     def enforce_gamma(t):
         # re-impose c(-G) = c*(G): the real-algebra Γ path assumes it and
@@ -180,8 +200,8 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
         idx = torch.from_numpy(unconv - nlock).to(psi0.device)
         Zs = Z[:, idx]                                     # [N, n]
         e = evals[idx].real
-        hpsi = trans_f(Zs, hphi)                            # [n, nG]
-        spsi = trans_f(Zs, sphi if sphi is not None else phi)
+        hpsi = trans_f(to_T(Zs), hphi)                      # [n, nG]
+        spsi = trans_f(to_T(Zs), sphi if sphi is not None else phi)
         if hpsi.is_cuda and hpsi.dtype == torch.complex128:
             from . import ops
 
@@ -216,7 +236,7 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
              and len(unconv) < early_restart * lockable)
         if should_restart:
             # restart: collapse the active subspace to its Ritz vectors
-            Znb = Z[:, :nb_act]
+            Znb = to_T(Z[:, :nb_act])
             psi = trans_f(Znb, phi)
             hpsi_f = trans_f(Znb, hphi)
             spsi_f = trans_f(Znb, sphi) if sphi is not None else None
@@ -279,7 +299,7 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
         evp_work += (Nn / nb) ** 3
 
     nb_act = nb - nlock
-    psi_act = trans_f(Z[:, :nb_act], phi)
+    psi_act = trans_f(to_T(Z[:, :nb_act]), phi)
     if nlock:
         psi = torch.cat([phi_buf[:nlock], psi_act], dim=0)
         ev_out = np.concatenate([eval_locked,
