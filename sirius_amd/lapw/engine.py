@@ -1262,7 +1262,7 @@ class FPGroundState:
                 src = int(op.perm[ia])
                 T = Ts[(iop, lab)]
                 for ispn in range(dm[ia].shape[2]):
-                    out[ia][:, :, ispn] += T.conj().T @ dm[src][:, :, ispn] @ T
+                    out[ia][:, :, ispn] += T @ dm[src][:, :, ispn] @ T.conj().T
         for ia in range(len(out)):
             out[ia] /= len(ops)
         return out
