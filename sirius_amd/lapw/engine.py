@@ -163,6 +163,12 @@ def make_lapw_context(cfg: Config, base_dir: str = ".", device=None):
         cfg._data["parameters"]["gk_cutoff"] = float(p.aw_cutoff) / min_rmt
         cfg.parameters.gk_cutoff = float(p.aw_cutoff) / min_rmt
 
+    if device is None:
+        # the FP engine orchestrates on host (radial machinery, dense
+        # set_fv_h_o); keep tensors on CPU unless a device is explicitly
+        # requested — moving the FV solve/FFTs to the GPU is the round-3
+        # item (NEXT.md)
+        device = "cpu"
     ctx = SimulationContext(cfg, unit_cell=uc, device=device)
     ctx.full_potential = True
     ctx.lmax_apw = int(p.lmax_apw)
