@@ -29,6 +29,13 @@ _DEFAULTS: dict[str, dict[str, Any]] = {
         "mpi_grid_dims": [1, 1],
         "rmt_max": 2.2,
         "verification": 0,
+        # MI355X extension (no reference counterpart): worker threads for
+        # the per-k-point Davidson loop, one HIP stream per worker.
+        # MEASURED on sto-uspp (64 k, 1 GPU): 1 thr 0.50 s/iter, 2 thr
+        # 0.62, 4 thr 1.06 — the loop is GIL/launch-bound, so threads only
+        # add contention.  0 = auto = serial; the knob stays for
+        # experiments (results are bit-identical, verified on GPU).
+        "num_kpoint_threads": 0,
     },
     "parameters": {
         "electronic_structure_method": "pseudopotential",

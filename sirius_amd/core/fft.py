@@ -62,6 +62,11 @@ class SphericalFFT:
         buf = self._pack_buf
         if (buf is None or buf.shape[0] < nb or buf.device != coeffs.device
                 or buf.dtype != coeffs.dtype):
+            if buf is not None and buf.is_cuda:
+                # threaded k-loop: the retiring buffer may belong to a
+                # different stream's allocator pool — keep it alive until
+                # this stream's enqueued reads complete
+                buf.record_stream(torch.cuda.current_stream())
             buf = torch.zeros(max(nb, 1), self.size, dtype=coeffs.dtype,
                               device=coeffs.device)
             self._pack_buf = buf

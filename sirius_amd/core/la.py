@@ -23,12 +23,19 @@ EIGH_GPU_MIN = 1024  # below this, host LAPACK beats rocSOLVER's launch storm
 _HOST_SOLVE_THREADS = min(16, torch.get_num_threads())
 
 
+PIN_HOST_THREADS = False  # threaded k-loop: caller pinned the pool already
+
+
 class _host_threads:
     def __enter__(self):
+        if PIN_HOST_THREADS:
+            return
         self.saved = torch.get_num_threads()
         torch.set_num_threads(_HOST_SOLVE_THREADS)
 
     def __exit__(self, *a):
+        if PIN_HOST_THREADS:
+            return
         torch.set_num_threads(self.saved)
 
 

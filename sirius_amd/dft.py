@@ -175,6 +175,64 @@ def diagonalize_exact(ctx, h0: Hamiltonian0, kset: KPointSet) -> bool:
     return True
 
 
+def _solve_kps_threaded(ctx, kps, solve_fn, nthreads):
+    """Run the independent per-k-point Davidson solves on a small thread
+    pool, one HIP stream per worker.  Each k-point is handled by exactly
+    one worker; per-kp state (kp.fft staging buffer, kp.psi) is only ever
+    touched by that worker within the region, and every stream is
+    host-synchronized before returning, so downstream serial code
+    (density generate) sees completed results.
+
+    Kept as an opt-in experiment (control.num_kpoint_threads > 1):
+    measured on sto-uspp (64 k, 1 MI355X) the per-k loop is GIL-bound —
+    thousands of small launches per SCF iteration — and 2/4 workers run
+    1.2x/2.1x SLOWER than serial while producing bit-identical energies.
+    The GPU-side fix for this regime is batching k-points into single
+    tensor ops, not host threads (NEXT.md)."""
+    import itertools
+    from concurrent.futures import ThreadPoolExecutor
+
+    streams = getattr(ctx, "_kp_streams", None)
+    if streams is None or len(streams) < nthreads:
+        streams = [torch.cuda.Stream(device=ctx.device)
+                   for _ in range(nthreads)]
+        ctx._kp_streams = streams
+    nxt = itertools.count()          # CPython: __next__ is atomic
+    outs = [None] * len(kps)
+
+    def run_shard(wid):
+        s = streams[wid]
+        s.wait_stream(torch.cuda.default_stream(ctx.device))
+        with torch.cuda.stream(s):
+            while True:
+                i = next(nxt)
+                if i >= len(kps):
+                    break
+                outs[i] = solve_fn(kps[i])
+        s.synchronize()
+
+    import sys
+    from .core import la
+    saved_nt = torch.get_num_threads()
+    saved_si = sys.getswitchinterval()
+    try:
+        # pin the (global) intra-op pool once instead of per-eigh toggling
+        # from 4 threads, and shorten the GIL switch interval: the workers
+        # interleave many short C-extension calls
+        torch.set_num_threads(la._HOST_SOLVE_THREADS)
+        la.PIN_HOST_THREADS = True
+        sys.setswitchinterval(0.0005)
+        with ThreadPoolExecutor(max_workers=nthreads) as ex:
+            futs = [ex.submit(run_shard, w) for w in range(nthreads)]
+            for f in futs:
+                f.result()
+    finally:
+        la.PIN_HOST_THREADS = False
+        sys.setswitchinterval(saved_si)
+        torch.set_num_threads(saved_nt)
+    return outs
+
+
 def diagonalize(ctx, h0: Hamiltonian0, kset: KPointSet, itsol_tol: float,
                 wf_dtype=None) -> bool:
     """Davidson for all local k-points/spins (reference diagonalize.hpp).
@@ -192,12 +250,13 @@ def diagonalize(ctx, h0: Hamiltonian0, kset: KPointSet, itsol_tol: float,
         and not ctx.nc_magnetism and ctx.hubbard is None
     bc = getattr(ctx, "band_comm", None)
     band_par = bc is not None and bc.active
-    all_conv = True
-    for kp in kset:
+
+    def _solve_kp(kp):
         hk = h0(kp)
         o_diag = hk.o_diag()
         if fp32:
             o_diag = o_diag.to(torch.float32)
+        conv, nit, work = True, 0, 0
         for ispn in range(ctx.num_spin_steps):
             h_diag = hk.h_diag(ispn)
             psi0 = kp.psi[ispn]
@@ -236,9 +295,25 @@ def diagonalize(ctx, h0: Hamiltonian0, kset: KPointSet, itsol_tol: float,
                 else None)
             kp.psi[ispn] = res.psi.to(torch.complex128) if fp32 else res.psi
             kp.eigvals[ispn] = res.eval
-            ctx.counters["num_itsol_steps"] += res.niter
-            ctx.counters["band_evp_work_count"] += res.evp_work
-            all_conv = all_conv and res.converged
+            nit += res.niter
+            work += res.evp_work
+            conv = conv and res.converged
+        return conv, nit, work
+
+    kps = list(kset)
+    # auto = serial: measured on sto-uspp/MI355X the loop is GIL-bound and
+    # threads only add contention (see control.num_kpoint_threads)
+    nthreads = max(1, int(ctx.cfg.control.num_kpoint_threads or 0))
+    nthreads = min(nthreads, max(1, len(kps)))
+    if nthreads > 1 and str(ctx.device) != "cpu" and not band_par:
+        outs = _solve_kps_threaded(ctx, kps, _solve_kp, nthreads)
+    else:
+        outs = [_solve_kp(kp) for kp in kps]
+    all_conv = True
+    for conv, nit, work in outs:
+        ctx.counters["num_itsol_steps"] += nit
+        ctx.counters["band_evp_work_count"] += work
+        all_conv = all_conv and conv
     comm = get_comm()
     if comm.active:
         all_conv = bool(comm.allreduce_scalar(float(all_conv)) == comm.size)

@@ -225,7 +225,12 @@ def make_named_context(model: str, device: str | None = None, **overrides):
         uc = make_synthetic_cell(natoms)
         return SimulationContext(cfg, unit_cell=uc, device=device)
     if model == "sto-uspp":
-        kw = dict(gk_cutoff=6.0, pw_cutoff=20.0, ngridk=(4, 4, 4))
+        # use_symmetry=True matches the reference default (SIRIUS reduces
+        # the 4x4x4 MP mesh of the ideal cubic perovskite to 10 IBZ
+        # points); converged etot verified identical to the full mesh to
+        # 1e-9 Ha
+        kw = dict(gk_cutoff=6.0, pw_cutoff=20.0, ngridk=(4, 4, 4),
+                  use_symmetry=True)
         kw.update(overrides)
         cfg, _ = make_synthetic_config(natoms=5, **kw)
         return SimulationContext(cfg, unit_cell=make_sto_cell(), device=device)
@@ -242,7 +247,10 @@ def make_named_context(model: str, device: str | None = None, **overrides):
                 "smearing": "gaussian",
                 "smearing_width": kw["smearing_width"],
                 "num_mag_dims": 1,
-                "use_symmetry": False,
+                # reference-default IBZ reduction (magnetic subgroup,
+                # moment along z); converged etot verified identical to
+                # the full 4x4x4 mesh to 1e-9 Ha (64 -> 8 points)
+                "use_symmetry": True,
             },
             "mixer": {"type": "anderson", "beta": 0.7},
         })
@@ -252,7 +260,8 @@ def make_named_context(model: str, device: str | None = None, **overrides):
 
 def make_synthetic_config(natoms: int = 8, gk_cutoff: float = 5.0,
                           pw_cutoff: float = 14.0, ngridk=(1, 1, 1),
-                          num_bands: int = -1, smearing_width: float = 0.01) -> tuple:
+                          num_bands: int = -1, smearing_width: float = 0.01,
+                          use_symmetry: bool = False) -> tuple:
     cfg = Config({
         "parameters": {
             "xc_functionals": ["XC_LDA_X", "XC_LDA_C_PZ"],
@@ -261,7 +270,7 @@ def make_synthetic_config(natoms: int = 8, gk_cutoff: float = 5.0,
             "ngridk": list(ngridk),
             "smearing_width": smearing_width,
             "num_bands": num_bands,
-            "use_symmetry": False,
+            "use_symmetry": use_symmetry,
         },
         "mixer": {"type": "anderson", "beta": 0.7},
     })
