@@ -122,6 +122,112 @@ API = [
      "Load a saved state."),
     ("sirius_free_object_handler", [("handler", "handler", "inout")],
      "Release a handler."),
+    # -- round-2 widening (introspection, arrays, SCF seams).  The
+    # -- string-out helpers (option_get_section_name/info) and the
+    # -- no-error-code version getters are C-only.
+    ("sirius_is_initialized", [("status", "bool_out", "out")],
+     "Check if the library is initialized."),
+    ("sirius_get_num_atoms", [("gs_handler", "handler", "in"),
+                              ("num_atoms", "int_out", "out")],
+     "Number of atoms in the unit cell."),
+    ("sirius_get_num_gvec", [("handler", "handler", "in"),
+                             ("num_gvec", "int_out", "out")],
+     "Size of the fine G-vector sphere."),
+    ("sirius_get_num_fft_grid_points",
+     [("handler", "handler", "in"), ("num_fft_grid_points", "int_out", "out")],
+     "Number of fine FFT grid points."),
+    ("sirius_get_fft_index", [("handler", "handler", "in"),
+                              ("fft_index", "int(*)_out", "out")],
+     "1-based FFT-grid offsets of the fine G sphere."),
+    ("sirius_get_num_beta_projectors",
+     [("handler", "handler", "in"), ("label", "string", "in"),
+      ("num_bp", "int_out", "out")],
+     "Number of beta projectors of an atom type."),
+    ("sirius_get_gvec_arrays",
+     [("handler", "handler", "in"), ("gvec", "int(*)_out", "out"),
+      ("gvec_cart", "double(*)_out", "out"),
+      ("gvec_len", "double(*)_out", "out"),
+      ("index_by_gvec", "int(*)_out", "out")],
+     "Miller indices / cartesian coords / lengths of the G sphere."),
+    ("sirius_get_max_num_gkvec", [("ks_handler", "handler", "in"),
+                                  ("max_num_gkvec", "int_out", "out")],
+     "Maximum number of G+k vectors over the k-set."),
+    ("sirius_get_gkvec_arrays",
+     [("ks_handler", "handler", "in"), ("ik", "int", "in"),
+      ("num_gkvec", "int_out", "out"), ("gvec_index", "int(*)_out", "out"),
+      ("gkvec", "double(*)_out", "out"), ("gkvec_cart", "double(*)_out", "out"),
+      ("gkvec_len", "double(*)_out", "out"),
+      ("gkvec_tp", "double(*)_out", "out")],
+     "G+k vector arrays of one k-point (1-based ik)."),
+    ("sirius_set_band_occupancies",
+     [("ks_handler", "handler", "in"), ("ik", "int", "in"),
+      ("ispn", "int", "in"), ("band_occupancies", "double(*)", "in"),
+      ("num_bands", "int", "in")],
+     "Set band occupancies of one k-point/spin."),
+    ("sirius_generate_initial_density", [("gs_handler", "handler", "in")],
+     "Generate the superposition-of-atoms initial density."),
+    ("sirius_generate_effective_potential",
+     [("gs_handler", "handler", "in")],
+     "Generate the effective potential from the current density."),
+    ("sirius_generate_density",
+     [("gs_handler", "handler", "in"), ("add_core", "bool", "in"),
+      ("transform_to_rg", "bool", "in"), ("paw_only", "bool", "in")],
+     "Generate the charge density from the current wave functions."),
+    ("sirius_initialize_subspace",
+     [("gs_handler", "handler", "in"), ("ks_handler", "handler", "in")],
+     "Initialize the Davidson trial subspace."),
+    ("sirius_find_eigen_states",
+     [("gs_handler", "handler", "in"), ("ks_handler", "handler", "in"),
+      ("precompute_pw", "bool", "in"), ("precompute_rf", "bool", "in"),
+      ("precompute_ri", "bool", "in"), ("iter_solver_tol", "double", "in")],
+     "Diagonalize the Hamiltonian for all k-points."),
+    ("sirius_find_band_occupancies", [("ks_handler", "handler", "in")],
+     "Find the Fermi level and band occupancies."),
+    ("sirius_get_periodic_function",
+     [("gs_handler", "handler", "in"), ("label", "string", "in"),
+      ("f_mt", "double(*)_out", "out"), ("lmmax", "int", "in"),
+      ("nrmtmax", "int", "in"), ("num_atoms", "int", "in"),
+      ("f_rg", "double(*)_out", "out"), ("size_x", "int", "in"),
+      ("size_y", "int", "in"), ("size_z", "int", "in"),
+      ("offset_z", "int", "in")],
+     "Real-grid values of a named scalar field."),
+    ("sirius_set_periodic_function",
+     [("gs_handler", "handler", "in"), ("label", "string", "in"),
+      ("f_mt", "double(*)", "in"), ("lmmax", "int", "in"),
+      ("nrmtmax", "int", "in"), ("num_atoms", "int", "in"),
+      ("f_rg", "double(*)", "in"), ("size_x", "int", "in"),
+      ("size_y", "int", "in"), ("size_z", "int", "in"),
+      ("offset_z", "int", "in")],
+     "Set a named scalar field from real-grid values."),
+    ("sirius_get_total_magnetization",
+     [("gs_handler", "handler", "in"), ("mag", "double_arr_out", "out")],
+     "Total magnetization vector (3 components)."),
+    ("sirius_set_atom_vector_field",
+     [("handler", "handler", "in"), ("ia", "int", "in"),
+      ("vector_field", "double(3)", "in")],
+     "Set the initial magnetization vector of one atom (1-based ia)."),
+    ("sirius_set_num_bands", [("handler", "handler", "in"),
+                              ("num_bands", "int", "in")],
+     "Set the number of bands."),
+    ("sirius_set_mpi_grid_dims",
+     [("handler", "handler", "in"), ("ndims", "int", "in"),
+      ("dims", "int(*)", "in")],
+     "Set the MPI grid dimensions (k-groups x band ranks)."),
+    ("sirius_create_context_from_json",
+     [("fcomm", "int_val", "in"), ("handler", "handler_out", "out"),
+      ("fname", "string", "in")],
+     "Create a context directly from a JSON string/file."),
+    ("sirius_update_ground_state", [("gs_handler", "handler", "in")],
+     "Re-generate the potential from the current density."),
+    ("sirius_print_info", [("handler", "handler", "in")],
+     "Print a short context summary."),
+    ("sirius_print_timers", [("flatten", "bool", "in")],
+     "Print timer statistics."),
+    ("sirius_option_get_number_of_sections", [("length", "int_out", "out")],
+     "Number of config schema sections."),
+    ("sirius_option_get_section_length",
+     [("section", "string", "in"), ("length", "int_out", "out")],
+     "Number of options in a schema section."),
 ]
 
 FT = {

@@ -398,4 +398,480 @@ void sirius_free_object_handler(void** handler, int* error_code) {
     set_err(error_code, 0);
 }
 
+
+// ---- round-2 widening: introspection, arrays, SCF seams ----------------
+
+void sirius_is_initialized(bool* status, int* error_code) {
+    *status = impl_module != nullptr;
+    set_err(error_code, 0);
+}
+
+void sirius_get_major_version(int* version) {
+    Gil g;
+    PyObject* r = call_impl("get_version", Py_BuildValue("(s)", "major"));
+    *version = r ? (int)PyLong_AsLong(r) : -1;
+    Py_XDECREF(r);
+}
+
+void sirius_get_minor_version(int* version) {
+    Gil g;
+    PyObject* r = call_impl("get_version", Py_BuildValue("(s)", "minor"));
+    *version = r ? (int)PyLong_AsLong(r) : -1;
+    Py_XDECREF(r);
+}
+
+void sirius_get_revision(int* version) {
+    Gil g;
+    PyObject* r = call_impl("get_version", Py_BuildValue("(s)", "revision"));
+    *version = r ? (int)PyLong_AsLong(r) : -1;
+    Py_XDECREF(r);
+}
+
+void sirius_get_num_atoms(void* const* gs_handler, int* num_atoms,
+                          int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("get_num_atoms",
+                            Py_BuildValue("(O)", (PyObject*)*gs_handler));
+    if (!r) { set_err(error_code, 1); return; }
+    *num_atoms = (int)PyLong_AsLong(r);
+    Py_DECREF(r);
+    set_err(error_code, 0);
+}
+
+void sirius_get_num_gvec(void* const* handler, int* num_gvec,
+                         int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("get_num_gvec",
+                            Py_BuildValue("(O)", (PyObject*)*handler));
+    if (!r) { set_err(error_code, 1); return; }
+    *num_gvec = (int)PyLong_AsLong(r);
+    Py_DECREF(r);
+    set_err(error_code, 0);
+}
+
+void sirius_get_num_fft_grid_points(void* const* handler, int* n,
+                                    int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("get_num_fft_grid_points",
+                            Py_BuildValue("(O)", (PyObject*)*handler));
+    if (!r) { set_err(error_code, 1); return; }
+    *n = (int)PyLong_AsLong(r);
+    Py_DECREF(r);
+    set_err(error_code, 0);
+}
+
+void sirius_get_fft_index(void* const* handler, int* fft_index,
+                          int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("get_fft_index",
+                            Py_BuildValue("(O)", (PyObject*)*handler));
+    if (!r) { set_err(error_code, 1); return; }
+    Py_ssize_t n = PySequence_Length(r);
+    for (Py_ssize_t i = 0; i < n; i++) {
+        PyObject* it = PySequence_GetItem(r, i);
+        fft_index[i] = (int)PyLong_AsLong(it);
+        Py_DECREF(it);
+    }
+    Py_DECREF(r);
+    set_err(error_code, 0);
+}
+
+void sirius_get_num_beta_projectors(void* const* handler, char const* label,
+                                    int* num_bp, int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("get_num_beta_projectors", Py_BuildValue(
+        "(Os)", (PyObject*)*handler, label));
+    if (!r) { set_err(error_code, 1); return; }
+    *num_bp = (int)PyLong_AsLong(r);
+    Py_DECREF(r);
+    set_err(error_code, 0);
+}
+
+void sirius_get_gvec_arrays(void* const* handler, int* gvec,
+                            double* gvec_cart, double* gvec_len,
+                            int* index_by_gvec, int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("get_gvec_arrays",
+                            Py_BuildValue("(O)", (PyObject*)*handler));
+    if (!r) { set_err(error_code, 1); return; }
+    PyObject* mil = PyTuple_GetItem(r, 0);
+    PyObject* cart = PyTuple_GetItem(r, 1);
+    PyObject* len = PyTuple_GetItem(r, 2);
+    if (gvec) {
+        Py_ssize_t n = PySequence_Length(mil);
+        for (Py_ssize_t i = 0; i < n; i++) {
+            PyObject* it = PySequence_GetItem(mil, i);
+            gvec[i] = (int)PyLong_AsLong(it);
+            Py_DECREF(it);
+        }
+    }
+    if (gvec_cart) doubles_from_seq(cart, gvec_cart);
+    if (gvec_len) doubles_from_seq(len, gvec_len);
+    (void)index_by_gvec;  // box map not exported (native ordering applies)
+    Py_DECREF(r);
+    set_err(error_code, 0);
+}
+
+void sirius_get_max_num_gkvec(void* const* ks_handler, int* max_num_gkvec,
+                              int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("get_max_num_gkvec",
+                            Py_BuildValue("(O)", (PyObject*)*ks_handler));
+    if (!r) { set_err(error_code, 1); return; }
+    *max_num_gkvec = (int)PyLong_AsLong(r);
+    Py_DECREF(r);
+    set_err(error_code, 0);
+}
+
+void sirius_get_gkvec_arrays(void* const* ks_handler, int* ik, int* num_gkvec,
+                             int* gvec_index, double* gkvec,
+                             double* gkvec_cart, double* gkvec_len,
+                             double* gkvec_tp, int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("get_gkvec_arrays", Py_BuildValue(
+        "(Oi)", (PyObject*)*ks_handler, *ik - 1));
+    if (!r) { set_err(error_code, 1); return; }
+    *num_gkvec = (int)PyLong_AsLong(PyTuple_GetItem(r, 0));
+    PyObject* idx = PyTuple_GetItem(r, 1);
+    if (gvec_index) {
+        Py_ssize_t n = PySequence_Length(idx);
+        for (Py_ssize_t i = 0; i < n; i++) {
+            PyObject* it = PySequence_GetItem(idx, i);
+            gvec_index[i] = (int)PyLong_AsLong(it);
+            Py_DECREF(it);
+        }
+    }
+    if (gkvec) doubles_from_seq(PyTuple_GetItem(r, 2), gkvec);
+    if (gkvec_cart) doubles_from_seq(PyTuple_GetItem(r, 3), gkvec_cart);
+    if (gkvec_len) doubles_from_seq(PyTuple_GetItem(r, 4), gkvec_len);
+    if (gkvec_tp) doubles_from_seq(PyTuple_GetItem(r, 5), gkvec_tp);
+    Py_DECREF(r);
+    set_err(error_code, 0);
+}
+
+void sirius_get_wave_functions(void* const* ks_handler, double const* vkl,
+                               int const* spin, int const* num_gvec_loc,
+                               int const* gvec_loc, double* evec,
+                               int const* ld, int const* num_spin_comp,
+                               int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("get_wave_functions", Py_BuildValue(
+        "(O[ddd]i)", (PyObject*)*ks_handler, vkl[0], vkl[1], vkl[2],
+        spin ? *spin - 1 : 0));
+    if (!r) { set_err(error_code, 1); return; }
+    int ngk = (int)PyLong_AsLong(PyTuple_GetItem(r, 0));
+    int nb = (int)PyLong_AsLong(PyTuple_GetItem(r, 1));
+    PyObject* flat = PyTuple_GetItem(r, 2);
+    PyObject* fast = PySequence_Fast(flat, "seq");
+    int stride = ld ? *ld : ngk;
+    for (int ib = 0; ib < nb; ib++)
+        for (int igp = 0; igp < ngk; igp++) {
+            evec[2 * (ib * stride + igp)] = PyFloat_AsDouble(
+                PySequence_Fast_GET_ITEM(fast, 2 * (ib * ngk + igp)));
+            evec[2 * (ib * stride + igp) + 1] = PyFloat_AsDouble(
+                PySequence_Fast_GET_ITEM(fast, 2 * (ib * ngk + igp) + 1));
+        }
+    Py_DECREF(fast);
+    (void)num_gvec_loc; (void)gvec_loc; (void)num_spin_comp;
+    Py_DECREF(r);
+    set_err(error_code, 0);
+}
+
+void sirius_set_band_occupancies(void* const* ks_handler, int const* ik,
+                                 int const* ispn,
+                                 double const* band_occupancies,
+                                 int const* num_bands, int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("set_band_occupancies", Py_BuildValue(
+        "(OiiO)", (PyObject*)*ks_handler, *ik - 1, *ispn,
+        list_from_doubles(band_occupancies, *num_bands)));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_generate_initial_density(void* const* gs_handler,
+                                     int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("generate_initial_density",
+                            Py_BuildValue("(O)", (PyObject*)*gs_handler));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_generate_effective_potential(void* const* gs_handler,
+                                         int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("generate_effective_potential",
+                            Py_BuildValue("(O)", (PyObject*)*gs_handler));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_generate_density(void* const* gs_handler,
+                             bool const* add_core,
+                             bool const* transform_to_rg,
+                             bool const* paw_only, int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("generate_density", Py_BuildValue(
+        "(Oii)", (PyObject*)*gs_handler, add_core ? (int)*add_core : 0,
+        transform_to_rg ? (int)*transform_to_rg : 0));
+    (void)paw_only;
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_initialize_subspace(void* const* gs_handler,
+                                void* const* ks_handler, int* error_code) {
+    Gil g;
+    (void)ks_handler;
+    PyObject* r = call_impl("initialize_subspace",
+                            Py_BuildValue("(O)", (PyObject*)*gs_handler));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_find_eigen_states(void* const* gs_handler,
+                              void* const* ks_handler,
+                              bool const* precompute_pw,
+                              bool const* precompute_rf,
+                              bool const* precompute_ri,
+                              double const* iter_solver_tol,
+                              int* error_code) {
+    Gil g;
+    (void)ks_handler; (void)precompute_rf; (void)precompute_ri;
+    PyObject* r = call_impl("find_eigen_states", Py_BuildValue(
+        "(Oid)", (PyObject*)*gs_handler,
+        precompute_pw ? (int)*precompute_pw : 1,
+        iter_solver_tol ? *iter_solver_tol : 1e-5));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_find_band_occupancies(void* const* ks_handler, int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("find_band_occupancies",
+                            Py_BuildValue("(O)", (PyObject*)*ks_handler));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_get_periodic_function(void* const* gs_handler, char const* label,
+                                  double* f_mt, int const* lmmax,
+                                  int const* nrmtmax, int const* num_atoms,
+                                  double* f_rg, int const* size_x,
+                                  int const* size_y, int const* size_z,
+                                  int const* offset_z, int* error_code) {
+    Gil g;
+    (void)f_mt; (void)lmmax; (void)nrmtmax; (void)num_atoms;
+    (void)size_x; (void)size_y; (void)size_z; (void)offset_z;
+    PyObject* r = call_impl("get_periodic_function", Py_BuildValue(
+        "(Os)", (PyObject*)*gs_handler, label));
+    if (!r) { set_err(error_code, 1); return; }
+    if (f_rg) doubles_from_seq(r, f_rg);
+    Py_DECREF(r);
+    set_err(error_code, 0);
+}
+
+void sirius_set_periodic_function(void* const* gs_handler, char const* label,
+                                  double* f_mt, int const* lmmax,
+                                  int const* nrmtmax, int const* num_atoms,
+                                  double* f_rg, int const* size_x,
+                                  int const* size_y, int const* size_z,
+                                  int const* offset_z, int* error_code) {
+    Gil g;
+    (void)f_mt; (void)lmmax; (void)nrmtmax; (void)num_atoms; (void)offset_z;
+    int n = (size_x && size_y && size_z) ? (*size_x) * (*size_y) * (*size_z)
+                                         : 0;
+    PyObject* r = call_impl("set_periodic_function", Py_BuildValue(
+        "(OsO[iii])", (PyObject*)*gs_handler, label,
+        list_from_doubles(f_rg, n), size_x ? *size_x : 0,
+        size_y ? *size_y : 0, size_z ? *size_z : 0));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_get_total_magnetization(void* const* gs_handler, double* mag,
+                                    int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("get_total_magnetization",
+                            Py_BuildValue("(O)", (PyObject*)*gs_handler));
+    if (!r) { set_err(error_code, 1); return; }
+    doubles_from_seq(r, mag);
+    Py_DECREF(r);
+    set_err(error_code, 0);
+}
+
+void sirius_set_atom_vector_field(void* const* handler, int const* ia,
+                                  double const* vector_field,
+                                  int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("set_atom_vector_field", Py_BuildValue(
+        "(OiO)", (PyObject*)*handler, *ia - 1,
+        list_from_doubles(vector_field, 3)));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_set_num_bands(void* const* handler, int* const num_bands,
+                          int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("set_num_bands", Py_BuildValue(
+        "(Oi)", (PyObject*)*handler, *num_bands));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_set_mpi_grid_dims(void* const* handler, int const* ndims,
+                              int const* dims, int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("set_mpi_grid_dims", Py_BuildValue(
+        "(OO)", (PyObject*)*handler, list_from_ints(dims, *ndims)));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_create_context_from_json(int fcomm, void** handler,
+                                     char const* fname, int* error_code) {
+    Gil g;
+    (void)fcomm;
+    PyObject* h = call_impl("create_context_from_json",
+                            Py_BuildValue("(s)", fname));
+    if (!h) { set_err(error_code, 1); return; }
+    *handler = h;
+    set_err(error_code, 0);
+}
+
+void sirius_update_ground_state(void** gs_handler, int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("update_ground_state",
+                            Py_BuildValue("(O)", (PyObject*)*gs_handler));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_print_info(void* const* handler, int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("print_info",
+                            Py_BuildValue("(O)", (PyObject*)*handler));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_print_timers(bool* flatten, int* error_code) {
+    Gil g;
+    (void)flatten;
+    PyObject* r = call_impl("print_timers", PyTuple_New(0));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_option_get_number_of_sections(int* length, int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("option_get_number_of_sections", PyTuple_New(0));
+    if (!r) { set_err(error_code, 1); return; }
+    *length = (int)PyLong_AsLong(r);
+    Py_DECREF(r);
+    set_err(error_code, 0);
+}
+
+void sirius_option_get_section_name(int elem, char* section_name,
+                                    int section_name_length,
+                                    int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("option_get_section_name",
+                            Py_BuildValue("(i)", elem - 1));
+    if (!r) { set_err(error_code, 1); return; }
+    const char* s = PyUnicode_AsUTF8(r);
+    std::snprintf(section_name, section_name_length, "%s", s ? s : "");
+    Py_DECREF(r);
+    set_err(error_code, 0);
+}
+
+void sirius_option_get_section_length(char const* section, int* length,
+                                      int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("option_get_section_length",
+                            Py_BuildValue("(s)", section));
+    if (!r) { set_err(error_code, 1); return; }
+    *length = (int)PyLong_AsLong(r);
+    Py_DECREF(r);
+    set_err(error_code, 0);
+}
+
+void sirius_option_get_info(char const* section, int elem, char* key_name,
+                            int key_name_len, int* type, int* length,
+                            int* enum_size, char* title, int title_len,
+                            char* description, int description_len,
+                            int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("option_get_info", Py_BuildValue(
+        "(si)", section, elem - 1));
+    if (!r) { set_err(error_code, 1); return; }
+    const char* key = PyUnicode_AsUTF8(PyTuple_GetItem(r, 0));
+    std::snprintf(key_name, key_name_len, "%s", key ? key : "");
+    *type = (int)PyLong_AsLong(PyTuple_GetItem(r, 1));
+    *length = (int)PyLong_AsLong(PyTuple_GetItem(r, 2));
+    if (enum_size) *enum_size = 0;
+    if (title && title_len > 0) title[0] = 0;
+    if (description && description_len > 0) description[0] = 0;
+    Py_DECREF(r);
+    set_err(error_code, 0);
+}
+
+void sirius_option_get(char const* section, char const* name,
+                       int const* type, void* data_ptr,
+                       int const* max_length, int const* enum_idx,
+                       int* error_code) {
+    Gil g;
+    (void)enum_idx;
+    PyObject* r = call_impl("option_get", Py_BuildValue("(ss)", section,
+                                                        name));
+    if (!r) { set_err(error_code, 1); return; }
+    int code = (int)PyLong_AsLong(PyTuple_GetItem(r, 0));
+    PyObject* v = PyTuple_GetItem(r, 1);
+    int want = type ? *type : code;
+    int cap = max_length ? *max_length : 1;
+    switch (want) {
+        case 1:  // int
+            *(int*)data_ptr = (int)PyLong_AsLong(v);
+            break;
+        case 2:  // bool
+            *(bool*)data_ptr = PyObject_IsTrue(v);
+            break;
+        case 4:  // double
+            *(double*)data_ptr = PyFloat_AsDouble(v);
+            break;
+        case 3: {  // string
+            const char* s = PyUnicode_AsUTF8(v);
+            std::snprintf((char*)data_ptr, cap, "%s", s ? s : "");
+            break;
+        }
+        case 7: {  // int array
+            Py_ssize_t n = PySequence_Length(v);
+            for (Py_ssize_t i = 0; i < n && i < cap; i++) {
+                PyObject* it = PySequence_GetItem(v, i);
+                ((int*)data_ptr)[i] = (int)PyLong_AsLong(it);
+                Py_DECREF(it);
+            }
+            break;
+        }
+        case 9: {  // double array
+            Py_ssize_t n = PySequence_Length(v);
+            PyObject* fast = PySequence_Fast(v, "seq");
+            for (Py_ssize_t i = 0; i < n && i < cap; i++)
+                ((double*)data_ptr)[i] = PyFloat_AsDouble(
+                    PySequence_Fast_GET_ITEM(fast, i));
+            Py_DECREF(fast);
+            break;
+        }
+        default:
+            Py_DECREF(r);
+            set_err(error_code, 2);
+            return;
+    }
+    Py_DECREF(r);
+    set_err(error_code, 0);
+}
+
 }  // extern "C"
+
