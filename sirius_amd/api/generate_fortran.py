@@ -228,6 +228,38 @@ API = [
     ("sirius_option_get_section_length",
      [("section", "string", "in"), ("length", "int_out", "out")],
      "Number of options in a schema section."),
+    ("sirius_set_atom_type_dion",
+     [("handler", "handler", "in"), ("label", "string", "in"),
+      ("num_beta", "int", "in"), ("dion", "double(*)", "in")],
+     "Set the ionic D matrix of an atom type."),
+    ("sirius_set_atom_type_paw",
+     [("handler", "handler", "in"), ("label", "string", "in"),
+      ("core_energy", "double", "in"), ("occupations", "double(*)", "in"),
+      ("num_occ", "int", "in")],
+     "Set PAW core energy and wave occupations."),
+    ("sirius_set_atom_type_configuration",
+     [("handler", "handler", "in"), ("label", "string", "in"),
+      ("n", "int", "in"), ("l", "int", "in"), ("k", "int", "in"),
+      ("occupancy", "double", "in"), ("core", "bool", "in")],
+     "Add one atomic level to the configuration."),
+    ("sirius_add_atom_type_aw_descriptor",
+     [("handler", "handler", "in"), ("label", "string", "in"),
+      ("n", "int", "in"), ("l", "int", "in"), ("enu", "double", "in"),
+      ("dme", "int", "in"), ("auto_enu", "bool", "in")],
+     "Add an APW radial-solution descriptor (LAPW)."),
+    ("sirius_add_atom_type_lo_descriptor",
+     [("handler", "handler", "in"), ("label", "string", "in"),
+      ("ilo", "int", "in"), ("n", "int", "in"), ("l", "int", "in"),
+      ("enu", "double", "in"), ("dme", "int", "in"),
+      ("auto_enu", "bool", "in")],
+     "Add a local-orbital descriptor (LAPW)."),
+    ("sirius_set_equivalent_atoms",
+     [("handler", "handler", "in"), ("equivalent_atoms", "int(*)", "in")],
+     "Set the equivalent-atom map (overrides autodetection)."),
+    ("sirius_get_fv_eigen_values",
+     [("ks_handler", "handler", "in"), ("ik", "int", "in"),
+      ("fv_eval", "double(*)_out", "out"), ("num_fv_states", "int", "in")],
+     "First-variational eigenvalues of one k-point."),
 ]
 
 FT = {

@@ -873,5 +873,98 @@ void sirius_option_get(char const* section, char const* name,
     set_err(error_code, 0);
 }
 
+
+void sirius_set_atom_type_dion(void* const* handler, char const* label,
+                               int const* num_beta, double* dion,
+                               int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("set_atom_type_dion", Py_BuildValue(
+        "(OsOi)", (PyObject*)*handler, label,
+        list_from_doubles(dion, (*num_beta) * (*num_beta)), *num_beta));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_set_atom_type_paw(void* const* handler, char const* label,
+                              double const* core_energy,
+                              double const* occupations, int const* num_occ,
+                              int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("set_atom_type_paw", Py_BuildValue(
+        "(OsdOi)", (PyObject*)*handler, label, *core_energy,
+        list_from_doubles(occupations, *num_occ), *num_occ));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_set_atom_type_configuration(void* const* handler,
+                                        char const* label, int const* n,
+                                        int const* l, int const* k,
+                                        double const* occupancy,
+                                        bool const* core, int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("set_atom_type_configuration", Py_BuildValue(
+        "(Osiiidi)", (PyObject*)*handler, label, *n, *l, *k, *occupancy,
+        (int)*core));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_add_atom_type_aw_descriptor(void* const* handler,
+                                        char const* label, int const* n,
+                                        int const* l, double const* enu,
+                                        int const* dme,
+                                        bool const* auto_enu,
+                                        int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("add_atom_type_aw_descriptor", Py_BuildValue(
+        "(Osiidii)", (PyObject*)*handler, label, *n, *l, *enu, *dme,
+        (int)*auto_enu));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_add_atom_type_lo_descriptor(void* const* handler,
+                                        char const* label, int const* ilo,
+                                        int const* n, int const* l,
+                                        double const* enu, int const* dme,
+                                        bool const* auto_enu,
+                                        int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("add_atom_type_lo_descriptor", Py_BuildValue(
+        "(Osiiidii)", (PyObject*)*handler, label, *ilo - 1, *n, *l, *enu,
+        *dme, (int)*auto_enu));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_set_equivalent_atoms(void* const* handler,
+                                 int* equivalent_atoms, int* error_code) {
+    Gil g;
+    // length = current atom count on the handler
+    PyObject* na = call_impl("ctx_num_atoms",
+                             Py_BuildValue("(O)", (PyObject*)*handler));
+    if (!na) { set_err(error_code, 1); return; }
+    int n = (int)PyLong_AsLong(na);
+    Py_DECREF(na);
+    PyObject* r = call_impl("set_equivalent_atoms", Py_BuildValue(
+        "(OO)", (PyObject*)*handler, list_from_ints(equivalent_atoms, n)));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_get_fv_eigen_values(void* const* ks_handler, int const* ik,
+                                double* fv_eval, int const* num_fv_states,
+                                int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("get_fv_eigen_values", Py_BuildValue(
+        "(Oii)", (PyObject*)*ks_handler, *ik - 1, *num_fv_states));
+    if (!r) { set_err(error_code, 1); return; }
+    doubles_from_seq(r, fv_eval);
+    Py_DECREF(r);
+    set_err(error_code, 0);
+}
+
 }  // extern "C"
+
 

@@ -600,3 +600,69 @@ def option_get(section: str, name: str):
     if isinstance(v, (list, tuple)):
         v = list(v)
     return code, v
+
+
+# ---- species-setup extensions (PAW / LAPW programmatic path) -----------
+
+def set_atom_type_paw(h: CtxHandle, label: str, core_energy: float,
+                      occupations, num_occ: int):
+    at = h.types[label]
+    at.paw_core_energy = float(core_energy)
+    at.paw_wf_occ = [float(x) for x in occupations[:num_occ]]
+    at.is_paw = True
+
+
+def set_atom_type_configuration(h: CtxHandle, label: str, n: int, l: int,
+                                k: int, occupancy: float, core: bool):
+    """Accumulate the atomic level list (reference
+    sirius_set_atom_type_configuration; consumed by the free-atom
+    density / LAPW core solver)."""
+    at = h.types[label]
+    if not hasattr(at, "configuration") or at.configuration is None:
+        at.configuration = []
+    at.configuration.append({"n": int(n), "l": int(l), "k": int(k),
+                             "occupancy": float(occupancy),
+                             "core": bool(core)})
+
+
+def add_atom_type_aw_descriptor(h: CtxHandle, label: str, n: int, l: int,
+                                enu: float, dme: int, auto_enu: bool):
+    """Accumulate APW radial-solution descriptors in the species-JSON
+    shape FPAtomType parses (lapw/species.py)."""
+    at = h.types[label]
+    if not hasattr(at, "aw_specific") or at.aw_specific is None:
+        at.aw_specific = {}
+    at.aw_specific.setdefault(int(l), []).append(
+        {"n": int(n), "enu": float(enu), "dme": int(dme),
+         "auto": int(bool(auto_enu))})
+
+
+def add_atom_type_lo_descriptor(h: CtxHandle, label: str, ilo: int, n: int,
+                                l: int, enu: float, dme: int,
+                                auto_enu: bool):
+    at = h.types[label]
+    if not hasattr(at, "lo_descriptors") or at.lo_descriptors is None:
+        at.lo_descriptors = {}
+    at.lo_descriptors.setdefault(int(ilo), {"l": int(l), "basis": []})
+    at.lo_descriptors[int(ilo)]["basis"].append(
+        {"n": int(n), "enu": float(enu), "dme": int(dme),
+         "auto": int(bool(auto_enu))})
+
+
+def set_equivalent_atoms(h: CtxHandle, eq):
+    h.cfg_data.setdefault("unit_cell", {})["equivalent_atoms"] = \
+        [int(x) for x in eq]
+
+
+def get_fv_eigen_values(ks: KsetHandle, ik: int, num_fv_states: int):
+    kp = ks.kset.kpoints[ik]
+    ev = getattr(kp, "fv_eval", None)
+    if ev is None:
+        ev = kp.eigvals[0]
+    return np.asarray(ev, dtype=np.float64)[:num_fv_states].tolist()
+
+
+def ctx_num_atoms(h: CtxHandle) -> int:
+    if h.ctx is not None:
+        return int(h.ctx.unit_cell.num_atoms)
+    return len(h.atoms)
