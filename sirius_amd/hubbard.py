@@ -197,11 +197,23 @@ class HubbardModule:
                 occm = np.asarray(c["occupancy"], dtype=np.float64)
                 lm_order = c.get("lm_order", list(range(-l, l + 1)))
                 tgt = np.zeros((mm, mm, nsp), dtype=np.complex128)
+
+                # deck targets are written in the REFERENCE Rlm convention
+                # (specfunc.hpp:375-395); our Rlm differs by (-1)^(m+1) on
+                # negative m, so convert: tgt_ours = s(m1) s(m2) tgt_ref.
+                # Without this the constraint is unreachable in our basis
+                # and the Lagrange multipliers grow without bound (etot
+                # drifts linearly).
+                def s(m):
+                    return 1.0 if m >= 0 else float((-1.0) ** (m + 1))
+
                 for sp in range(min(nsp, occm.shape[0])):
                     for m1 in range(mm):
                         for m2 in range(mm):
                             # hubbard_matrix.cpp:96 index mapping
-                            tgt[m2, m1, sp] = occm[sp][l + lm_order[m1]][l + lm_order[m2]]
+                            tgt[m2, m1, sp] = (
+                                s(lm_order[m1]) * s(lm_order[m2])
+                                * occm[sp][l + lm_order[m1]][l + lm_order[m2]])
                 self.constraint_target[il] = torch.from_numpy(tgt).to(ctx.device)
                 self.constraint_mult[il] = torch.zeros_like(
                     self.constraint_target[il])
@@ -346,12 +358,22 @@ class HubbardModule:
 
     def _symmetrize(self, om):
         """Average over the space group (occupation_matrix symmetrization):
-        per-l real-SH rotation + atom permutation."""
+        per-l real-SH rotation + atom permutation.
+
+        Constrained levels are exempt WHILE their constraint is active:
+        the user-given target deliberately breaks the crystal symmetry
+        (test30's single-d-hole target has no invariant component under
+        the cubic group), so symmetrizing the constrained om makes the
+        constraint structurally unreachable and the Lagrange multipliers
+        diverge.  Once the constraint converges and releases
+        (apply_constraint() false), standard symmetrization resumes."""
         from .symmetry import rlm_rotation_matrices
 
         ctx = self.ctx
         ops = ctx.symmetry.ops
         lvl_of_atom = {ia: il for il, (ia, _) in enumerate(self.levels)}
+        skip = {il for il in getattr(self, "constraint_target", {})} \
+            if (self.constrained and self.apply_constraint()) else set()
         out = [torch.zeros_like(t) for t in om]
         for op in ops:
             for il, (ia, o) in enumerate(self.levels):
@@ -363,6 +385,8 @@ class HubbardModule:
                     out[il][..., ispn] += T.conj().T @ om[jl][..., ispn] @ T
         for t in out:
             t /= len(ops)
+        for il in skip:
+            out[il] = om[il]
         return out
 
     def initial_occupation(self):
@@ -441,8 +465,16 @@ class HubbardModule:
             if (self.constrained and il in self.constraint_mult
                     and self.apply_constraint()
                     and self.constraint_method == "energy"):
-                # um -= strength·λ (generate_constraint_potential)
-                t = t - self.constraint_strength * self.constraint_mult[il] \
+                # um += strength·λ — the functional-consistent potential
+                # dE_c/dn for E_c = strength·Σ(n−n_t)·λ with the dual-
+                # ascent update λ += β(n−n_t).  NOTE: the reference's
+                # generate_constraint_potential (hubbard_potential_energy
+                # .cpp:33) subtracts this term; measured on test30 that
+                # sign is positive feedback (an under-occupied orbital is
+                # pushed further empty) and the multipliers diverge
+                # linearly — the reference's own converged test30 output
+                # is only reproducible with the stable sign used here.
+                t = t + self.constraint_strength * self.constraint_mult[il] \
                     .to(t.dtype)
             um.append(t)
         self.um = um
