@@ -45,7 +45,64 @@ MODEL_LABEL = {
     "si512": "Si512-NC-LDA-Gamma",
     "sto-uspp": "SrTiO3-USPP-4x4x4k",
     "fe-paw": "Fe-bcc-PAW-collinear-12x12x12k",
+    "nio-lapw": "NiO4-FP-LAPW-AFM-6x6x6k",
 }
+
+
+def bench_nio_lapw(args, comm):
+    """BASELINE config 5 (partial coverage): NiO 4-atom FP-LAPW AFM on the
+    real test16 species, 6x6x6 k default.  This engine's LAPW branch runs
+    the collinear AFM LSDA problem (the reference raises for LAPW+Hubbard;
+    non-collinear LAPW is a round-3 item) and orchestrates on the CPU —
+    an honest but unaccelerated number; see NEXT.md."""
+    if comm.size > 1:
+        raise SystemExit("nio-lapw bench supports 1 rank (k-parallel LAPW "
+                         "bench is a round-3 item)")
+    from sirius_amd.lapw.engine import make_lapw_context, FPGroundState
+    from sirius_amd.config import Config
+
+    deck = json.load(open("verification/test16/sirius.json"))
+    deck["parameters"]["ngridk"] = list(args.ngridk) if args.ngridk \
+        else [6, 6, 6]
+    deck["control"] = dict(deck.get("control", {}), verbosity=0)
+    ctx = make_lapw_context(Config(deck), base_dir="verification/test16")
+    kset = KPointSet(ctx)
+    gs = FPGroundState(kset).initial_state()
+    print(f"# model=nio-lapw natoms=4 nk={kset.num_kpoints} "
+          f"nfv={ctx.cfg.parameters.num_fv_states} (cpu-orchestrated)",
+          flush=True)
+    for _ in range(args.warmup):
+        gs.scf_iteration()
+    t0 = time.time()
+    for _ in range(args.steps):
+        gs.scf_iteration()
+    dt = time.time() - t0
+    print(json.dumps({
+        "metric": "sec/SCF-iteration",
+        "value": dt / args.steps,
+        "unit": "s",
+        "n_gpus": comm.size,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": dt / args.steps * 1000.0,
+        "higher_is_better": False,
+        "scaling": "strong",
+        "vs_baseline": None,
+        "dtype": "fp64",
+        "data": "test16 deck (staged reference verification data)",
+        "config": {
+            "model": MODEL_LABEL["nio-lapw"],
+            "baseline_config": 5,
+            "natoms": 4,
+            "num_fv_states": int(ctx.cfg.parameters.num_fv_states),
+            "num_mag_dims": 1,
+            "ngridk": deck["parameters"]["ngridk"],
+            "num_kpoints": kset.num_kpoints,
+            "parallelism": "cpu-orchestrated (LAPW GPU path: NEXT.md)",
+            "coverage": "collinear AFM LSDA; no Hubbard U (reference "
+                        "raises for LAPW+U), no nc-magnetism yet",
+        },
+    }), flush=True)
 
 
 def main():
@@ -77,6 +134,9 @@ def main():
         overrides["pw_cutoff"] = args.pw_cutoff
     if args.ngridk is not None:
         overrides["ngridk"] = tuple(args.ngridk)
+
+    if args.model == "nio-lapw":
+        return bench_nio_lapw(args, comm)
 
     ctx = make_named_context(args.model, device=device, **overrides)
     kset = KPointSet(ctx)
