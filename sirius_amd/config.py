@@ -84,6 +84,11 @@ _DEFAULTS: dict[str, dict[str, Any]] = {
         "extra_ortho": False,
         "min_tolerance": 1e-13,
         "tolerance_ratio": 0.0,
+        # MI355X extension: cap the Davidson expansion block (0 = all
+        # unconverged residuals).  Large-band Γ-only cells are subspace-
+        # transform bound; a cap trades more (cheaper) iterations for
+        # smaller transforms.
+        "max_block_size": 0,
         "tolerance_scale": [0.1, 0.5],
         "relative_tolerance": 0,
         "init_eval_old": True,

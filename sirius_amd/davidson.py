@@ -99,6 +99,7 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
              o_diag: torch.Tensor, occ: np.ndarray | None,
              tol_occ: float, tol_empty: float,
              num_steps: int = 20, subspace_size: int = 2,
+             max_block: int = 0,
              min_occupancy: float = 1e-14,
              extra_ortho: bool = False, locking: bool = True,
              early_restart: float = 0.5, gamma: bool = False,
@@ -183,12 +184,14 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
     nlock = 0                      # number of locked (frozen) leading bands
     eval_locked = np.zeros(0)
 
+    pending = np.array([], dtype=np.int64)   # capped-out, not yet expanded
     for it in range(num_steps):
         niter = it + 1
         nb_act = nb - nlock        # bands still solved in the active block
         ev = np.concatenate([eval_locked,
                              evals[:nb_act].real.cpu().numpy()])
         unconv = np.nonzero(np.abs(ev - eval_old[: len(ev)]) > tol)[0]
+        unconv = np.union1d(unconv, pending).astype(np.int64)
         unconv = unconv[unconv >= nlock]
         eval_old = ev.copy()
         if len(unconv) == 0:
@@ -224,6 +227,19 @@ def davidson(apply_h_s, psi0: torch.Tensor, h_diag: torch.Tensor,
             converged = True
             break
         res = res[keep]
+        if max_block and res.shape[0] > max_block:
+            # cap the expansion block (lowest bands first): trades more,
+            # cheaper steps for smaller subspace transforms.  Bands cut
+            # from the block stay marked pending — an unexpanded band's
+            # eigenvalue barely moves, which would otherwise fool the
+            # |de| convergence test
+            cut = unconv[keep.cpu().numpy()][max_block:]
+            pending = np.union1d(pending, cut)
+            res = res[:max_block]
+            taken = unconv[keep.cpu().numpy()][:max_block]
+            pending = np.setdiff1d(pending, taken)
+        elif len(pending):
+            pending = np.setdiff1d(pending, unconv[keep.cpu().numpy()])
         res = res / torch.linalg.vector_norm(res, dim=1, keepdim=True).to(res.dtype)
         res = res.contiguous()
         enforce_gamma(res)

@@ -286,6 +286,7 @@ def diagonalize(ctx, h0: Hamiltonian0, kset: KPointSet, itsol_tol: float,
                 psi0, h_diag, o_diag, occ=kp.occ[ispn],
                 tol_occ=itsol_tol, tol_empty=empy_tol,
                 num_steps=itso.num_steps, subspace_size=itso.subspace_size,
+                max_block=int(itso.get("max_block_size", 0)),
                 min_occupancy=itso.min_occupancy, extra_ortho=itso.extra_ortho,
                 locking=bool(itso.locking),
                 early_restart=float(itso.early_restart),
