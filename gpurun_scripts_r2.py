@@ -1,16 +1,43 @@
+"""Scratch GPU A/B: Davidson expansion-block cap on si512 (not packaged)."""
 import json
-from sirius_amd.models.synthetic import make_synthetic_config, make_synthetic_cell
-from sirius_amd.context import SimulationContext
+import time
+
+import torch
+
+from sirius_amd.models.synthetic import make_named_context
 from sirius_amd.kpoint import KPointSet
-from sirius_amd.dft import DFTGroundState
+from sirius_amd.dft import DFTGroundState, diagonalize
+from sirius_amd.hamiltonian import Hamiltonian0
+
 outs = {}
-for gamma in (False, True):
-    cfg, _ = make_synthetic_config(natoms=8, gk_cutoff=4.0, pw_cutoff=10.0, ngridk=(1,1,1))
-    cfg._data["parameters"]["gamma_point"] = gamma
-    cfg.parameters.gamma_point = gamma
-    ctx = SimulationContext(cfg, unit_cell=make_synthetic_cell(8), device="cuda:0")
-    dft = DFTGroundState(KPointSet(ctx)).initial_state()
-    hist = []
-    r = dft.find(num_dft_iter=25, callback=lambda it,e,rm: hist.append((it, e, rm)))
-    outs[str(gamma)] = {"etot": r["energy"]["total"], "hist": [(h[0], round(h[1],8), float(h[2])) for h in hist[-6:]]}
-print(json.dumps(outs, indent=1))
+for mb in (0, 384, 256):
+    torch.manual_seed(0)
+    ctx = make_named_context("si512", device="cuda:0")
+    ctx.cfg.iterative_solver.set("max_block_size", mb)
+    kset = KPointSet(ctx)
+    dft = DFTGroundState(kset).initial_state()
+    dft.density.mixer_init(ctx.cfg.mixer)
+
+    def step():
+        h0 = Hamiltonian0(ctx, dft.potential, dft.density)
+        diagonalize(ctx, h0, kset, 1e-4)
+        kset.find_band_occupancies()
+        dft.density.generate(kset, h0)
+        dft.density.mix()
+        dft.potential.generate(dft.density)
+
+    for _ in range(2):
+        step()
+    torch.cuda.synchronize()
+    t0 = time.time()
+    for _ in range(3):
+        step()
+    torch.cuda.synchronize()
+    dt = (time.time() - t0) / 3
+    ev = kset.kpoints[0].eigvals[0]
+    outs[mb] = {"s_per_iter": round(dt, 3), "ev0": float(ev[0]),
+                "ev_last_occ": float(ev[1023])}
+    print(mb, outs[mb], flush=True)
+    del dft, kset, ctx
+    torch.cuda.empty_cache()
+print(json.dumps(outs))

@@ -85,9 +85,10 @@ _DEFAULTS: dict[str, dict[str, Any]] = {
         "min_tolerance": 1e-13,
         "tolerance_ratio": 0.0,
         # MI355X extension: cap the Davidson expansion block (0 = all
-        # unconverged residuals).  Large-band Γ-only cells are subspace-
-        # transform bound; a cap trades more (cheaper) iterations for
-        # smaller transforms.
+        # unconverged residuals).  MEASURED on si512 (1228 bands, 1 GPU):
+        # cap 256/384 runs ~3x SLOWER per SCF iteration than uncapped —
+        # the pending bands keep every Davidson call at its step limit.
+        # Kept as an experiment knob; default stays off.
         "max_block_size": 0,
         "tolerance_scale": [0.1, 0.5],
         "relative_tolerance": 0,
