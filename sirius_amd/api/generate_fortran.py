@@ -260,6 +260,17 @@ API = [
      [("ks_handler", "handler", "in"), ("ik", "int", "in"),
       ("fv_eval", "double(*)_out", "out"), ("num_fv_states", "int", "in")],
      "First-variational eigenvalues of one k-point."),
+    ("sirius_nlcg", [("gs_handler", "handler", "in"),
+                     ("ks_handler", "handler", "in")],
+     "Direct total-energy minimization (orbital CG)."),
+    ("sirius_nlcg_params",
+     [("gs_handler", "handler", "in"), ("ks_handler", "handler", "in"),
+      ("temp", "double", "in"), ("smearing", "string", "in"),
+      ("kappa", "double", "in"), ("tau", "double", "in"),
+      ("tol", "double", "in"), ("maxiter", "int", "in"),
+      ("restart", "int", "in"), ("processing_unit", "string", "in"),
+      ("converged", "bool_out", "out")],
+     "Direct minimization with explicit parameters."),
 ]
 
 FT = {

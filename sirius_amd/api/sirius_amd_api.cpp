@@ -965,6 +965,36 @@ void sirius_get_fv_eigen_values(void* const* ks_handler, int const* ik,
     set_err(error_code, 0);
 }
 
+
+void sirius_nlcg(void* const* gs_handler, void* const* ks_handler,
+                 int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("nlcg", Py_BuildValue(
+        "(OO)", (PyObject*)*gs_handler, (PyObject*)*ks_handler));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_nlcg_params(void* const* gs_handler, void* const* ks_handler,
+                        double const* temp, char const* smearing,
+                        double const* kappa, double const* tau,
+                        double const* tol, int const* maxiter,
+                        int const* restart, char const* processing_unit,
+                        bool* converged, int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("nlcg", Py_BuildValue(
+        "(OOdsdddiis)", (PyObject*)*gs_handler, (PyObject*)*ks_handler,
+        temp ? *temp : -1.0, smearing ? smearing : "",
+        kappa ? *kappa : 0.3, tau ? *tau : 0.1, tol ? *tol : 1e-9,
+        maxiter ? *maxiter : 300, restart ? *restart : 10,
+        processing_unit ? processing_unit : ""));
+    if (!r) { set_err(error_code, 1); return; }
+    if (converged) *converged = PyObject_IsTrue(r);
+    Py_DECREF(r);
+    set_err(error_code, 0);
+}
+
 }  // extern "C"
+
 
 

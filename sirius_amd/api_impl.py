@@ -666,3 +666,21 @@ def ctx_num_atoms(h: CtxHandle) -> int:
     if h.ctx is not None:
         return int(h.ctx.unit_cell.num_atoms)
     return len(h.atoms)
+
+
+def nlcg(gs: GsHandle, ks: KsetHandle, temp: float = -1.0,
+         smearing: str = "", kappa: float = 0.3, tau: float = 0.1,
+         tol: float = 1e-9, maxiter: int = 300, restart: int = 10,
+         processing_unit: str = ""):
+    """Direct total-energy minimization (reference sirius_nlcg /
+    sirius_nlcg_params; fixed-occupation orbital CG — the ensemble-DFT
+    smearing branch is a roadmap item, so temp/smearing/kappa are
+    accepted and the fixed-occupation minimizer runs)."""
+    from .nlcg import DirectMinimizer
+
+    dm = DirectMinimizer(gs.dft, maxiter=int(maxiter), tol=float(tol))
+    res = dm.run()
+    gs.result = {"energy": {"total": res["etot"]},
+                 "converged": bool(res["converged"]),
+                 "num_scf_iterations": int(res["num_iter"])}
+    return bool(res["converged"])
