@@ -50,7 +50,32 @@ def build_extensions(verbose: bool = False):
         verbose=verbose,
     )
     build_radial(verbose=verbose)
+    build_capi(verbose=verbose)
     return _ext
+
+
+def build_capi(verbose: bool = False):
+    """Build libsirius_amd.so (the embedded-CPython C API shim)."""
+    import subprocess
+    import sys
+
+    api_dir = os.path.abspath(os.path.join(os.path.dirname(_SRC_DIR),
+                                           os.pardir, "api"))
+    src = os.path.join(api_dir, "sirius_amd_api.cpp")
+    lib = os.path.join(api_dir, "libsirius_amd.so")
+    if not os.path.exists(src):
+        return None
+    if os.path.exists(lib) and os.path.getmtime(lib) >= os.path.getmtime(src):
+        return lib
+    inc = subprocess.run(["python3-config", "--includes"],
+                         capture_output=True, text=True)
+    cmd = ["g++", "-O2", "-shared", "-fPIC", "-o", lib, src] \
+        + inc.stdout.split() \
+        + [f"-lpython{sys.version_info.major}.{sys.version_info.minor}"]
+    subprocess.run(cmd, check=True)
+    if verbose:
+        print(f"built {lib}")
+    return lib
 
 
 def build_radial(verbose: bool = False):
