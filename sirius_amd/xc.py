@@ -151,18 +151,25 @@ _PBE_BETA = 0.06672455060314922
 _PBE_GAMMA = (1.0 - math.log(2.0)) / math.pi**2
 
 
-def gga_x_pbe(rho: torch.Tensor, sigma: torch.Tensor, mu: float = _PBE_MU):
-    """XC_GGA_X_PBE (or PBEsol with mu=10/81), unpolarized.
+def gga_x_pbe(rho: torch.Tensor, sigma: torch.Tensor, mu: float = _PBE_MU,
+              kappa: float = _PBE_KAPPA, rpbe: bool = False):
+    """XC_GGA_X_PBE family, unpolarized: PBE (default), PBEsol
+    (mu=10/81), revPBE (kappa=1.245, XC_GGA_X_PBE_R) and RPBE
+    (Hammer-Hansen-Norskov exponential form, XC_GGA_X_RPBE).
     Returns (eps, vrho, vsigma)."""
     rho = _safe_rho(rho)
     sigma = torch.clamp(sigma, min=1e-40)
     kf = (3.0 * math.pi**2 * rho) ** _THIRD
     # s^2 = sigma / (2 kf rho)^2
     s2 = sigma / (2.0 * kf * rho) ** 2
-    kappa = _PBE_KAPPA
-    fdenom = 1.0 + mu * s2 / kappa
-    fx = 1.0 + kappa - kappa / fdenom
-    dfx_ds2 = mu / fdenom**2
+    if rpbe:
+        ex = torch.exp(-mu * s2 / kappa)
+        fx = 1.0 + kappa * (1.0 - ex)
+        dfx_ds2 = mu * ex
+    else:
+        fdenom = 1.0 + mu * s2 / kappa
+        fx = 1.0 + kappa - kappa / fdenom
+        dfx_ds2 = mu / fdenom**2
     cx = (3.0 / 4.0) * (3.0 / math.pi) ** _THIRD
     eps_unif = -cx * rho ** _THIRD
     eps = eps_unif * fx
@@ -312,14 +319,17 @@ def lda_c_pw_spin(ru: torch.Tensor, rd: torch.Tensor):
     return eps, vu, vd
 
 
-def gga_x_pbe_spin(ru, rd, s_uu, s_dd, mu: float = _PBE_MU):
-    """PBE exchange via exact spin scaling. Returns
+def gga_x_pbe_spin(ru, rd, s_uu, s_dd, mu: float = _PBE_MU,
+                   kappa: float = _PBE_KAPPA, rpbe: bool = False):
+    """PBE-family exchange via exact spin scaling. Returns
     (eps, vu, vd, vs_uu, vs_dd); vsigma_ud = 0 for exchange."""
     ru = _safe_rho(ru)
     rd = _safe_rho(rd)
     n = ru + rd
-    e_u, v_u, vs_u = gga_x_pbe(2.0 * ru, 4.0 * s_uu, mu=mu)
-    e_d, v_d, vs_d = gga_x_pbe(2.0 * rd, 4.0 * s_dd, mu=mu)
+    e_u, v_u, vs_u = gga_x_pbe(2.0 * ru, 4.0 * s_uu, mu=mu, kappa=kappa,
+                               rpbe=rpbe)
+    e_d, v_d, vs_d = gga_x_pbe(2.0 * rd, 4.0 * s_dd, mu=mu, kappa=kappa,
+                               rpbe=rpbe)
     eps = (ru * e_u + rd * e_d) / n
     return eps, v_u, v_d, 2.0 * vs_u, 2.0 * vs_d
 
@@ -409,8 +419,19 @@ def gga_c_pbesol(rho, sigma):
 
 _LDA = {"XC_LDA_X": lda_x, "XC_LDA_C_PZ": lda_c_pz, "XC_LDA_C_PW": lda_c_pw,
         "XC_LDA_C_VWN": lda_c_vwn}
+def gga_x_revpbe(rho, sigma):
+    """XC_GGA_X_PBE_R (revPBE, Zhang-Yang kappa=1.245)."""
+    return gga_x_pbe(rho, sigma, kappa=1.245)
+
+
+def gga_x_rpbe(rho, sigma):
+    """XC_GGA_X_RPBE (Hammer-Hansen-Norskov)."""
+    return gga_x_pbe(rho, sigma, rpbe=True)
+
+
 _GGA = {"XC_GGA_X_PBE": gga_x_pbe, "XC_GGA_C_PBE": gga_c_pbe,
-        "XC_GGA_X_PBE_SOL": gga_x_pbesol, "XC_GGA_C_PBE_SOL": gga_c_pbesol}
+        "XC_GGA_X_PBE_SOL": gga_x_pbesol, "XC_GGA_C_PBE_SOL": gga_c_pbesol,
+        "XC_GGA_X_PBE_R": gga_x_revpbe, "XC_GGA_X_RPBE": gga_x_rpbe}
 _LDA_SPIN = {"XC_LDA_X": lda_x_spin, "XC_LDA_C_PZ": lda_c_pz_spin,
              "XC_LDA_C_PW": lda_c_pw_spin, "XC_LDA_C_VWN": lda_c_vwn_spin}
 
@@ -433,9 +454,13 @@ def evaluate_spin(names: list[str], ru: torch.Tensor, rd: torch.Tensor,
             eps = eps + e
             vu = vu + a
             vd = vd + b
-        elif name in ("XC_GGA_X_PBE", "XC_GGA_X_PBE_SOL"):
-            mu = _PBE_MU if name == "XC_GGA_X_PBE" else _PBESOL_MU
-            e, a, b, su, sd = gga_x_pbe_spin(ru, rd, s_uu, s_dd, mu=mu)
+        elif name in ("XC_GGA_X_PBE", "XC_GGA_X_PBE_SOL",
+                      "XC_GGA_X_PBE_R", "XC_GGA_X_RPBE"):
+            mu = _PBESOL_MU if name == "XC_GGA_X_PBE_SOL" else _PBE_MU
+            kap = 1.245 if name == "XC_GGA_X_PBE_R" else _PBE_KAPPA
+            e, a, b, su, sd = gga_x_pbe_spin(
+                ru, rd, s_uu, s_dd, mu=mu, kappa=kap,
+                rpbe=name == "XC_GGA_X_RPBE")
             eps = eps + e
             vu = vu + a
             vd = vd + b

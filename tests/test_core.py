@@ -346,3 +346,28 @@ def test_radial_solver_scalar_relativistic():
     c = 137.035999084
     e_dirac = c * c * (np.sqrt(1 - (30.0 / c) ** 2) - 1.0)
     assert abs(e30 - e_dirac) / abs(e_dirac) < 1e-3, (e30, e_dirac)
+
+
+def test_xc_revpbe_rpbe_derivatives():
+    """revPBE / RPBE exchange: finite-difference check of vrho/vsigma and
+    exact spin-scaling consistency at zeta=0."""
+    import torch
+    from sirius_amd import xc
+
+    rho = torch.tensor([0.2, 1.0, 3.0], dtype=torch.float64)
+    sig = torch.tensor([0.01, 0.5, 4.0], dtype=torch.float64)
+    h = 1e-6
+    for name in ("XC_GGA_X_PBE_R", "XC_GGA_X_RPBE"):
+        e, v, vs = xc.evaluate([name], rho, sig)
+        ep, _, _ = xc.evaluate([name], rho + h, sig)
+        em, _, _ = xc.evaluate([name], rho - h, sig)
+        num = ((rho + h) * ep - (rho - h) * em) / (2 * h)
+        assert torch.allclose(num, v, atol=1e-5)
+        es1, _, _ = xc.evaluate([name], rho, sig + h)
+        es2, _, _ = xc.evaluate([name], rho, sig - h)
+        assert torch.allclose(rho * (es1 - es2) / (2 * h), vs, atol=1e-6)
+        e0, v0, _ = xc.evaluate([name], rho, sig)
+        es_, vu, vd, *_ = xc.evaluate_spin([name], rho / 2, rho / 2,
+                                           sig / 4, sig / 4, sig)
+        assert torch.allclose(e0, es_, atol=1e-12)
+        assert torch.allclose(v0, vu, atol=1e-12)
