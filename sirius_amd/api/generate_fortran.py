@@ -271,6 +271,47 @@ API = [
       ("restart", "int", "in"), ("processing_unit", "string", "in"),
       ("converged", "bool_out", "out")],
      "Direct minimization with explicit parameters."),
+    ("sirius_start_timer", [("name", "string", "in")], "Start a timer."),
+    ("sirius_stop_timer", [("name", "string", "in")], "Stop a timer."),
+    ("sirius_serialize_timers", [("fname", "string", "in")],
+     "Write timer statistics to a JSON file."),
+    ("sirius_update_context", [("handler", "handler", "in")],
+     "Re-initialize the context from the accumulated config."),
+    ("sirius_dump_runtime_setup",
+     [("handler", "handler", "in"), ("filename", "string", "in")],
+     "Dump the merged runtime configuration to a JSON file."),
+    ("sirius_get_kp_params_from_ctx",
+     [("handler", "handler", "in"), ("k_grid", "int(3)_out", "out"),
+      ("k_shift", "int(3)_out", "out"), ("use_symmetry", "bool_out", "out")],
+     "k-grid parameters stored in the context."),
+    ("sirius_get_scf_params_from_ctx",
+     [("handler", "handler", "in"), ("density_tol", "double_out", "out"),
+      ("energy_tol", "double_out", "out"),
+      ("iter_solver_tol", "double_out", "out"),
+      ("max_niter", "int_out", "out")],
+     "SCF convergence parameters stored in the context."),
+    ("sirius_fft_transform",
+     [("gs_handler", "handler", "in"), ("label", "string", "in"),
+      ("direction", "int", "in")],
+     "Transform a named field between PW and real grid."),
+    ("sirius_generate_coulomb_potential",
+     [("gs_handler", "handler", "in"), ("vh_el", "double(*)_out", "out")],
+     "Generate the Coulomb potential from the current density."),
+    ("sirius_generate_xc_potential", [("gs_handler", "handler", "in")],
+     "Generate the XC potential from the current density."),
+    ("sirius_get_rg_values",
+     [("gs_handler", "handler", "in"), ("label", "string", "in"),
+      ("grid_dims", "int(3)", "in"), ("local_box_origin", "int(3)", "in"),
+      ("local_box_size", "int(3)", "in"), ("fcomm", "int", "in"),
+      ("values", "double(*)_out", "out"),
+      ("transform_to_rg", "bool", "in")],
+     "Real-grid values of a named field inside a box."),
+    ("sirius_set_rg_values",
+     [("gs_handler", "handler", "in"), ("label", "string", "in"),
+      ("grid_dims", "int(3)", "in"), ("local_box_origin", "int(3)", "in"),
+      ("local_box_size", "int(3)", "in"), ("fcomm", "int", "in"),
+      ("values", "double(*)", "in"), ("transform_to_pw", "bool", "in")],
+     "Set real-grid values of a named field inside a box."),
 ]
 
 FT = {

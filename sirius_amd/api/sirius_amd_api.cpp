@@ -994,7 +994,268 @@ void sirius_nlcg_params(void* const* gs_handler, void* const* ks_handler,
     set_err(error_code, 0);
 }
 
+
+void sirius_start_timer(char const* name, int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("start_timer", Py_BuildValue("(s)", name));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_stop_timer(char const* name, int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("stop_timer", Py_BuildValue("(s)", name));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_serialize_timers(char const* fname, int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("serialize_timers", Py_BuildValue("(s)", fname));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_set_parameters(
+    void* const* handler, int const* lmax_apw, int const* lmax_rho,
+    int const* lmax_pot, int const* num_fv_states, int const* num_bands,
+    int const* num_mag_dims, double const* pw_cutoff,
+    double const* gk_cutoff, int const* fft_grid_size, int const* auto_rmt,
+    bool const* gamma_point, bool const* use_symmetry,
+    bool const* so_correction, char const* valence_rel,
+    char const* core_rel, double const* iter_solver_tol_empty,
+    char const* iter_solver_type, int const* verbosity,
+    bool const* hubbard_correction, int const* hubbard_correction_kind,
+    bool const* hubbard_full_orthogonalization,
+    bool const* hubbard_constrained_calculation,
+    char const* hubbard_orbitals, int const* sht_coverage,
+    double const* min_occupancy, char const* smearing,
+    double const* smearing_width, double const* spglib_tol,
+    char const* electronic_structure_method, int* error_code) {
+    Gil g;
+    PyObject* d = PyDict_New();
+    auto seti = [&](const char* k, int const* v) {
+        if (v) PyDict_SetItemString(d, k, PyLong_FromLong(*v));
+    };
+    auto setd = [&](const char* k, double const* v) {
+        if (v) PyDict_SetItemString(d, k, PyFloat_FromDouble(*v));
+    };
+    auto setb = [&](const char* k, bool const* v) {
+        if (v) PyDict_SetItemString(d, k, PyBool_FromLong(*v));
+    };
+    auto sets = [&](const char* k, char const* v) {
+        if (v && v[0]) PyDict_SetItemString(d, k, PyUnicode_FromString(v));
+    };
+    seti("lmax_apw", lmax_apw);
+    seti("lmax_rho", lmax_rho);
+    seti("lmax_pot", lmax_pot);
+    seti("num_fv_states", num_fv_states);
+    seti("num_bands", num_bands);
+    seti("num_mag_dims", num_mag_dims);
+    setd("pw_cutoff", pw_cutoff);
+    setd("gk_cutoff", gk_cutoff);
+    if (fft_grid_size)
+        PyDict_SetItemString(d, "fft_grid_size",
+                             list_from_ints(fft_grid_size, 3));
+    seti("auto_rmt", auto_rmt);
+    setb("gamma_point", gamma_point);
+    setb("use_symmetry", use_symmetry);
+    setb("so_correction", so_correction);
+    sets("valence_rel", valence_rel);
+    sets("core_rel", core_rel);
+    setd("iter_solver_tol_empty", iter_solver_tol_empty);
+    sets("iter_solver_type", iter_solver_type);
+    seti("verbosity", verbosity);
+    setb("hubbard_correction", hubbard_correction);
+    seti("hubbard_correction_kind", hubbard_correction_kind);
+    setb("hubbard_full_orthogonalization", hubbard_full_orthogonalization);
+    setb("hubbard_constrained_calculation", hubbard_constrained_calculation);
+    sets("hubbard_orbitals", hubbard_orbitals);
+    seti("sht_coverage", sht_coverage);
+    setd("min_occupancy", min_occupancy);
+    sets("smearing", smearing);
+    setd("smearing_width", smearing_width);
+    setd("spglib_tol", spglib_tol);
+    sets("electronic_structure_method", electronic_structure_method);
+    PyObject* r = call_impl("set_parameters", Py_BuildValue(
+        "(ON)", (PyObject*)*handler, d));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_update_context(void* const* handler, int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("update_context",
+                            Py_BuildValue("(O)", (PyObject*)*handler));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_option_set(void* const* handler, char const* section,
+                       char const* name, int const* type,
+                       void const* data_ptr, int const* max_length,
+                       bool const* append, int* error_code) {
+    Gil g;
+    PyObject* v = nullptr;
+    int ty = type ? *type : 0;
+    int n = max_length ? *max_length : 1;
+    switch (ty) {
+        case 1: v = PyLong_FromLong(*(int const*)data_ptr); break;
+        case 2: v = PyBool_FromLong(*(bool const*)data_ptr); break;
+        case 3: v = PyUnicode_FromString((char const*)data_ptr); break;
+        case 4: v = PyFloat_FromDouble(*(double const*)data_ptr); break;
+        case 7: v = list_from_ints((int const*)data_ptr, n); break;
+        case 9: v = list_from_doubles((double const*)data_ptr, n); break;
+        default: set_err(error_code, 2); return;
+    }
+    PyObject* r = call_impl("option_set", Py_BuildValue(
+        "(OssNi)", (PyObject*)*handler, section, name, v,
+        append ? (int)*append : 0));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_dump_runtime_setup(void* const* handler, char* filename,
+                               int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("dump_runtime_setup", Py_BuildValue(
+        "(Os)", (PyObject*)*handler, filename));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_get_kp_params_from_ctx(void* const* handler, int* k_grid,
+                                   int* k_shift, bool* use_symmetry,
+                                   int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("get_kp_params_from_ctx",
+                            Py_BuildValue("(O)", (PyObject*)*handler));
+    if (!r) { set_err(error_code, 1); return; }
+    PyObject* kg = PyTuple_GetItem(r, 0);
+    PyObject* ks = PyTuple_GetItem(r, 1);
+    for (int i = 0; i < 3; i++) {
+        PyObject* a = PySequence_GetItem(kg, i);
+        PyObject* b = PySequence_GetItem(ks, i);
+        k_grid[i] = (int)PyLong_AsLong(a);
+        k_shift[i] = (int)PyLong_AsLong(b);
+        Py_DECREF(a);
+        Py_DECREF(b);
+    }
+    *use_symmetry = PyObject_IsTrue(PyTuple_GetItem(r, 2));
+    Py_DECREF(r);
+    set_err(error_code, 0);
+}
+
+void sirius_get_scf_params_from_ctx(void* const* handler,
+                                    double* density_tol, double* energy_tol,
+                                    double* iter_solver_tol, int* max_niter,
+                                    int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("get_scf_params_from_ctx",
+                            Py_BuildValue("(O)", (PyObject*)*handler));
+    if (!r) { set_err(error_code, 1); return; }
+    *density_tol = PyFloat_AsDouble(PyTuple_GetItem(r, 0));
+    *energy_tol = PyFloat_AsDouble(PyTuple_GetItem(r, 1));
+    *iter_solver_tol = PyFloat_AsDouble(PyTuple_GetItem(r, 2));
+    *max_niter = (int)PyLong_AsLong(PyTuple_GetItem(r, 3));
+    Py_DECREF(r);
+    set_err(error_code, 0);
+}
+
+void sirius_get_pw_coeffs(void* const* gs_handler, char const* label,
+                          void* pw_coeffs, int const* ngv, int* gvl,
+                          int const* comm, int* error_code) {
+    Gil g;
+    (void)comm;
+    int n = ngv ? *ngv : 0;
+    PyObject* r = call_impl("get_pw_coeffs", Py_BuildValue(
+        "(OsN)", (PyObject*)*gs_handler, label, list_from_ints(gvl, 3 * n)));
+    if (!r) { set_err(error_code, 1); return; }
+    doubles_from_seq(r, (double*)pw_coeffs);
+    Py_DECREF(r);
+    set_err(error_code, 0);
+}
+
+void sirius_set_pw_coeffs(void* const* gs_handler, char const* label,
+                          void const* pw_coeffs, bool const* transform_to_rg,
+                          int const* ngv, int* gvl, int const* comm,
+                          int* error_code) {
+    Gil g;
+    (void)comm;
+    int n = ngv ? *ngv : 0;
+    PyObject* r = call_impl("set_pw_coeffs", Py_BuildValue(
+        "(OsNNi)", (PyObject*)*gs_handler, label,
+        list_from_doubles((double const*)pw_coeffs, 2 * n),
+        list_from_ints(gvl, 3 * n),
+        transform_to_rg ? (int)*transform_to_rg : 0));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_fft_transform(void* const* gs_handler, char const* label,
+                          int* direction, int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("fft_transform", Py_BuildValue(
+        "(Osi)", (PyObject*)*gs_handler, label, *direction));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_get_rg_values(void* const* gs_handler, char const* label,
+                          int const* grid_dims, int const* local_box_origin,
+                          int const* local_box_size, int const* fcomm,
+                          double* values, bool const* transform_to_rg,
+                          int* error_code) {
+    Gil g;
+    (void)grid_dims; (void)fcomm; (void)transform_to_rg;
+    PyObject* r = call_impl("get_rg_values", Py_BuildValue(
+        "(OsNN)", (PyObject*)*gs_handler, label,
+        list_from_ints(local_box_origin, 3),
+        list_from_ints(local_box_size, 3)));
+    if (!r) { set_err(error_code, 1); return; }
+    doubles_from_seq(r, values);
+    Py_DECREF(r);
+    set_err(error_code, 0);
+}
+
+void sirius_set_rg_values(void* const* gs_handler, char const* label,
+                          int const* grid_dims, int const* local_box_origin,
+                          int const* local_box_size, int const* fcomm,
+                          double const* values, bool const* transform_to_pw,
+                          int* error_code) {
+    Gil g;
+    (void)grid_dims; (void)fcomm;
+    int n = local_box_size[0] * local_box_size[1] * local_box_size[2];
+    PyObject* r = call_impl("set_rg_values", Py_BuildValue(
+        "(OsNNNi)", (PyObject*)*gs_handler, label,
+        list_from_ints(local_box_origin, 3),
+        list_from_ints(local_box_size, 3), list_from_doubles(values, n),
+        transform_to_pw ? (int)*transform_to_pw : 0));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_generate_coulomb_potential(void* const* gs_handler,
+                                       double* vh_el, int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("generate_coulomb_potential",
+                            Py_BuildValue("(O)", (PyObject*)*gs_handler));
+    if (!r) { set_err(error_code, 1); return; }
+    if (vh_el) doubles_from_seq(r, vh_el);
+    Py_DECREF(r);
+    set_err(error_code, 0);
+}
+
+void sirius_generate_xc_potential(void* const* gs_handler, int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("generate_xc_potential",
+                            Py_BuildValue("(O)", (PyObject*)*gs_handler));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
 }  // extern "C"
+
 
 
 

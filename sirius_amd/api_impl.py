@@ -451,7 +451,9 @@ def get_periodic_function(gs: GsHandle, label: str):
         "veff": lambda: dft.potential.veff_r,
         "bz": lambda: dft.potential.bz_r,
         "magz": lambda: dft.density.mag_r,
-        "vha": lambda: dft.potential.vha_r,
+        "vha": lambda: dft.ctx.fft_fine.to_real(
+            dft.potential.vha_g).real
+        if getattr(dft.potential, "vha_g", None) is not None else None,
         "vxc": lambda: dft.potential.vxc_r,
         "exc": lambda: dft.potential.exc_r,
     }
@@ -684,3 +686,236 @@ def nlcg(gs: GsHandle, ks: KsetHandle, temp: float = -1.0,
                  "converged": bool(res["converged"]),
                  "num_scf_iterations": int(res["num_iter"])}
     return bool(res["converged"])
+
+
+# ---- timers, parameters, field access (batch 3) ------------------------
+
+_timers: dict = {}
+
+
+def start_timer(name: str):
+    import time
+
+    _timers.setdefault(name, {"total": 0.0, "count": 0})["start"] = \
+        time.time()
+
+
+def stop_timer(name: str):
+    import time
+
+    t = _timers.get(name)
+    if t and "start" in t:
+        t["total"] += time.time() - t.pop("start")
+        t["count"] += 1
+
+
+def serialize_timers(fname: str):
+    with open(fname, "w") as f:
+        json.dump({k: {kk: vv for kk, vv in v.items() if kk != "start"}
+                   for k, v in _timers.items()}, f, indent=1)
+
+
+def set_parameters(h: CtxHandle, params: dict):
+    """Scalar parameter batch (reference sirius_set_parameters; only the
+    non-null pointers arrive here)."""
+    p = h.cfg_data.setdefault("parameters", {})
+    simple = {"lmax_apw", "lmax_rho", "lmax_pot", "num_fv_states",
+              "num_bands", "num_mag_dims", "pw_cutoff", "gk_cutoff",
+              "auto_rmt", "gamma_point", "use_symmetry", "so_correction",
+              "hubbard_correction", "smearing", "smearing_width"}
+    for k, v in params.items():
+        if k in simple:
+            p[k] = v
+        elif k == "valence_rel":
+            p["valence_relativity"] = v
+        elif k == "core_rel":
+            p["core_relativity"] = v
+        elif k == "verbosity":
+            h.cfg_data.setdefault("control", {})["verbosity"] = v
+        elif k == "iter_solver_type":
+            h.cfg_data.setdefault("iterative_solver", {})["type"] = v
+        elif k == "iter_solver_tol_empty":
+            h.cfg_data.setdefault("iterative_solver", {})[
+                "empty_states_tolerance"] = v
+        elif k == "fft_grid_size":
+            h.cfg_data.setdefault("settings", {})["fft_grid_size"] = v
+        elif k == "hubbard_full_orthogonalization":
+            h.cfg_data.setdefault("hubbard", {})[
+                "hubbard_subspace_method"] = "full_orthogonalization" \
+                if v else "none"
+        elif k == "hubbard_constrained_calculation":
+            h.cfg_data.setdefault("hubbard", {})[
+                "constrained_calculation"] = v
+        # unknown keys are kept for introspection
+        else:
+            p[k] = v
+
+
+def update_context(h: CtxHandle):
+    """Re-initialize the simulation context from the accumulated config
+    (reference sirius_update_context)."""
+    return initialize_context(h)
+
+
+def option_set(h: CtxHandle, section: str, name: str, value, append=False):
+    sec = h.cfg_data.setdefault(section.lower(), {})
+    if append and isinstance(sec.get(name), list):
+        sec[name].append(value)
+    else:
+        sec[name] = value
+
+
+def dump_runtime_setup(h: CtxHandle, fname: str):
+    data = h.ctx.cfg.to_dict() if h.ctx is not None else h.cfg_data
+    with open(fname, "w") as f:
+        json.dump(data, f, indent=1)
+
+
+def get_kp_params_from_ctx(h: CtxHandle):
+    p = h.ctx.cfg.parameters
+    return list(p.ngridk), list(p.shiftk), bool(p.use_symmetry)
+
+
+def get_scf_params_from_ctx(h: CtxHandle):
+    p = h.ctx.cfg.parameters
+    tol = h.ctx.cfg.iterative_solver.get("energy_tolerance", 1e-2)
+    return (float(p.density_tol), float(p.energy_tol), float(tol),
+            int(p.num_dft_iter))
+
+
+def _named_pw_field(dft, label: str):
+    m = {"rho": ("density", "rho_g"), "veff": ("potential", "veff_g"),
+         "magz": ("density", "mag_g"), "bz": ("potential", "bz_g"),
+         "rhoc": ("density", "rho_core_g"), "vloc": ("potential", "vloc_g")}
+    if label not in m:
+        raise ValueError(f"unknown pw field label: {label}")
+    owner, attr = m[label]
+    return getattr(dft, owner), attr
+
+
+def get_pw_coeffs(gs: GsHandle, label: str, gvl):
+    """PW coefficients of a named field at the caller's Miller indices
+    (reference sirius_get_pw_coeffs; serial comm).  gvl is a flat list of
+    3*ngv ints; returns interleaved re/im."""
+    dft = gs.dft
+    owner, attr = _named_pw_field(dft, label)
+    f = getattr(owner, attr)
+    if f is None:
+        raise ValueError(f"field {label} not present")
+    fine = dft.ctx.gvec_fine
+    key = {tuple(mm): i for i, mm in enumerate(fine.miller)}
+    g = np.asarray(gvl, dtype=np.int64).reshape(-1, 3)
+    fc = f.detach().cpu().numpy()
+    out = np.zeros(2 * len(g))
+    for i, mm in enumerate(g):
+        j = key.get(tuple(mm))
+        if j is not None:
+            out[2 * i] = fc[j].real
+            out[2 * i + 1] = fc[j].imag
+    return out.tolist()
+
+
+def set_pw_coeffs(gs: GsHandle, label: str, coeffs, gvl,
+                  transform_to_rg: bool = False):
+    import torch
+
+    dft = gs.dft
+    owner, attr = _named_pw_field(dft, label)
+    fine = dft.ctx.gvec_fine
+    key = {tuple(mm): i for i, mm in enumerate(fine.miller)}
+    g = np.asarray(gvl, dtype=np.int64).reshape(-1, 3)
+    c = np.asarray(coeffs, dtype=np.float64)
+    f = torch.zeros(fine.num_gvec, dtype=dft.ctx.dtype)
+    for i, mm in enumerate(g):
+        j = key.get(tuple(mm))
+        if j is not None:
+            f[j] = complex(c[2 * i], c[2 * i + 1])
+    f = f.to(dft.ctx.device)
+    setattr(owner, attr, f)
+    if transform_to_rg and label == "rho":
+        owner.rho_r = dft.ctx.fft_fine.to_real(f).real
+    elif transform_to_rg and label == "magz":
+        owner.mag_r = dft.ctx.fft_fine.to_real(f).real
+
+
+def fft_transform(gs: GsHandle, label: str, direction: int):
+    """direction +1: PW -> real grid; -1: real grid -> PW (reference
+    sirius_fft_transform)."""
+    dft = gs.dft
+    rg_attr = {"rho": "rho_r", "magz": "mag_r", "veff": "veff_r"}
+    pw_attr = {"rho": "rho_g", "magz": "mag_g", "veff": "veff_g"}
+    if label not in rg_attr:
+        raise ValueError(f"fft_transform: unknown label {label}")
+    owner = dft.density if label in ("rho", "magz") else dft.potential
+    if direction > 0:
+        setattr(owner, rg_attr[label], dft.ctx.fft_fine.to_real(
+            getattr(owner, pw_attr[label])).real)
+    else:
+        setattr(owner, pw_attr[label], dft.ctx.fft_fine.to_pw(
+            getattr(owner, rg_attr[label]).to(dft.ctx.dtype)))
+
+
+def _rg_field(dft, label: str):
+    m = {"rho": (dft.density, "rho_r"), "magz": (dft.density, "mag_r"),
+         "veff": (dft.potential, "veff_r"), "bz": (dft.potential, "bz_r"),
+         "vxc": (dft.potential, "vxc_r"), "exc": (dft.potential, "exc_r")}
+    if label not in m:
+        raise ValueError(f"unknown rg field label: {label}")
+    return m[label]
+
+
+def get_rg_values(gs: GsHandle, label: str, box_origin, box_size):
+    """Real-grid values inside a (1-based, Fortran-order) box."""
+    owner, attr = _rg_field(gs.dft, label)
+    t = getattr(owner, attr)
+    if t is None:
+        raise ValueError(f"field {label} not present")
+    a = t.detach().cpu().numpy()
+    o = [int(x) - 1 for x in box_origin]
+    s = [int(x) for x in box_size]
+    box = a[o[0]:o[0] + s[0], o[1]:o[1] + s[1], o[2]:o[2] + s[2]]
+    return np.asarray(box, dtype=np.float64).transpose(2, 1, 0) \
+        .reshape(-1).tolist()
+
+
+def set_rg_values(gs: GsHandle, label: str, box_origin, box_size, values,
+                  transform_to_pw: bool = False):
+    import torch
+
+    dft = gs.dft
+    owner, attr = _rg_field(dft, label)
+    t = getattr(owner, attr)
+    if t is None:
+        raise ValueError(f"field {label} not present")
+    o = [int(x) - 1 for x in box_origin]
+    s = [int(x) for x in box_size]
+    box = np.asarray(values, dtype=np.float64).reshape(
+        s[2], s[1], s[0]).transpose(2, 1, 0)
+    t[o[0]:o[0] + s[0], o[1]:o[1] + s[1], o[2]:o[2] + s[2]] = \
+        torch.from_numpy(box.copy()).to(t.device)
+    if transform_to_pw:
+        pw = {"rho": "rho_g", "magz": "mag_g", "veff": "veff_g"}.get(label)
+        if pw:
+            setattr(owner, pw, dft.ctx.fft_fine.to_pw(
+                getattr(owner, attr).to(dft.ctx.dtype)))
+
+
+def generate_coulomb_potential(gs: GsHandle):
+    """Regenerate the effective potential from the current density and
+    return vh_el per atom when the branch computes it (reference
+    sirius_generate_coulomb_potential; the PP branch has no MT spheres,
+    so vh_el is zeros there — documented deviation)."""
+    dft = gs.dft
+    dft.potential.generate(dft.density)
+    vh_el = getattr(dft.potential, "vh_el", None)
+    n = dft.ctx.unit_cell.num_atoms
+    if vh_el is None:
+        return [0.0] * n
+    return [float(x) for x in np.asarray(vh_el).reshape(-1)[:n]]
+
+
+def generate_xc_potential(gs: GsHandle):
+    """Regenerate the potential (includes the XC part; reference
+    sirius_generate_xc_potential regenerates only XC — here the full
+    potential is rebuilt, which subsumes it)."""
+    gs.dft.potential.generate(gs.dft.density)

@@ -318,3 +318,125 @@ def test_c_api_introspection(tmp_path):
 
     for hh in (gs, ks, h):
         lib.sirius_free_object_handler(ctypes.byref(hh), ctypes.byref(ec))
+
+
+def test_c_api_fields_and_params(tmp_path):
+    """Batch-3 surface: set_parameters, scf/kp params, pw-coeff and
+    rg-value round trips, fft_transform, timers."""
+    _build()
+    lib = ctypes.CDLL(LIB)
+    ec = ctypes.c_int(0)
+    t = ctypes.c_bool(False)
+    lib.sirius_initialize(ctypes.byref(t), ctypes.byref(ec))
+
+    lib.sirius_start_timer(b"outer", ctypes.byref(ec))
+    assert ec.value == 0
+
+    h = ctypes.c_void_p()
+    fk = ctypes.c_int()
+    fb = ctypes.c_int()
+    lib.sirius_create_context(0, ctypes.byref(h), ctypes.byref(fk),
+                              ctypes.byref(fb), ctypes.byref(ec))
+    deck = json.load(open("verification/test08/sirius.json"))
+    base = os.path.abspath("verification/test08")
+    for k, v in deck["unit_cell"]["atom_files"].items():
+        deck["unit_cell"]["atom_files"][k] = os.path.join(base, v)
+    lib.sirius_import_parameters(ctypes.byref(h), json.dumps(deck).encode(),
+                                 ctypes.byref(ec))
+    # override one scalar through sirius_set_parameters (nullable args)
+    nb = ctypes.c_int(20)
+    lib.sirius_set_parameters(ctypes.byref(h), None, None, None, None,
+                              ctypes.byref(nb), None, None, None, None,
+                              None, None, None, None, None, None, None,
+                              None, None, None, None, None, None, None,
+                              None, None, None, None, None, None,
+                              ctypes.byref(ec))
+    assert ec.value == 0
+    lib.sirius_initialize_context(ctypes.byref(h), ctypes.byref(ec))
+    assert ec.value == 0
+
+    kg = (ctypes.c_int * 3)()
+    ks_ = (ctypes.c_int * 3)()
+    us = ctypes.c_bool(False)
+    lib.sirius_get_kp_params_from_ctx(ctypes.byref(h), kg, ks_,
+                                      ctypes.byref(us), ctypes.byref(ec))
+    assert ec.value == 0 and list(kg) == list(deck["parameters"]["ngridk"])
+    dtol = ctypes.c_double()
+    etol = ctypes.c_double()
+    itol = ctypes.c_double()
+    mx = ctypes.c_int()
+    lib.sirius_get_scf_params_from_ctx(ctypes.byref(h), ctypes.byref(dtol),
+                                       ctypes.byref(etol), ctypes.byref(itol),
+                                       ctypes.byref(mx), ctypes.byref(ec))
+    assert ec.value == 0 and dtol.value > 0
+
+    grid = (ctypes.c_int * 3)(1, 1, 1)
+    shift = (ctypes.c_int * 3)(0, 0, 0)
+    usym = ctypes.c_bool(False)
+    ks = ctypes.c_void_p()
+    lib.sirius_create_kset_from_grid(ctypes.byref(h), grid, shift,
+                                     ctypes.byref(usym), ctypes.byref(ks),
+                                     ctypes.byref(ec))
+    gs = ctypes.c_void_p()
+    lib.sirius_create_ground_state(ctypes.byref(ks), ctypes.byref(gs),
+                                   ctypes.byref(ec))
+    lib.sirius_generate_initial_density(ctypes.byref(gs), ctypes.byref(ec))
+    assert ec.value == 0
+    vh = (ctypes.c_double * 2)()
+    lib.sirius_generate_coulomb_potential(ctypes.byref(gs), vh,
+                                          ctypes.byref(ec))
+    assert ec.value == 0
+    lib.sirius_generate_xc_potential(ctypes.byref(gs), ctypes.byref(ec))
+    assert ec.value == 0
+
+    # pw-coeff round trip on rho at the first 10 G vectors
+    from sirius_amd import api_impl
+    gs_py = ctypes.cast(gs, ctypes.py_object).value
+    fine = gs_py.dft.ctx.gvec_fine
+    n = 10
+    gvl = (ctypes.c_int * (3 * n))(*fine.miller[:n].reshape(-1).tolist())
+    ngv = ctypes.c_int(n)
+    cz = ctypes.c_int(0)
+    buf = (ctypes.c_double * (2 * n))()
+    lib.sirius_get_pw_coeffs(ctypes.byref(gs), b"rho", buf,
+                             ctypes.byref(ngv), gvl, ctypes.byref(cz),
+                             ctypes.byref(ec))
+    assert ec.value == 0
+    import numpy as _np
+    rho0 = gs_py.dft.density.rho_g[:n].cpu().numpy()
+    got = _np.array(buf[0::2]) + 1j * _np.array(buf[1::2])
+    assert _np.abs(got - rho0).max() < 1e-12
+
+    # rg-value box read matches the tensor
+    d1, d2, d3 = gs_py.dft.ctx.fft_fine.dims
+    org = (ctypes.c_int * 3)(1, 1, 1)
+    sz = (ctypes.c_int * 3)(2, 2, 2)
+    gd = (ctypes.c_int * 3)(d1, d2, d3)
+    vals = (ctypes.c_double * 8)()
+    tb = ctypes.c_bool(False)
+    lib.sirius_get_rg_values(ctypes.byref(gs), b"rho", gd, org, sz,
+                             ctypes.byref(cz), vals, ctypes.byref(tb),
+                             ctypes.byref(ec))
+    assert ec.value == 0
+    ref = gs_py.dft.density.rho_r[:2, :2, :2].cpu().numpy()
+    got = _np.array(vals).reshape(2, 2, 2).transpose(2, 1, 0)
+    assert _np.abs(got - ref).max() < 1e-12
+
+    # fft_transform -1 then +1 reproduces rho_r
+    before = gs_py.dft.density.rho_r.clone()
+    dm1 = ctypes.c_int(-1)
+    dp1 = ctypes.c_int(1)
+    lib.sirius_fft_transform(ctypes.byref(gs), b"rho", ctypes.byref(dm1),
+                             ctypes.byref(ec))
+    lib.sirius_fft_transform(ctypes.byref(gs), b"rho", ctypes.byref(dp1),
+                             ctypes.byref(ec))
+    assert ec.value == 0
+    assert float((gs_py.dft.density.rho_r - before).abs().max()) < 1e-10
+
+    lib.sirius_stop_timer(b"outer", ctypes.byref(ec))
+    p = str(tmp_path / "timers.json").encode()
+    lib.sirius_serialize_timers(p, ctypes.byref(ec))
+    assert ec.value == 0 and os.path.exists(p)
+
+    for hh in (gs, ks, h):
+        lib.sirius_free_object_handler(ctypes.byref(hh), ctypes.byref(ec))
