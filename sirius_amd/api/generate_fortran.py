@@ -312,6 +312,37 @@ API = [
       ("local_box_size", "int(3)", "in"), ("fcomm", "int", "in"),
       ("values", "double(*)", "in"), ("transform_to_pw", "bool", "in")],
      "Set real-grid values of a named field inside a box."),
+    ("sirius_set_atom_type_hubbard",
+     [("handler", "handler", "in"), ("label", "string", "in"),
+      ("l", "int", "in"), ("n", "int", "in"), ("occ", "double", "in"),
+      ("U", "double", "in"), ("J", "double", "in"),
+      ("alpha", "double", "in"), ("beta", "double", "in"),
+      ("J0", "double", "in")],
+     "Add a Hubbard-corrected orbital to an atom type."),
+    ("sirius_add_hubbard_atom_pair",
+     [("handler", "handler", "in"), ("atom_pair", "int(*)", "in"),
+      ("translation", "int(3)", "in"), ("n", "int(*)", "in"),
+      ("l", "int(*)", "in"), ("coupling", "double", "in")],
+     "Add an inter-site Hubbard V pair."),
+    ("sirius_get_kpoint_inner_comm",
+     [("handler", "handler", "in"), ("fcomm", "int_out", "out")],
+     "Communicator handle inside a k-group (serial embedding: 0)."),
+    ("sirius_get_kpoint_inter_comm",
+     [("handler", "handler", "in"), ("fcomm", "int_out", "out")],
+     "Communicator handle across k-groups (serial embedding: 0)."),
+    ("sirius_get_fft_comm",
+     [("handler", "handler", "in"), ("fcomm", "int_out", "out")],
+     "FFT communicator handle (serial embedding: 0)."),
+    ("sirius_set_energy_fermi",
+     [("ks_handler", "handler", "in"), ("energy_fermi", "double", "in")],
+     "Set the Fermi energy."),
+    ("sirius_check_scf_density", [("gs_handler", "handler", "in")],
+     "Regenerate rho from the wave functions and report max |drho(G)|."),
+    ("sirius_get_step_function",
+     [("handler", "handler", "in"), ("cfunig", "double(*)_out", "out"),
+      ("cfunrg", "double(*)_out", "out"),
+      ("num_rg_points", "int_out", "out")],
+     "LAPW unit-step function (PW re/im interleaved + real grid)."),
 ]
 
 FT = {

@@ -1254,7 +1254,157 @@ void sirius_generate_xc_potential(void* const* gs_handler, int* error_code) {
     Py_XDECREF(r);
 }
 
+
+void sirius_set_atom_type_hubbard(void* const* handler, char const* label,
+                                  int const* l, int const* n,
+                                  double const* occ, double const* U,
+                                  double const* J, double const* alpha,
+                                  double const* beta, double const* J0,
+                                  int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("set_atom_type_hubbard", Py_BuildValue(
+        "(Osiidddddd)", (PyObject*)*handler, label, *l, *n,
+        occ ? *occ : 0.0, U ? *U : 0.0, J ? *J : 0.0,
+        alpha ? *alpha : 0.0, beta ? *beta : 0.0, J0 ? *J0 : 0.0));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_add_hubbard_atom_pair(void* const* handler, int* const atom_pair,
+                                  int* const translation, int* const n,
+                                  int* const l, const double* const coupling,
+                                  int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("add_hubbard_atom_pair", Py_BuildValue(
+        "(ONNNNd)", (PyObject*)*handler, list_from_ints(atom_pair, 2),
+        list_from_ints(translation, 3), list_from_ints(n, 2),
+        list_from_ints(l, 2), *coupling));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_add_hubbard_atom_constraint(void* const* handler,
+                                        int* const atom_id, int* const n,
+                                        int* const l, int* const lmax_at,
+                                        const double* const occ,
+                                        int* const orbital_order,
+                                        int* const error_code) {
+    Gil g;
+    int mm = 2 * (*l) + 1;
+    // the occupancy block is (nsp, mm, mm); nsp inferred Python-side —
+    // pass the larger collinear size, extra values are ignored there
+    PyObject* occl = list_from_doubles(occ, 2 * mm * mm);
+    PyObject* ord = orbital_order ? list_from_ints(orbital_order, mm)
+                                  : (Py_INCREF(Py_None), Py_None);
+    PyObject* r = call_impl("add_hubbard_atom_constraint", Py_BuildValue(
+        "(OiiiiNN)", (PyObject*)*handler, *atom_id, *n, *l,
+        lmax_at ? *lmax_at : mm, occl, ord));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_get_kpoint_inner_comm(void* const* handler, int* fcomm,
+                                  int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("get_comm_handle", Py_BuildValue(
+        "(Os)", (PyObject*)*handler, "inner"));
+    if (!r) { set_err(error_code, 1); return; }
+    *fcomm = (int)PyLong_AsLong(r);
+    Py_DECREF(r);
+    set_err(error_code, 0);
+}
+
+void sirius_get_kpoint_inter_comm(void* const* handler, int* fcomm,
+                                  int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("get_comm_handle", Py_BuildValue(
+        "(Os)", (PyObject*)*handler, "inter"));
+    if (!r) { set_err(error_code, 1); return; }
+    *fcomm = (int)PyLong_AsLong(r);
+    Py_DECREF(r);
+    set_err(error_code, 0);
+}
+
+void sirius_get_fft_comm(void* const* handler, int* fcomm, int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("get_comm_handle", Py_BuildValue(
+        "(Os)", (PyObject*)*handler, "fft"));
+    if (!r) { set_err(error_code, 1); return; }
+    *fcomm = (int)PyLong_AsLong(r);
+    Py_DECREF(r);
+    set_err(error_code, 0);
+}
+
+void sirius_set_energy_fermi(void* const* ks_handler, double* energy_fermi,
+                             int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("set_energy_fermi", Py_BuildValue(
+        "(Od)", (PyObject*)*ks_handler, *energy_fermi));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_check_scf_density(void* const* gs_handler, int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("check_scf_density",
+                            Py_BuildValue("(O)", (PyObject*)*gs_handler));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_get_step_function(void* const* handler, void* cfunig,
+                              double* cfunrg, int* num_rg_points,
+                              int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("get_step_function",
+                            Py_BuildValue("(O)", (PyObject*)*handler));
+    if (!r) { set_err(error_code, 1); return; }
+    if (cfunig) doubles_from_seq(PyTuple_GetItem(r, 0), (double*)cfunig);
+    if (cfunrg) doubles_from_seq(PyTuple_GetItem(r, 1), cfunrg);
+    if (num_rg_points)
+        *num_rg_points = (int)PyLong_AsLong(PyTuple_GetItem(r, 2));
+    Py_DECREF(r);
+    set_err(error_code, 0);
+}
+
+void sirius_get_fv_eigen_vectors(void* const* ks_handler, int const* ik,
+                                 void* fv_evec, int const* ld,
+                                 int const* num_fv_states,
+                                 int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("get_fv_eigen_vectors", Py_BuildValue(
+        "(Oii)", (PyObject*)*ks_handler, *ik - 1, *num_fv_states));
+    if (!r) { set_err(error_code, 1); return; }
+    int nbasis = (int)PyLong_AsLong(PyTuple_GetItem(r, 0));
+    PyObject* flat = PyTuple_GetItem(r, 1);
+    PyObject* fast = PySequence_Fast(flat, "seq");
+    double* out = (double*)fv_evec;
+    int stride = ld ? *ld : nbasis;
+    for (int j = 0; j < *num_fv_states; j++)
+        for (int i = 0; i < nbasis; i++) {
+            out[2 * (j * stride + i)] = PyFloat_AsDouble(
+                PySequence_Fast_GET_ITEM(fast, 2 * (j * nbasis + i)));
+            out[2 * (j * stride + i) + 1] = PyFloat_AsDouble(
+                PySequence_Fast_GET_ITEM(fast, 2 * (j * nbasis + i) + 1));
+        }
+    Py_DECREF(fast);
+    Py_DECREF(r);
+    set_err(error_code, 0);
+}
+
+void sirius_get_psi(void* const* ks_handler, int* ik, int* ispin,
+                    void* psi, int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("get_psi", Py_BuildValue(
+        "(Oii)", (PyObject*)*ks_handler, *ik - 1, *ispin - 1));
+    if (!r) { set_err(error_code, 1); return; }
+    doubles_from_seq(PyTuple_GetItem(r, 2), (double*)psi);
+    Py_DECREF(r);
+    set_err(error_code, 0);
+}
+
 }  // extern "C"
+
 
 
 

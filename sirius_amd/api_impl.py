@@ -919,3 +919,115 @@ def generate_xc_potential(gs: GsHandle):
     sirius_generate_xc_potential regenerates only XC — here the full
     potential is rebuilt, which subsumes it)."""
     gs.dft.potential.generate(gs.dft.density)
+
+
+# ---- Hubbard setup, eigen-vector export, misc (batch 4) ----------------
+
+def set_atom_type_hubbard(h: CtxHandle, label: str, l: int, n: int,
+                          occ: float, U: float, J: float, alpha: float,
+                          beta: float, J0: float):
+    """Add a Hubbard-corrected orbital for an atom type (reference
+    sirius_set_atom_type_hubbard; accumulates hubbard.local entries)."""
+    hub = h.cfg_data.setdefault("hubbard", {})
+    hub.setdefault("local", []).append({
+        "atom_type": label, "l": int(l), "n": int(n),
+        "total_initial_occupancy": float(occ), "U": float(U),
+        "J": float(J), "alpha": float(alpha), "beta": float(beta),
+        "J0": float(J0)})
+    h.cfg_data.setdefault("parameters", {})["hubbard_correction"] = True
+
+
+def add_hubbard_atom_pair(h: CtxHandle, atom_pair, translation, n, l,
+                          coupling: float):
+    hub = h.cfg_data.setdefault("hubbard", {})
+    hub.setdefault("nonlocal", []).append({
+        "atom_pair": [int(atom_pair[0]) - 1, int(atom_pair[1]) - 1],
+        "T": [int(x) for x in translation],
+        "n": [int(x) for x in n], "l": [int(x) for x in l],
+        "V": float(coupling)})
+
+
+def add_hubbard_atom_constraint(h: CtxHandle, atom_id: int, n: int, l: int,
+                                lmax_at: int, occ, orbital_order=None):
+    hub = h.cfg_data.setdefault("hubbard", {})
+    mm = 2 * int(l) + 1
+    nsp = 2 if h.cfg_data.get("parameters", {}).get("num_mag_dims", 0) \
+        else 1
+    occm = np.asarray(occ, dtype=np.float64).reshape(nsp, mm, mm)
+    entry = {"atom_index": int(atom_id) - 1, "n": int(n), "l": int(l),
+             "occupancy": occm.tolist()}
+    if orbital_order is not None:
+        entry["lm_order"] = [int(x) for x in orbital_order]
+    hub.setdefault("local_constraint", []).append(entry)
+    hub["constrained_calculation"] = True
+
+
+def get_comm_handle(h: CtxHandle, which: str) -> int:
+    """Fortran communicator handles (reference get_kpoint_inner/inter/
+    fft_comm).  This engine drives distribution through torch.distributed
+    (RCCL/gloo), not MPI — callers embedding serially get MPI_COMM_SELF's
+    conventional f-handle 0."""
+    return 0
+
+
+def set_energy_fermi(ks: KsetHandle, ef: float):
+    ks.kset.energy_fermi = float(ef)
+
+
+def check_scf_density(gs: GsHandle) -> float:
+    """Regenerate rho from the current wave functions and report the max
+    |Δrho(G)| against the stored density (reference
+    sirius_check_scf_density prints the same check)."""
+    import torch
+
+    from .hamiltonian import Hamiltonian0
+
+    dft = gs.dft
+    old = dft.density.rho_g.clone()
+    h0 = Hamiltonian0(dft.ctx, dft.potential, dft.density)
+    dft.density.generate(dft.kset, h0)
+    d = float((dft.density.rho_g - old).abs().max())
+    print(f"sirius_amd: check_scf_density |drho(G)|_max = {d:.3e}")
+    return d
+
+
+def get_step_function(h: CtxHandle):
+    """LAPW unit-step function Θ: (PW coeffs interleaved, real-grid
+    values, n_rg) — FP context only."""
+    ctx = h.ctx
+    if not getattr(ctx, "theta_pw", None) is not None:
+        raise ValueError("step function only exists for the LAPW branch")
+    pw = ctx.theta_pw.detach().cpu().numpy()
+    rg = ctx.theta_rg.detach().cpu().numpy()
+    flat = np.empty(2 * pw.size)
+    flat[0::2] = pw.real
+    flat[1::2] = pw.imag
+    return (flat.tolist(),
+            np.asarray(rg, dtype=np.float64).transpose(2, 1, 0)
+            .reshape(-1).tolist(), int(rg.size))
+
+
+def get_fv_eigen_vectors(ks: KsetHandle, ik: int, num_fv_states: int):
+    """First-variational eigenvectors [N_basis, nfv] (LAPW), interleaved
+    re/im, column-major like the reference."""
+    kp = ks.kset.kpoints[ik]
+    v = getattr(kp, "fv_evec", None)
+    if v is None:
+        raise ValueError("no first-variational eigenvectors on this kp")
+    v = np.asarray(v)[:, :num_fv_states]
+    flat = np.empty(2 * v.size)
+    a = v.reshape(-1, order="F")
+    flat[0::2] = a.real
+    flat[1::2] = a.imag
+    return int(v.shape[0]), flat.tolist()
+
+
+def get_psi(ks: KsetHandle, ik: int, ispin: int):
+    """Full wave-function coefficients of one k-point/spin, interleaved
+    re/im, band-major (reference sirius_get_psi)."""
+    kp = ks.kset.kpoints[ik]
+    psi = kp.psi[ispin].detach().cpu().numpy()
+    flat = np.empty(2 * psi.size)
+    flat[0::2] = psi.real.reshape(-1)
+    flat[1::2] = psi.imag.reshape(-1)
+    return int(psi.shape[1]), int(psi.shape[0]), flat.tolist()
