@@ -1403,7 +1403,197 @@ void sirius_get_psi(void* const* ks_handler, int* ik, int* ispin,
     set_err(error_code, 0);
 }
 
+
+void sirius_get_parameters(
+    void* const* handler, int* lmax_apw, int* lmax_rho, int* lmax_pot,
+    int* num_fv_states, int* num_bands, int* num_spins, int* num_mag_dims,
+    double* pw_cutoff, double* gk_cutoff, int* fft_grid_size, int* auto_rmt,
+    bool* gamma_point, bool* use_symmetry, bool* so_correction,
+    double* iter_solver_tol, double* iter_solver_tol_empty, int* verbosity,
+    bool* hubbard_correction, double* evp_work_count,
+    int* num_loc_op_applied, int* num_sym_op,
+    char* electronic_structure_method, int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("get_parameters",
+                            Py_BuildValue("(O)", (PyObject*)*handler));
+    if (!r) { set_err(error_code, 1); return; }
+    auto gi = [&](const char* k, int* out) {
+        if (out) {
+            PyObject* v = PyDict_GetItemString(r, k);
+            if (v) *out = (int)PyLong_AsLong(v);
+        }
+    };
+    auto gd = [&](const char* k, double* out) {
+        if (out) {
+            PyObject* v = PyDict_GetItemString(r, k);
+            if (v) *out = PyFloat_AsDouble(v);
+        }
+    };
+    auto gb = [&](const char* k, bool* out) {
+        if (out) {
+            PyObject* v = PyDict_GetItemString(r, k);
+            if (v) *out = PyObject_IsTrue(v);
+        }
+    };
+    gi("lmax_apw", lmax_apw);
+    gi("lmax_rho", lmax_rho);
+    gi("lmax_pot", lmax_pot);
+    gi("num_fv_states", num_fv_states);
+    gi("num_bands", num_bands);
+    gi("num_spins", num_spins);
+    gi("num_mag_dims", num_mag_dims);
+    gd("pw_cutoff", pw_cutoff);
+    gd("gk_cutoff", gk_cutoff);
+    if (fft_grid_size) {
+        PyObject* v = PyDict_GetItemString(r, "fft_grid_size");
+        for (int i = 0; v && i < 3; i++) {
+            PyObject* it = PySequence_GetItem(v, i);
+            fft_grid_size[i] = (int)PyLong_AsLong(it);
+            Py_DECREF(it);
+        }
+    }
+    gi("auto_rmt", auto_rmt);
+    gb("gamma_point", gamma_point);
+    gb("use_symmetry", use_symmetry);
+    gb("so_correction", so_correction);
+    gd("iter_solver_tol", iter_solver_tol);
+    gd("iter_solver_tol_empty", iter_solver_tol_empty);
+    gi("verbosity", verbosity);
+    gb("hubbard_correction", hubbard_correction);
+    gd("evp_work_count", evp_work_count);
+    gi("num_loc_op_applied", num_loc_op_applied);
+    gi("num_sym_op", num_sym_op);
+    if (electronic_structure_method) {
+        PyObject* v = PyDict_GetItemString(r,
+                                           "electronic_structure_method");
+        const char* sv = v ? PyUnicode_AsUTF8(v) : "";
+        std::strcpy(electronic_structure_method, sv ? sv : "");
+    }
+    Py_DECREF(r);
+    set_err(error_code, 0);
+}
+
+void sirius_initialize_kset(void* const* ks_handler, int* count,
+                            int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("initialize_kset",
+                            Py_BuildValue("(O)", (PyObject*)*ks_handler));
+    if (!r) { set_err(error_code, 1); return; }
+    if (count) *count = (int)PyLong_AsLong(r);
+    Py_DECREF(r);
+    set_err(error_code, 0);
+}
+
+void sirius_create_hamiltonian(void* const* gs_handler, void** H0_handler,
+                               int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("create_hamiltonian",
+                            Py_BuildValue("(O)", (PyObject*)*gs_handler));
+    if (!r) { set_err(error_code, 1); return; }
+    *H0_handler = r;
+    set_err(error_code, 0);
+}
+
+void sirius_diagonalize_hamiltonian(
+    void* const* handler, void* const* gs_handler, void* const* H0_handler,
+    double* const iter_solver_tol, int* const max_steps,
+    int* converge_by_energy, bool* const exact_diagonalization,
+    bool* converged, int* niter, int* error_code) {
+    Gil g;
+    (void)handler; (void)converge_by_energy;
+    PyObject* r = call_impl("diagonalize_hamiltonian", Py_BuildValue(
+        "(OOdii)", (PyObject*)*gs_handler, (PyObject*)*H0_handler,
+        iter_solver_tol ? *iter_solver_tol : 1e-5,
+        max_steps ? *max_steps : 20,
+        exact_diagonalization ? (int)*exact_diagonalization : 0));
+    if (!r) { set_err(error_code, 1); return; }
+    if (converged) *converged = PyObject_IsTrue(PyTuple_GetItem(r, 0));
+    if (niter) *niter = (int)PyLong_AsLong(PyTuple_GetItem(r, 1));
+    Py_DECREF(r);
+    set_err(error_code, 0);
+}
+
+void sirius_generate_d_operator_matrix(void* const* gs_handler,
+                                       int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("generate_d_operator_matrix",
+                            Py_BuildValue("(O)", (PyObject*)*gs_handler));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_set_atom_type_radial_grid_inf(void* const* handler,
+                                          char const* label,
+                                          int const* num_radial_points,
+                                          double const* radial_points,
+                                          int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("set_atom_type_radial_grid_inf", Py_BuildValue(
+        "(OsN)", (PyObject*)*handler, label,
+        list_from_doubles(radial_points, *num_radial_points)));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_get_gkvec(void* const* ks_handler, int* ik, double* gvec,
+                      int* error_code) {
+    Gil g;
+    PyObject* r = call_impl("get_gkvec", Py_BuildValue(
+        "(Oi)", (PyObject*)*ks_handler, *ik - 1));
+    if (!r) { set_err(error_code, 1); return; }
+    doubles_from_seq(r, gvec);
+    Py_DECREF(r);
+    set_err(error_code, 0);
+}
+
+void sirius_set_local_occupation_matrix(void** handler, int const* ia,
+                                        int const* n, int const* l,
+                                        int const* spin, void* occ_mtrx,
+                                        int const* ld, int* error_code) {
+    Gil g;
+    int mm = 2 * (*l) + 1;
+    PyObject* r = call_impl("set_local_occupation_matrix", Py_BuildValue(
+        "(OiiiiNi)", (PyObject*)*handler, *ia - 1, *n, *l, *spin - 1,
+        list_from_doubles((double*)occ_mtrx, 2 * (*ld) * mm), *ld));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_set_nonlocal_occupation_matrix(void** handler,
+                                           int const* atom_pair,
+                                           int const* n, int const* l,
+                                           int const* spin, int const* T,
+                                           void* occ_mtrx, int const* ld1,
+                                           int const* ld2,
+                                           int* error_code) {
+    Gil g;
+    // 1-based atom indices from Fortran
+    int ap[2] = {atom_pair[0] - 1, atom_pair[1] - 1};
+    PyObject* r = call_impl("set_nonlocal_occupation_matrix", Py_BuildValue(
+        "(ONNNiNNii)", (PyObject*)*handler, list_from_ints(ap, 2),
+        list_from_ints(n, 2), list_from_ints(l, 2), *spin - 1,
+        list_from_ints(T, 3),
+        list_from_doubles((double*)occ_mtrx, 2 * (*ld1) * (*ld2)),
+        *ld1, *ld2));
+    set_err(error_code, r ? 0 : 1);
+    Py_XDECREF(r);
+}
+
+void sirius_get_sv_eigen_vectors(void* const* ks_handler, int const* ik,
+                                 void* sv_evec, int const* num_bands,
+                                 int* error_code) {
+    Gil g;
+    (void)num_bands;
+    PyObject* r = call_impl("get_sv_eigen_vectors", Py_BuildValue(
+        "(Oi)", (PyObject*)*ks_handler, *ik - 1));
+    if (!r) { set_err(error_code, 1); return; }
+    doubles_from_seq(PyTuple_GetItem(r, 2), (double*)sv_evec);
+    Py_DECREF(r);
+    set_err(error_code, 0);
+}
+
 }  // extern "C"
+
 
 
 

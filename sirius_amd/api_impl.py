@@ -514,16 +514,30 @@ def get_parameters(h: CtxHandle):
     """Scalar parameter snapshot (reference sirius_get_parameters)."""
     ctx = h.ctx
     p = ctx.cfg.parameters
+    itso = ctx.cfg.iterative_solver
     return {
+        "lmax_apw": int(p.lmax_apw),
+        "lmax_rho": int(p.lmax_rho),
+        "lmax_pot": int(p.lmax_pot),
         "num_bands": int(ctx.num_bands),
         "num_spins": int(ctx.num_spins),
         "num_mag_dims": int(ctx.num_mag_dims),
         "pw_cutoff": float(ctx.pw_cutoff),
         "gk_cutoff": float(ctx.gk_cutoff),
         "fft_grid_size": list(ctx.fft_fine.dims),
+        "auto_rmt": int(p.auto_rmt),
         "gamma_point": bool(p.gamma_point),
         "use_symmetry": bool(p.use_symmetry),
         "so_correction": bool(getattr(p, "so_correction", False)),
+        "iter_solver_tol": float(itso.get("energy_tolerance", 1e-2)),
+        "iter_solver_tol_empty": float(
+            itso.get("empty_states_tolerance", 0.0)),
+        "verbosity": int(ctx.cfg.control.verbosity),
+        "hubbard_correction": bool(ctx.hubbard is not None),
+        "evp_work_count": float(ctx.counters.get(
+            "band_evp_work_count", 0.0)),
+        "num_loc_op_applied": int(ctx.counters.get(
+            "local_operator_num_applied", 0)),
         "electronic_structure_method": str(p.electronic_structure_method),
         "num_sym_op": int(len(ctx.symmetry.ops)) if getattr(
             ctx, "symmetry", None) else 0,
@@ -1031,3 +1045,126 @@ def get_psi(ks: KsetHandle, ik: int, ispin: int):
     flat[0::2] = psi.real.reshape(-1)
     flat[1::2] = psi.imag.reshape(-1)
     return int(psi.shape[1]), int(psi.shape[0]), flat.tolist()
+
+
+# ---- batch 5: Hamiltonian handlers, kset init, occupation setters ------
+
+class HamHandle:
+    def __init__(self, h0):
+        self.h0 = h0
+
+
+def initialize_kset(ks: KsetHandle):
+    """No-op: this engine's k-set is fully built at creation (reference
+    sirius_initialize_kset splits construction in two)."""
+    return int(ks.kset.num_kpoints)
+
+
+def create_hamiltonian(gs: GsHandle):
+    from .hamiltonian import Hamiltonian0
+
+    dft = gs.dft
+    return HamHandle(Hamiltonian0(dft.ctx, dft.potential, dft.density))
+
+
+def diagonalize_hamiltonian(gs: GsHandle, H0: HamHandle,
+                            iter_solver_tol: float, max_steps: int,
+                            exact: bool = False):
+    from .dft import diagonalize
+
+    dft = gs.dft
+    itso = dft.ctx.cfg.iterative_solver
+    saved = itso.get("num_steps"), itso.get("type")
+    try:
+        itso.set("num_steps", int(max_steps))
+        if exact:
+            itso.set("type", "exact")
+        conv = diagonalize(dft.ctx, H0.h0, dft.kset,
+                           float(iter_solver_tol))
+    finally:
+        itso.set("num_steps", saved[0])
+        itso.set("type", saved[1])
+    nit = int(dft.ctx.counters.get("num_itsol_steps", 0))
+    return bool(conv), nit
+
+
+def generate_d_operator_matrix(gs: GsHandle):
+    """The D matrices are (re)built with the Hamiltonian from the current
+    potential (reference sirius_generate_d_operator_matrix) — rebuild and
+    cache on the handler so a following create_hamiltonian is warm."""
+    create_hamiltonian(gs)
+
+
+def set_atom_type_radial_grid_inf(h: CtxHandle, label: str, points):
+    at = h.types[label]
+    at.r_inf = np.asarray(points, dtype=np.float64)
+
+
+def get_gkvec(ks: KsetHandle, ik: int):
+    """Cartesian G+k vectors of one k-point, flat [3*ngk]."""
+    kp = ks.kset.kpoints[ik]
+    return np.asarray(kp.gkvec.gkvec_cart,
+                      dtype=np.float64).reshape(-1).tolist()
+
+
+def set_local_occupation_matrix(h, ia: int, n: int, l: int, spin: int,
+                                occ_flat, ld: int):
+    """Set one local Hubbard occupation block (reference
+    sirius_set_local_occupation_matrix; handler is the ground state)."""
+    import torch
+
+    dft = h.dft if isinstance(h, GsHandle) else h
+    hub = dft.ctx.hubbard
+    mm = 2 * int(l) + 1
+    il = hub._find_level(int(ia), int(n), int(l))
+    c = np.asarray(occ_flat, dtype=np.float64)
+    m = (c[0::2] + 1j * c[1::2]).reshape(ld, -1)[:mm, :mm]
+    if hub.om is None:
+        hub.om = hub.initial_occupation()
+    hub.om[il][..., int(spin)] = torch.from_numpy(m).to(
+        hub.om[il].device, hub.om[il].dtype)
+
+
+def set_nonlocal_occupation_matrix(h, atom_pair, n, l, spin: int, T,
+                                   occ_flat, ld1: int, ld2: int):
+    """Set one inter-site occupation block."""
+    import torch
+
+    dft = h.dft if isinstance(h, GsHandle) else h
+    hub = dft.ctx.hubbard
+    m1 = 2 * int(l[0]) + 1
+    m2 = 2 * int(l[1]) + 1
+    c = np.asarray(occ_flat, dtype=np.float64)
+    blk = (c[0::2] + 1j * c[1::2]).reshape(ld1, ld2)[:m1, :m2]
+    tgt = (int(atom_pair[0]), int(atom_pair[1]), tuple(int(x) for x in T))
+    for i, p in enumerate(hub.nonlocal_pairs):
+        if (p.ia, p.ja, p.T) == tgt and p.n1 == int(n[0]) \
+                and p.n2 == int(n[1]):
+            if hub.om_nl is None:
+                hub.om_nl = [torch.zeros(2 * q.il + 1, 2 * q.jl + 1,
+                                         dft.ctx.num_spins,
+                                         dtype=dft.ctx.dtype)
+                             for q in hub.nonlocal_pairs]
+            hub.om_nl[i][..., int(spin)] = torch.from_numpy(blk).to(
+                hub.om_nl[i].device, hub.om_nl[i].dtype)
+            return
+    raise ValueError(f"nonlocal pair {tgt} not found")
+
+
+def get_sv_eigen_vectors(ks: KsetHandle, ik: int):
+    """Second-variational eigenvectors (LAPW collinear: block-diagonal
+    per-spin rotation), interleaved re/im, column-major per spin."""
+    kp = ks.kset.kpoints[ik]
+    sv = getattr(kp, "sv_evec", None)
+    if sv is None or sv[0] is None:
+        raise ValueError("no second-variational eigenvectors on this kp")
+    mats = [np.asarray(m) for m in sv if m is not None]
+    flat_parts = []
+    for v in mats:
+        a = v.reshape(-1, order="F")
+        f = np.empty(2 * a.size)
+        f[0::2] = a.real
+        f[1::2] = a.imag
+        flat_parts.append(f)
+    return int(mats[0].shape[0]), len(mats), \
+        np.concatenate(flat_parts).tolist()
