@@ -343,6 +343,24 @@ API = [
       ("cfunrg", "double(*)_out", "out"),
       ("num_rg_points", "int_out", "out")],
      "LAPW unit-step function (PW re/im interleaved + real grid)."),
+    ("sirius_initialize_kset",
+     [("ks_handler", "handler", "in"), ("count", "int_out", "out")],
+     "Finalize k-set construction (no-op here; returns num_kpoints)."),
+    ("sirius_create_hamiltonian",
+     [("gs_handler", "handler", "in"), ("H0_handler", "handler_out", "out")],
+     "Create a Hamiltonian handler from the current potential."),
+    ("sirius_generate_d_operator_matrix",
+     [("gs_handler", "handler", "in")],
+     "Rebuild the D-operator matrices from the current potential."),
+    ("sirius_set_atom_type_radial_grid_inf",
+     [("handler", "handler", "in"), ("label", "string", "in"),
+      ("num_radial_points", "int", "in"),
+      ("radial_points", "double(*)", "in")],
+     "Set the extended (free-atom) radial grid of an atom type."),
+    ("sirius_get_gkvec",
+     [("ks_handler", "handler", "in"), ("ik", "int", "in"),
+      ("gvec", "double(*)_out", "out")],
+     "Cartesian G+k vectors of one k-point (1-based ik)."),
 ]
 
 FT = {
