@@ -48,7 +48,9 @@ def test08_si_uspp_lda():
 @requires_reference
 @pytest.mark.slow
 def test06_fe_uspp_lda_fm():
-    res, eref = run_case("test06")
+    # 60 iterations: the FM trajectory is marginal at 40 (roundoff-level
+    # changes flip convergence between iteration 39 and 41)
+    res, eref = run_case("test06", num_iter=60)
     assert res["converged"]
     assert abs(res["energy"]["total"] - eref) < 1e-5
     assert abs(res["magnetization"] - 6.760705375907565) < 1e-3
